@@ -1,0 +1,151 @@
+"""HIP (gfx950) kernel extension: loading and high-level wrappers.
+
+The extension is built IN-TREE (sparse/kernels/_build/_sparse_hip.so) so the
+.so travels to GPU boxes with the repo snapshot.  Build with
+`python -m sparse.kernels.build` or __graft_entry__.build().
+
+Policy: on a GPU, ops REQUIRE this extension — require() raises if it is
+missing.  There is no eager/PyTorch fallback on the GPU path.
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+_EXT = None
+_TRIED = False
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+BUILD_DIR = os.path.join(_DIR, "_build")
+SO_PATH = os.path.join(BUILD_DIR, "sparse_hip.so")
+
+
+def ext():
+    """Return the loaded extension module, or None if unavailable."""
+    global _EXT, _TRIED
+    if _EXT is not None or _TRIED:
+        return _EXT
+    _TRIED = True
+    if os.path.exists(SO_PATH):
+        torch.ops.load_library(SO_PATH)
+        _EXT = torch.ops.sparse_hip
+    return _EXT
+
+
+def require():
+    if ext() is None:
+        raise RuntimeError(
+            "sparse: GPU op invoked but the HIP extension is not built "
+            f"(expected {SO_PATH}). Build it with `python -m sparse.kernels.build`."
+        )
+
+
+# -- high-level wrappers ------------------------------------------------------
+def spmv(A, x, y, col_lo: int, beta: float):
+    ext().spmv(A.indptr, A.indices, A.values, x, y, int(col_lo), float(beta))
+
+
+def spmv_dot(A, x, y, p):
+    return ext().spmv_dot(A.indptr, A.indices, A.values, x, y, p, 0)
+
+
+def spmm(A, B, C, col_lo: int):
+    ext().spmm(A.indptr, A.indices, A.values, B, C, int(col_lo))
+
+
+def rspmm(B, A_dense, C):
+    ext().rspmm(B.indptr, B.indices, B.values, A_dense, C)
+
+
+def sddmm(A, C, D, out):
+    ext().sddmm(A.indptr, A.indices, A.values, C, D, out)
+
+
+def mult_dense(A, D, out):
+    ext().mult_dense(A.indptr, A.indices, A.values, D, out)
+
+
+def csr_to_dense(A, out):
+    ext().csr_to_dense(A.indptr, A.indices, A.values, out)
+
+
+def csr_diagonal(A, out, row_offset: int):
+    ext().csr_diagonal(A.indptr, A.indices, A.values, out, int(row_offset))
+
+
+def csc_spmv(colptr, rowidx, values, x, y, rlo: int):
+    ext().csc_spmv(colptr, rowidx, values, x, y, int(rlo))
+
+
+def csc_spmm(colptr, rowidx, values, B, C, rlo: int):
+    ext().csc_spmm(colptr, rowidx, values, B, C, int(rlo))
+
+
+def tropical_spmv(A, x, y, col_lo: int):
+    ext().tropical_spmv(A.indptr, A.indices, A.values, x, y, int(col_lo))
+
+
+def add_csr(A, B, alpha, beta, vdt):
+    from ..ops.local import LocalCSR
+
+    nnz_per_row = torch.empty(A.nrows, dtype=torch.int64, device=A.device)
+    ext().add_nnz(A.indptr, A.indices, B.indptr, B.indices, nnz_per_row)
+    indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=A.device)
+    torch.cumsum(nnz_per_row, 0, out=indptr[1:])
+    nnz = int(indptr[-1].item())
+    indices = torch.empty(nnz, dtype=A.indices.dtype, device=A.device)
+    values = torch.empty(nnz, dtype=vdt, device=A.device)
+    ext().add_compute(A.indptr, A.indices, A.values.to(vdt),
+                      B.indptr, B.indices, B.values.to(vdt),
+                      indptr, indices, values, float(alpha), float(beta))
+    return LocalCSR(indptr, indices, values, A.nrows, A.ncols)
+
+
+def elem_mult_csr(A, B, vdt):
+    from ..ops.local import LocalCSR
+
+    nnz_per_row = torch.empty(A.nrows, dtype=torch.int64, device=A.device)
+    ext().mult_nnz(A.indptr, A.indices, B.indptr, B.indices, nnz_per_row)
+    indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=A.device)
+    torch.cumsum(nnz_per_row, 0, out=indptr[1:])
+    nnz = int(indptr[-1].item())
+    indices = torch.empty(nnz, dtype=A.indices.dtype, device=A.device)
+    values = torch.empty(nnz, dtype=vdt, device=A.device)
+    ext().mult_compute(A.indptr, A.indices, A.values.to(vdt),
+                       B.indptr, B.indices, B.values.to(vdt),
+                       indptr, indices, values)
+    return LocalCSR(indptr, indices, values, A.nrows, A.ncols)
+
+
+def spgemm_csr(A, B, a_col_lo, vdt):
+    from ..ops.local import LocalCSR
+
+    nnz_per_row = torch.empty(A.nrows, dtype=torch.int64, device=A.device)
+    ext().spgemm_nnz(A.indptr, A.indices, B.indptr, B.indices, nnz_per_row,
+                     int(a_col_lo), int(B.ncols))
+    indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=A.device)
+    torch.cumsum(nnz_per_row, 0, out=indptr[1:])
+    nnz = int(indptr[-1].item())
+    indices = torch.empty(nnz, dtype=A.indices.dtype, device=A.device)
+    values = torch.empty(nnz, dtype=vdt, device=A.device)
+    ext().spgemm_compute(A.indptr, A.indices, A.values.to(vdt),
+                         B.indptr, B.indices, B.values.to(vdt),
+                         indptr, indices, values, int(a_col_lo), int(B.ncols))
+    return LocalCSR(indptr, indices, values, A.nrows, B.ncols)
+
+
+def axpby(y, x, a, b, isalpha: bool, negate: bool):
+    """Fused y = y + (±a/b) x  (isalpha) or y = x + (±a/b) y.
+    Reference: AXPBY task (axpby.cu:25-42, linalg.py:479-496)."""
+    ext().axpby(y, x, a, b, bool(isalpha), bool(negate))
+
+
+def rk_calc_dy(K, a, h, dy):
+    """dy[i] = h * sum_j K[j,i] * a[j] (reference runge_kutta.cu:26-42)."""
+    ext().rk_calc_dy(K, a, float(h), dy)
+
+
+def cdist(XA, XB, out):
+    ext().cdist(XA, XB, out)

@@ -1,0 +1,347 @@
+"""Local (per-rank) sparse kernels: dispatch between the HIP extension and a
+CPU path.
+
+Every GPU entry point REQUIRES the in-tree HIP extension (sparse/kernels) —
+there is no silent eager fallback on a GPU box.  The CPU path uses
+scipy/torch and exists for the GPU-less CI environment and as the numerics
+oracle.
+
+Column indices are GLOBAL; ops that consume a gathered x-window take a
+`col_lo` offset and subtract it on the fly (free ALU on the GPU; the
+reference does the same rebase in convertGlobalPosToLocalIndPtr,
+src/sparse/util/cusparse_utils.h:28-38).
+
+Reference parity map: each function here corresponds to one or
+more task opcodes of src/sparse/sparse_c.h; citations on each function.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import numpy as np
+import torch
+
+from ..types import index_dtype_for
+
+_scipy = None
+
+
+def sp():
+    global _scipy
+    if _scipy is None:
+        import scipy.sparse as s
+
+        _scipy = s
+    return _scipy
+
+
+def hip():
+    """The HIP kernel wrapper module; raises loudly if not built."""
+    from .. import kernels
+
+    kernels.require()
+    return kernels
+
+
+@dataclass
+class LocalCSR:
+    """One rank's row slab: local 0-based indptr (int64), GLOBAL column ids."""
+
+    indptr: torch.Tensor
+    indices: torch.Tensor
+    values: torch.Tensor
+    nrows: int
+    ncols: int  # global number of columns
+
+    @property
+    def nnz(self) -> int:
+        return int(self.values.numel())
+
+    @property
+    def device(self):
+        return self.values.device
+
+    @property
+    def dtype(self):
+        return self.values.dtype
+
+    def to_scipy(self, col_lo: int = 0, width: Optional[int] = None):
+        """CPU oracle view (moves to host); columns rebased by col_lo."""
+        idx = self.indices.detach().cpu().numpy()
+        if col_lo:
+            idx = idx - col_lo
+        w = width if width is not None else self.ncols - col_lo
+        return sp().csr_matrix(
+            (self.values.detach().cpu().numpy(), idx,
+             self.indptr.detach().cpu().numpy()),
+            shape=(self.nrows, w),
+        )
+
+    @staticmethod
+    def from_scipy(m, device, vdtype=None, idtype=None, ncols=None) -> "LocalCSR":
+        m = m.tocsr()
+        m.sort_indices()
+        shape = (m.shape[0], ncols if ncols is not None else m.shape[1])
+        idt = idtype or index_dtype_for(shape)
+        vals = torch.as_tensor(m.data, device=device)
+        if vdtype is not None:
+            vals = vals.to(vdtype)
+        return LocalCSR(
+            torch.as_tensor(m.indptr.astype(np.int64), device=device),
+            torch.as_tensor(m.indices, device=device).to(idt),
+            vals, shape[0], shape[1])
+
+    def clone(self) -> "LocalCSR":
+        return LocalCSR(self.indptr.clone(), self.indices.clone(),
+                        self.values.clone(), self.nrows, self.ncols)
+
+
+def is_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+# -- SpMV ---------------------------------------------------------------------
+def spmv(A: LocalCSR, x: torch.Tensor, col_lo: int = 0,
+         y: Optional[torch.Tensor] = None, beta: float = 0.0) -> torch.Tensor:
+    """y = A @ x[window] (+ beta*y).  Reference: CSR_SPMV_ROW_SPLIT
+    (src/sparse/array/csr/spmv.cu:25-123).  GPU: nnz-split HIP kernel."""
+    vdt = torch.promote_types(A.dtype, x.dtype)
+    if y is None:
+        y = torch.empty(A.nrows, dtype=vdt, device=A.device)
+        beta = 0.0
+    if is_gpu(A.values):
+        hip().spmv(A, x, y, col_lo, float(beta))
+        return y
+    m = A.to_scipy(col_lo=col_lo, width=x.shape[0])
+    r = m @ x.detach().cpu().numpy()
+    rt = torch.as_tensor(r, device=A.device).to(y.dtype)
+    if beta == 0.0:
+        y.copy_(rt)
+    else:
+        y.mul_(beta).add_(rt)
+    return y
+
+
+# -- SpMM / rSpMM / SDDMM -----------------------------------------------------
+def spmm(A: LocalCSR, B: torch.Tensor, col_lo: int = 0) -> torch.Tensor:
+    """C = A @ B[window rows].  Reference: SPMM_CSR_DENSE (spmm.cu:26-117)."""
+    vdt = torch.promote_types(A.dtype, B.dtype)
+    if is_gpu(A.values):
+        C = torch.empty((A.nrows, B.shape[1]), dtype=vdt, device=A.device)
+        hip().spmm(A, B.contiguous(), C, col_lo)
+        return C
+    m = A.to_scipy(col_lo=col_lo, width=B.shape[0])
+    r = m @ B.detach().cpu().numpy()
+    return torch.as_tensor(np.ascontiguousarray(r), device=A.device).to(vdt)
+
+
+def rspmm(A_dense: torch.Tensor, B: LocalCSR) -> torch.Tensor:
+    """C = A_dense @ B_csr, dense k-slab x local B rows -> full-width partial C
+    (reduced by the caller).  Reference: SPMM_DENSE_CSR (spmm.cu:115-180)."""
+    vdt = torch.promote_types(B.dtype, A_dense.dtype)
+    if is_gpu(B.values):
+        C = torch.zeros((A_dense.shape[0], B.ncols), dtype=vdt, device=B.device)
+        hip().rspmm(B, A_dense.contiguous(), C)
+        return C
+    m = B.to_scipy()
+    r = A_dense.detach().cpu().numpy() @ m
+    return torch.as_tensor(np.ascontiguousarray(r), device=B.device).to(vdt)
+
+
+def sddmm(A: LocalCSR, C: torch.Tensor, D: torch.Tensor) -> torch.Tensor:
+    """out_vals[nz at (i,j)] = A.values[nz] * (C[i,:] @ D[:,j]).
+    Reference: CSR_SDDMM (sddmm.cu:25-85)."""
+    if is_gpu(A.values):
+        out = torch.empty_like(A.values)
+        hip().sddmm(A, C.contiguous(), D.contiguous(), out)
+        return out
+    m = A.to_scipy().tocoo()
+    Cn = C.detach().cpu().numpy()
+    Dn = D.detach().cpu().numpy()
+    vals = m.data * np.einsum("ij,ji->i", Cn[m.row], Dn[:, m.col])
+    return torch.as_tensor(vals, device=A.device)
+
+
+# -- elementwise --------------------------------------------------------------
+def add(A: LocalCSR, B: LocalCSR, alpha=1.0, beta=1.0) -> LocalCSR:
+    """Union add alpha*A + beta*B on aligned row slabs, two-phase.
+    Reference: ADD_CSR_CSR_NNZ / ADD_CSR_CSR (add.cu)."""
+    vdt = torch.promote_types(A.dtype, B.dtype)
+    if is_gpu(A.values):
+        return hip().add_csr(A, B, alpha, beta, vdt)
+    a, b = A.to_scipy(), B.to_scipy()
+    r = (a.astype(np.dtype(_np_of(vdt))) * alpha + b.astype(np.dtype(_np_of(vdt))) * beta).tocsr()
+    r.sort_indices()
+    return LocalCSR.from_scipy(r, A.device, vdtype=vdt, idtype=A.indices.dtype,
+                               ncols=A.ncols)
+
+
+def _np_of(t: torch.dtype):
+    from ..types import to_numpy_dtype
+
+    return to_numpy_dtype(t)
+
+
+def elem_mult(A: LocalCSR, B: LocalCSR) -> LocalCSR:
+    """Intersection multiply.  Reference: ELEM_MULT_CSR_CSR (mult.cu:27-109)."""
+    vdt = torch.promote_types(A.dtype, B.dtype)
+    if is_gpu(A.values):
+        return hip().elem_mult_csr(A, B, vdt)
+    a, b = A.to_scipy(), B.to_scipy()
+    r = a.multiply(b).tocsr()
+    r.sort_indices()
+    return LocalCSR.from_scipy(r, A.device, vdtype=vdt, idtype=A.indices.dtype,
+                               ncols=A.ncols)
+
+
+def mult_dense(A: LocalCSR, D: torch.Tensor) -> torch.Tensor:
+    """vals'[nz at (i,j)] = vals[nz] * D[i, j] (D = this slab's dense rows,
+    full global width).  Reference: ELEM_MULT_CSR_DENSE (mult_dense.cu)."""
+    if is_gpu(A.values):
+        out = torch.empty_like(A.values)
+        hip().mult_dense(A, D.contiguous(), out)
+        return out
+    m = A.to_scipy().tocoo()
+    Dn = D.detach().cpu().numpy()
+    vals = m.data * Dn[m.row, m.col]
+    return torch.as_tensor(vals, device=A.device)
+
+
+# -- SpGEMM -------------------------------------------------------------------
+def spgemm(A: LocalCSR, B: LocalCSR, a_col_lo: int = 0) -> LocalCSR:
+    """C = A @ B where B holds rows [a_col_lo, a_col_lo + B.nrows) of the
+    global B (gathered by the caller).  GPU: two-phase Gustavson hash.
+    Reference: SPGEMM_CSR_CSR_CSR_GPU (spgemm_csr_csr_csr.cu:33-272)."""
+    vdt = torch.promote_types(A.dtype, B.dtype)
+    if is_gpu(A.values):
+        return hip().spgemm_csr(A, B, a_col_lo, vdt)
+    a = A.to_scipy(col_lo=a_col_lo, width=B.nrows)
+    b = B.to_scipy()
+    r = (a @ b).tocsr()
+    r.sum_duplicates()
+    r.sort_indices()
+    return LocalCSR.from_scipy(r, A.device, vdtype=vdt, idtype=A.indices.dtype,
+                               ncols=B.ncols)
+
+
+# -- conversions --------------------------------------------------------------
+def csr_to_dense(A: LocalCSR) -> torch.Tensor:
+    """Reference: CSR_TO_DENSE (csr_to_dense.cu)."""
+    if is_gpu(A.values):
+        out = torch.zeros((A.nrows, A.ncols), dtype=A.dtype, device=A.device)
+        hip().csr_to_dense(A, out)
+        return out
+    return torch.as_tensor(A.to_scipy().toarray(), device=A.device)
+
+
+def dense_to_csr(D: torch.Tensor, ncols: Optional[int] = None) -> LocalCSR:
+    """Reference: DENSE_TO_CSR_NNZ / DENSE_TO_CSR (dense_to_csr.cu).
+    Torch-vectorized on both devices (mask + nonzero are HIP kernels under
+    torch; a dedicated kernel buys nothing — conversion is not hot)."""
+    ncols = D.shape[1] if ncols is None else ncols
+    D = D.contiguous()
+    mask = D != 0
+    nnz_per_row = mask.sum(dim=1)
+    indptr = torch.zeros(D.shape[0] + 1, dtype=torch.int64, device=D.device)
+    torch.cumsum(nnz_per_row, 0, out=indptr[1:])
+    idx = mask.nonzero(as_tuple=False)
+    idt = index_dtype_for((D.shape[0], ncols))
+    return LocalCSR(indptr, idx[:, 1].to(idt), D[mask], D.shape[0], ncols)
+
+
+def csr_diagonal(A: LocalCSR, row_offset: int = 0) -> torch.Tensor:
+    """d[i] = A[i+row_offset, i+row_offset] (k=0).  Reference: CSR_DIAGONAL
+    (get_diagonal.cu)."""
+    if is_gpu(A.values):
+        out = torch.zeros(A.nrows, dtype=A.dtype, device=A.device)
+        hip().csr_diagonal(A, out, int(row_offset))
+        return out
+    m = A.to_scipy().tocoo()
+    out = np.zeros(A.nrows, dtype=m.data.dtype)
+    hit = m.col == (m.row + row_offset)
+    out[m.row[hit]] = m.data[hit]
+    return torch.as_tensor(out, device=A.device)
+
+
+def tropical_spmv(A: LocalCSR, x: torch.Tensor, col_lo: int = 0) -> torch.Tensor:
+    """(max, lexicographic) semiring SpMV on int64 multi-field vectors.
+    Reference: CSR_SPMV_ROW_SPLIT_TROPICAL_SEMIRING (tropical_spmv.cu:26-56)."""
+    if is_gpu(A.values):
+        y = torch.zeros((A.nrows, x.shape[1]), dtype=torch.int64, device=A.device)
+        hip().tropical_spmv(A, x.contiguous(), y, col_lo)
+        return y
+    # CPU: per-row lexicographic max over neighbor fields
+    ip = A.indptr.cpu().numpy()
+    ix = (A.indices.cpu().numpy() - col_lo)
+    xn = x.cpu().numpy()
+    y = np.zeros((A.nrows, x.shape[1]), dtype=np.int64)
+    for i in range(A.nrows):
+        best = None
+        for p in range(ip[i], ip[i + 1]):
+            cand = xn[ix[p]]
+            if best is None or tuple(cand) > tuple(best):
+                best = cand
+        if best is not None:
+            y[i] = best
+    return torch.as_tensor(y, device=A.device)
+
+
+def expand_pos_to_coordinates(indptr: torch.Tensor, nnz: int, row_offset: int = 0) -> torch.Tensor:
+    """CSR row pointer -> explicit (global) row ids per nnz.
+    Reference: EXPAND_POS_TO_COORDINATES (pos_to_coordinates.cu)."""
+    counts = indptr[1:] - indptr[:-1]
+    return torch.repeat_interleave(
+        torch.arange(row_offset, row_offset + counts.numel(), device=indptr.device),
+        counts,
+    )
+
+
+def coords_to_indptr(sorted_rows: torch.Tensor, nrows: int, row_offset: int = 0) -> torch.Tensor:
+    """Sorted global row ids -> local indptr.  Reference:
+    SORTED_COORDS_TO_COUNTS + nnz_to_pos (sorted_coords_to_counts.cu,
+    base.py:30-48)."""
+    counts = torch.bincount((sorted_rows.long() - row_offset), minlength=nrows)
+    indptr = torch.zeros(nrows + 1, dtype=torch.int64, device=sorted_rows.device)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    return indptr
+
+
+# -- CSC col-split ops --------------------------------------------------------
+def csc_spmv(colptr: torch.Tensor, rowidx: torch.Tensor, values: torch.Tensor,
+             x_cols: torch.Tensor, rlo: int, rhi: int) -> torch.Tensor:
+    """Partial y over the row window [rlo,rhi): y[r-rlo] += v * x[col].
+    Reference: CSC_SPMV_COL_SPLIT (src/sparse/array/csc/spmv.cu:60-75, with
+    the beta=1 reduction semantics); the caller reduce-scatters the partial
+    into the owners of y."""
+    vdt = torch.promote_types(values.dtype, x_cols.dtype)
+    if is_gpu(values):
+        y = torch.zeros(max(0, rhi - rlo), dtype=vdt, device=values.device)
+        hip().csc_spmv(colptr, rowidx, values, x_cols, y, rlo)
+        return y
+    ncl = colptr.numel() - 1
+    m = sp().csc_matrix(
+        (values.detach().cpu().numpy(), (rowidx.detach().cpu().numpy() - rlo),
+         colptr.detach().cpu().numpy()),
+        shape=(max(0, rhi - rlo), ncl))
+    r = m @ x_cols.detach().cpu().numpy()
+    return torch.as_tensor(r, device=values.device).to(vdt)
+
+
+def csc_spmm(colptr: torch.Tensor, rowidx: torch.Tensor, values: torch.Tensor,
+             B_cols: torch.Tensor, rlo: int, rhi: int) -> torch.Tensor:
+    """Partial C over the row window: C[r-rlo, :] += v * B[col, :].
+    Reference: SPMM_CSC_DENSE (csc/spmm.cu)."""
+    vdt = torch.promote_types(values.dtype, B_cols.dtype)
+    if is_gpu(values):
+        C = torch.zeros((max(0, rhi - rlo), B_cols.shape[1]), dtype=vdt,
+                        device=values.device)
+        hip().csc_spmm(colptr, rowidx, values, B_cols.contiguous(), C, rlo)
+        return C
+    ncl = colptr.numel() - 1
+    m = sp().csc_matrix(
+        (values.detach().cpu().numpy(), (rowidx.detach().cpu().numpy() - rlo),
+         colptr.detach().cpu().numpy()),
+        shape=(max(0, rhi - rlo), ncl))
+    r = m @ B_cols.detach().cpu().numpy()
+    return torch.as_tensor(np.ascontiguousarray(r), device=values.device).to(vdt)

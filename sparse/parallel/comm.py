@@ -1,0 +1,130 @@
+"""Thin collective layer over torch.distributed.
+
+On GPU the backend is "nccl" (== RCCL over xGMI on ROCm); on CPU it is
+"gloo".  gloo has no all_to_all, so variable all-to-all falls back to batched
+isend/irecv pairs — the same pattern works on RCCL (p2p over xGMI).
+
+Replaces: NCCL calls in the reference's samplesort (src/sparse/sort/sort.cu:
+163-322) and every implicit Legion copy/reduction.
+
+Complex dtypes are viewed as real pairs before hitting the wire (gloo and
+RCCL reductions do not support complex).
+"""
+from __future__ import annotations
+
+from typing import List, Sequence
+
+import torch
+import torch.distributed as dist
+
+
+def initialized() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def world_size(group=None) -> int:
+    return dist.get_world_size(group) if initialized() else 1
+
+
+def rank(group=None) -> int:
+    return dist.get_rank(group) if initialized() else 0
+
+
+def _as_real(t: torch.Tensor) -> torch.Tensor:
+    return torch.view_as_real(t).flatten() if t.is_complex() else t
+
+
+def all_reduce_(t: torch.Tensor, op: str = "sum", group=None, async_op: bool = False):
+    """In-place all-reduce; no-op at world size 1. Complex -> viewed as real
+    (valid for sum/min/max-on-abs is NOT handled — sum only for complex)."""
+    if not initialized() or world_size(group) == 1:
+        return None
+    ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
+    if t.is_complex():
+        assert op == "sum", "complex all-reduce supports sum only"
+        return dist.all_reduce(torch.view_as_real(t), op=ops[op], group=group, async_op=async_op)
+    return dist.all_reduce(t, op=ops[op], group=group, async_op=async_op)
+
+
+def all_gather_rows(local: torch.Tensor, counts: Sequence[int], group=None) -> torch.Tensor:
+    """All-gather variable-size row slabs (dim 0) into one global tensor.
+
+    Pads to the max slab so plain equal-shape all_gather works on both
+    backends, then slices.  Fine for the near-equal slabs we use.
+    """
+    ws = world_size(group)
+    if not initialized() or ws == 1:
+        return local
+    local_c = local.contiguous()
+    cplx = local_c.is_complex()
+    wire = torch.view_as_real(local_c) if cplx else local_c
+    maxc = max(counts)
+    tail = wire.shape[1:]
+    pad = torch.empty((maxc, *tail), dtype=wire.dtype, device=wire.device)
+    if wire.shape[0] > 0:
+        pad[: wire.shape[0]].copy_(wire)
+    outs = [torch.empty_like(pad) for _ in range(ws)]
+    dist.all_gather(outs, pad, group=group)
+    pieces = [outs[r][: counts[r]] for r in range(ws)]
+    out = torch.cat(pieces, dim=0)
+    if cplx:
+        out = torch.view_as_complex(out)
+    return out
+
+
+def bcast_(t: torch.Tensor, src: int = 0, group=None) -> None:
+    if not initialized() or world_size(group) == 1:
+        return
+    dist.broadcast(torch.view_as_real(t) if t.is_complex() else t, src=src, group=group)
+
+
+def all_to_all_v(send: List[torch.Tensor], group=None) -> List[torch.Tensor]:
+    """Exchange send[r] -> rank r (1-D tensors); returns recv list indexed by
+    source.  Sizes exchanged via equal-size all_gather.  Zero-size guards
+    mirror the reference's NCCL hang workaround (sort.cu:259-263).
+    """
+    ws = world_size(group)
+    me = rank(group)
+    if ws == 1:
+        return [send[0]]
+    device = send[0].device
+    dtype = send[0].dtype
+    cplx = dtype.is_complex
+    wire = [(_as_real(s.contiguous())) for s in send]
+    wdtype = wire[0].dtype
+    backend = dist.get_backend(group)
+    cdev = device if backend == "nccl" else torch.device("cpu")
+    counts = torch.tensor([int(s.numel()) for s in wire], dtype=torch.int64, device=cdev)
+    all_counts = [torch.zeros(ws, dtype=torch.int64, device=cdev) for _ in range(ws)]
+    dist.all_gather(all_counts, counts, group=group)
+    recv_counts = [int(all_counts[src][me].item()) for src in range(ws)]
+    recv = [torch.empty(c, dtype=wdtype, device=device) for c in recv_counts]
+    if backend == "nccl":
+        dist.all_to_all(recv, wire, group=group)
+    else:
+        reqs = []
+        for peer in range(ws):
+            if peer != me and wire[peer].numel() > 0:
+                reqs.append(dist.isend(wire[peer], dst=peer, group=group))
+        recv[me].copy_(wire[me])
+        for peer in range(ws):
+            if peer != me and recv[peer].numel() > 0:
+                dist.recv(recv[peer], src=peer, group=group)
+        for r in reqs:
+            r.wait()
+    if cplx:
+        recv = [torch.view_as_complex(r.view(-1, 2)) for r in recv]
+    return recv
+
+
+_subgroup_cache = {}
+
+
+def subgroup(nranks: int):
+    """Communicator over ranks [0, nranks) — the reference's machine-scoping
+    (examples/gmg.py:212-218) equivalent: per-GMG-level sub-communicators."""
+    if not initialized() or nranks >= world_size():
+        return None
+    if nranks not in _subgroup_cache:
+        _subgroup_cache[nranks] = dist.new_group(ranks=list(range(nranks)))
+    return _subgroup_cache[nranks]

@@ -1,0 +1,115 @@
+"""The x-gather plan: the explicit MI355X replacement for the reference's
+image partitions.
+
+Reference parity: MinMaxImagePartition (sparse/partition.py:139-208) — each
+row slab of a CSR matrix needs only x[lo:hi) where [lo,hi] is the min/max of
+its column indices; the runtime there turns that into an implicit gather.
+Here the plan is computed once per (matrix structure, operand partition),
+cached on the matrix, and executed as one RCCL/xGMI alltoallv of the
+overlapping x slices.  (settings.precise_images selects an exact-index plan
+in the reference; the window plan is what its default build uses.)
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from . import comm
+from .partition import RowPartition
+
+
+class WindowGatherPlan:
+    def __init__(self, lo: int, hi: int, xpart: RowPartition, group=None):
+        self.lo = int(lo)
+        self.hi = int(hi)
+        self.xpart = xpart
+        self.group = group
+        ws = comm.world_size(group)
+        me = comm.rank(group)
+        if ws == 1:
+            self.windows = [(self.lo, self.hi)]
+        else:
+            w = torch.tensor([self.lo, self.hi], dtype=torch.int64)
+            outs = [torch.zeros(2, dtype=torch.int64) for _ in range(ws)]
+            import torch.distributed as dist
+
+            if dist.get_backend(group) == "nccl":
+                from ..runtime import runtime
+
+                w = w.to(runtime().device)
+                outs = [o.to(runtime().device) for o in outs]
+            dist.all_gather(outs, w, group=group)
+            self.windows = [(int(o[0].item()), int(o[1].item())) for o in outs]
+        # slices of MY slab that each peer needs
+        s0, s1 = xpart.start(me), xpart.stop(me)
+        self.send_ranges = []
+        for p in range(ws):
+            plo, phi = self.windows[p]
+            a, b = max(plo, s0), min(phi, s1)
+            self.send_ranges.append((a - s0, max(a, b) - s0))
+        # what I receive from each peer (ascending in p => contiguous window)
+        self.recv_counts = []
+        for p in range(ws):
+            a, b = max(self.lo, xpart.start(p)), min(self.hi, xpart.stop(p))
+            self.recv_counts.append(max(0, b - a))
+        assert sum(self.recv_counts) == self.hi - self.lo
+
+    def gather(self, xlocal: torch.Tensor) -> torch.Tensor:
+        """Return the window x[lo:hi) (dim 0 slices; works for 1-D and 2-D)."""
+        ws = comm.world_size(self.group)
+        me = comm.rank(self.group)
+        if ws == 1:
+            if self.lo == 0 and self.hi == xlocal.shape[0]:
+                return xlocal
+            return xlocal[self.lo: self.hi]
+        tail = xlocal.shape[1:]
+        k = 1
+        for t in tail:
+            k *= t
+        send = []
+        for p in range(ws):
+            a, b = self.send_ranges[p]
+            send.append(xlocal[a:b].reshape(-1))
+        recv = comm.all_to_all_v(send, group=self.group)
+        out = torch.cat([r for r in recv], dim=0)
+        return out.reshape(self.hi - self.lo, *tail) if tail else out
+
+
+class ReduceScatterPlan:
+    """Inverse of the window gather: each rank holds partial contributions to
+    y[lo:hi); owners receive and sum them.
+
+    Reference parity: the ADD reductions of col-split SpMV / CSC SpMV
+    (csr.py:869-927, csc/spmv.cu:60-75) that Legion performs implicitly.
+    """
+
+    def __init__(self, lo: int, hi: int, ypart: RowPartition, group=None):
+        self.fwd = WindowGatherPlan(lo, hi, ypart, group)
+
+    def scatter_add(self, partial: torch.Tensor, ylocal: torch.Tensor, beta: float = 1.0) -> torch.Tensor:
+        """partial covers [lo,hi); add into ylocal (owner slabs)."""
+        ws = comm.world_size(self.fwd.group)
+        me = comm.rank(self.fwd.group)
+        if beta == 0.0:
+            ylocal.zero_()
+        if ws == 1:
+            ylocal[self.fwd.lo: self.fwd.hi] += partial
+            return ylocal
+        # send partial pieces to owners: piece for peer p is the overlap of my
+        # window with p's slab — the same recv_counts layout, reversed.
+        xpart = self.fwd.xpart
+        send = []
+        off = 0
+        for p in range(ws):
+            c = self.fwd.recv_counts[p]
+            send.append(partial[off: off + c].reshape(-1))
+            off += c
+        recv = comm.all_to_all_v(send, group=self.fwd.group)
+        s0 = xpart.start(me)
+        tail = ylocal.shape[1:]
+        for p in range(ws):
+            a, b = self.fwd.send_ranges[p]  # my-slab-local target range for peer p's piece
+            if b > a:
+                ylocal[a:b] += recv[p].reshape(b - a, *tail)
+        return ylocal

@@ -1,0 +1,32 @@
+"""mmread/mmwrite (coverage parity: reference test_io.py)."""
+import numpy as np
+import scipy.io as sio
+
+import sparse.io as sio_ours
+
+from utils.common import test_mtx_files
+
+
+def test_mmread_matches_scipy():
+    for path in test_mtx_files:
+        ours = sio_ours.mmread(path)
+        ref = sio.mmread(path)
+        assert ours.shape == ref.shape, path
+        assert np.allclose(np.asarray(ours.todense()), ref.toarray()), path
+
+
+def test_mmread_tocsr():
+    path = test_mtx_files[0]
+    A = sio_ours.mmread(path).tocsr()
+    ref = sio.mmread(path).tocsr()
+    assert np.allclose(np.asarray(A.todense()), ref.toarray())
+
+
+def test_mmwrite_roundtrip(tmp_path):
+    import sparse
+
+    A = sparse.random(12, 9, 0.3, random_state=1, format="csr")
+    p = str(tmp_path / "out.mtx")
+    sio_ours.mmwrite(p, A)
+    back = sio_ours.mmread(p)
+    assert np.allclose(np.asarray(back.todense()), np.asarray(A.todense()))

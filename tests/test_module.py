@@ -1,0 +1,76 @@
+"""Module constructors (coverage parity: reference test_module.py)."""
+import numpy as np
+import pytest
+import scipy.sparse as sps
+
+import sparse
+from sparse import csr_array
+
+
+def test_eye():
+    for m, n, k in [(5, None, 0), (5, 7, 0), (6, 4, -1), (5, 5, 2)]:
+        ours = sparse.eye(m, n, k=k)
+        ref = sps.eye(m, n if n is not None else m, k=k)
+        assert np.allclose(np.asarray(ours.todense()), ref.toarray()), (m, n, k)
+
+
+def test_identity():
+    assert np.allclose(np.asarray(sparse.identity(6).todense()), np.eye(6))
+
+
+def test_diags_variants():
+    cases = [
+        ([[1, 2, 3, 4]], [0]),
+        ([[1, 2, 3], [4, 5, 6, 7], [8, 9]], [-1, 0, 2]),
+        ([np.ones(5), 2 * np.ones(4)], [0, -1]),
+    ]
+    for diags, offs in cases:
+        ours = sparse.diags(diags, offs)
+        ref = sps.diags(diags, offs)
+        assert np.allclose(np.asarray(ours.todense()), ref.toarray()), (diags, offs)
+
+
+def test_diags_scalar_broadcast():
+    ours = sparse.diags([2.0], [1], shape=(5, 5))
+    ref = sps.diags([2.0], [1], shape=(5, 5))
+    assert np.allclose(np.asarray(ours.todense()), ref.toarray())
+
+
+def test_diags_shape():
+    ours = sparse.diags([np.arange(1, 6)], [1], shape=(6, 6))
+    ref = sps.diags([np.arange(1, 6)], [1], shape=(6, 6))
+    assert np.allclose(np.asarray(ours.todense()), ref.toarray())
+
+
+def test_spdiags():
+    data = np.array([[1, 2, 3, 4.0], [5, 6, 7, 8.0]])
+    ours = sparse.spdiags(data, [0, 1], 4, 4)
+    ref = sps.spdiags(data, [0, 1], 4, 4)
+    assert np.allclose(np.asarray(ours.todense()), ref.toarray())
+
+
+def test_kron():
+    a = sps.random(4, 3, 0.5, random_state=1).tocsr()
+    b = sps.random(2, 5, 0.6, random_state=2).tocsr()
+    ours = sparse.kron(csr_array(a), csr_array(b), format="csr")
+    ref = sps.kron(a, b)
+    assert np.allclose(np.asarray(ours.todense()), ref.toarray())
+
+
+def test_random():
+    r = sparse.random(30, 20, density=0.1, random_state=3, format="csr")
+    assert r.shape == (30, 20)
+    assert r.nnz == int(round(0.1 * 600))
+    d = np.asarray(r.todense())
+    assert (d != 0).sum() == r.nnz
+
+
+def test_issparse_predicates():
+    A = sparse.eye(3)
+    assert sparse.issparse(A)
+    assert sparse.isspmatrix_csr(A)
+    assert not sparse.isspmatrix_coo(A)
+    assert sparse.isspmatrix_coo(A.tocoo())
+    assert sparse.isspmatrix_csc(A.tocsc())
+    assert sparse.isspmatrix_dia(A.tocoo().todia())
+    assert not sparse.issparse(np.eye(3))

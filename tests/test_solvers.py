@@ -1,0 +1,135 @@
+"""Solver suite (coverage parity: reference test_cg_solve.py,
+test_bicg_solve.py, test_cgs_solve.py, test_gmres_solve.py,
+test_lsqr_solve.py, test_eigsh.py)."""
+import numpy as np
+import pytest
+import scipy.sparse as sps
+
+from sparse import csr_array, linalg
+
+from utils.sample import sample_dense, spd_csr
+
+
+@pytest.mark.parametrize("dt", [np.float32, np.float64])
+def test_cg(dt):
+    n = 60
+    s = spd_csr(n, seed=1, dtype=dt)
+    b = sample_dense(n, seed=2, dtype=dt)
+    x, info = linalg.cg(csr_array(s), b, tol=1e-8, conv_test_iters=5)
+    assert info == 0
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-4 if dt == np.float32 else 1e-6)
+
+
+def test_cg_with_callback_and_x0():
+    n = 40
+    s = spd_csr(n, seed=3)
+    b = sample_dense(n, seed=4)
+    calls = []
+    x0 = np.ones(n) * 0.1
+    x, info = linalg.cg(csr_array(s), b, x0=x0, tol=1e-10,
+                        callback=lambda xk: calls.append(1), conv_test_iters=2)
+    assert info == 0
+    assert len(calls) > 0
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-6)
+
+
+def test_cg_identity_preconditioner():
+    n = 40
+    s = spd_csr(n, seed=5)
+    b = sample_dense(n, seed=6)
+    M = linalg.IdentityOperator((n, n), dtype=np.float64)
+    x, info = linalg.cg(csr_array(s), b, M=M, tol=1e-10, conv_test_iters=4)
+    assert info == 0
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-6)
+
+
+def test_cg_jacobi_preconditioner():
+    n = 50
+    s = spd_csr(n, seed=7)
+    A = csr_array(s)
+    dinv = 1.0 / s.diagonal()
+
+    def M(x, out=None):
+        import sparse
+
+        xd = sparse.asdistarray(x)
+        r = xd * sparse.asdistarray(dinv)
+        if out is not None:
+            out.local.copy_(r.local)
+            return out
+        return r
+
+    x, info = linalg.cg(A, sample_dense(n, seed=8),
+                        M=linalg.LinearOperator((n, n), matvec=M), tol=1e-10)
+    assert info == 0
+
+
+def test_cg_linear_operator():
+    n = 30
+    s = spd_csr(n, seed=9)
+    A = csr_array(s)
+    op = linalg.LinearOperator((n, n), matvec=lambda x, out=None: A.dot(x, out=out))
+    b = sample_dense(n, seed=10)
+    x, info = linalg.cg(op, b, tol=1e-10)
+    assert info == 0
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-6)
+
+
+def test_spsolve():
+    n = 30
+    s = spd_csr(n, seed=11)
+    b = sample_dense(n, seed=12)
+    x = linalg.spsolve(csr_array(s), b)
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-6)
+
+
+@pytest.mark.parametrize("solver", [linalg.cgs, linalg.bicg, linalg.bicgstab])
+def test_nonsymmetric_solvers(solver):
+    n = 40
+    rng = np.random.default_rng(13)
+    s = (sps.random(n, n, 0.3, random_state=14) + n * sps.eye(n)).tocsr()
+    b = rng.random(n)
+    x, info = solver(csr_array(s), b, tol=1e-10, maxiter=400, conv_test_iters=2)
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-5), solver.__name__
+
+
+def test_gmres():
+    n = 40
+    rng = np.random.default_rng(15)
+    s = (sps.random(n, n, 0.3, random_state=16) + n * sps.eye(n)).tocsr()
+    b = rng.random(n)
+    x, info = linalg.gmres(csr_array(s), b, tol=1e-10, restart=20, maxiter=400)
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-5)
+
+
+def test_lsqr():
+    # overdetermined least squares
+    rng = np.random.default_rng(17)
+    m, n = 50, 20
+    s = sps.random(m, n, 0.4, random_state=18).tocsr() + sps.random(
+        m, n, 0.01, random_state=19).tocsr()
+    b = rng.random(m)
+    r = linalg.lsqr(csr_array(s), b, atol=1e-12, btol=1e-12, iter_lim=200)
+    x = np.asarray(r[0])
+    xref = sps.linalg.lsqr(s, b, atol=1e-12, btol=1e-12, iter_lim=200)[0]
+    assert np.allclose(x, xref, atol=1e-5)
+
+
+@pytest.mark.parametrize("which", ["LM", "SA", "LA"])
+def test_eigsh(which):
+    n = 60
+    s = spd_csr(n, seed=20) - (n // 2) * sps.eye(n)  # mixed-sign spectrum
+    s = s.tocsr()
+    k = 4
+    w, V = linalg.eigsh(csr_array(s), k=k, which=which)
+    ws = np.linalg.eigvalsh(s.toarray())
+    if which == "LM":
+        expect = ws[np.argsort(np.abs(ws))[::-1][:k]]
+        assert np.allclose(np.sort(np.abs(w)), np.sort(np.abs(expect)), atol=1e-4)
+    elif which == "LA":
+        assert np.allclose(np.sort(w), np.sort(ws[-k:]), atol=1e-4)
+    else:
+        assert np.allclose(np.sort(w), ws[:k], atol=1e-4)
+    for i in range(k):
+        v = V[:, i]
+        assert np.linalg.norm(s @ v - w[i] * v) < 1e-3 * max(1, abs(w[i]))
