@@ -1,0 +1,96 @@
+"""Structured test matrices built directly as distributed CSR slabs.
+
+These construct each rank's rows with vectorized torch ops — no global
+arrays, no conversion sort — so a 16384^2 5-pt Poisson (268M rows, 1.34B
+nnz) materializes in seconds across 8 GPUs.  Role of the matrix-building
+prologues of the reference's examples (pde.py/gmg.py build via diags/kron).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+import torch
+
+from .csr import csr_array
+from .parallel import comm
+from .parallel.partition import RowPartition
+from .runtime import runtime
+from .types import index_dtype_for, to_torch_dtype
+
+__all__ = ["poisson2d", "poisson3d", "banded"]
+
+
+def _assemble(rows_i, cols_list, vals_list, valid_list, shape, part, dtype):
+    rt = runtime()
+    mloc = rows_i.numel()
+    idt = index_dtype_for(shape)
+    C = torch.stack(cols_list, dim=1)
+    V = torch.stack(vals_list, dim=1).to(to_torch_dtype(dtype))
+    M = torch.stack(valid_list, dim=1)
+    counts = M.sum(dim=1)
+    indptr = torch.zeros(mloc + 1, dtype=torch.int64)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    flat = M.reshape(-1)
+    indices = C.reshape(-1)[flat].to(idt)
+    values = V.reshape(-1)[flat]
+    return csr_array.from_local(indptr.to(rt.device), indices.to(rt.device),
+                                values.to(rt.device), part, shape)
+
+
+def poisson2d(nx: int, ny: Optional[int] = None, dtype=np.float64,
+              scale: float = 1.0) -> csr_array:
+    """5-point 2-D Laplacian on an nx*ny grid (Dirichlet): the pde.py /
+    BASELINE.json headline operator.  A[i,i]=4*scale, neighbors -scale."""
+    ny = nx if ny is None else ny
+    N = nx * ny
+    part = RowPartition.equal(N, comm.world_size())
+    r = comm.rank()
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    ix = rows % nx
+    one = torch.ones_like(rows, dtype=torch.float64)
+    cols = [rows - nx, rows - 1, rows, rows + 1, rows + nx]
+    vals = [-one * scale, -one * scale, 4.0 * one * scale, -one * scale, -one * scale]
+    valid = [rows - nx >= 0, ix > 0, torch.ones_like(ix, dtype=torch.bool),
+             ix < nx - 1, rows + nx < N]
+    return _assemble(rows, cols, vals, valid, (N, N), part, dtype)
+
+
+def poisson3d(nx: int, ny: Optional[int] = None, nz: Optional[int] = None,
+              dtype=np.float64, scale: float = 1.0) -> csr_array:
+    """7-point 3-D Laplacian (Dirichlet) — the GMG 3-D benchmark operator."""
+    ny = nx if ny is None else ny
+    nz = nx if nz is None else nz
+    N = nx * ny * nz
+    part = RowPartition.equal(N, comm.world_size())
+    r = comm.rank()
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    ix = rows % nx
+    iy = (rows // nx) % ny
+    one = torch.ones_like(rows, dtype=torch.float64)
+    nxy = nx * ny
+    cols = [rows - nxy, rows - nx, rows - 1, rows, rows + 1, rows + nx, rows + nxy]
+    vals = [-one * scale, -one * scale, -one * scale, 6.0 * one * scale,
+            -one * scale, -one * scale, -one * scale]
+    valid = [rows - nxy >= 0, iy > 0, ix > 0,
+             torch.ones_like(ix, dtype=torch.bool), ix < nx - 1, iy < ny - 1,
+             rows + nxy < N]
+    return _assemble(rows, cols, vals, valid, (N, N), part, dtype)
+
+
+def banded(n: int, ndiags: int = 11, dtype=np.float64) -> csr_array:
+    """Banded matrix with ndiags diagonals at offsets centered on 0 — the
+    dot_microbenchmark operator (reference examples/dot_microbenchmark.py:
+    11-diagonal CSR, fp64)."""
+    half = ndiags // 2
+    part = RowPartition.equal(n, comm.world_size())
+    r = comm.rank()
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    one = torch.ones_like(rows, dtype=torch.float64)
+    cols, vals, valid = [], [], []
+    for k in range(-half, ndiags - half):
+        c = rows + k
+        cols.append(c)
+        vals.append(one * (1.0 if k else float(ndiags)))
+        valid.append((c >= 0) & (c < n))
+    return _assemble(rows, cols, vals, valid, (n, n), part, dtype)

@@ -125,6 +125,9 @@ def spgemm_csr(A, B, a_col_lo, vdt):
     nnz_per_row = torch.empty(A.nrows, dtype=torch.int64, device=A.device)
     ext().spgemm_nnz(A.indptr, A.indices, B.indptr, B.indices, nnz_per_row,
                      int(a_col_lo), int(B.ncols))
+    if bool((nnz_per_row < 0).any().item()):
+        # rows denser than the LDS hash: vectorized expand-sort-reduce
+        return _spgemm_esc(A, B, a_col_lo, vdt)
     indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=A.device)
     torch.cumsum(nnz_per_row, 0, out=indptr[1:])
     nnz = int(indptr[-1].item())
@@ -133,7 +136,50 @@ def spgemm_csr(A, B, a_col_lo, vdt):
     ext().spgemm_compute(A.indptr, A.indices, A.values.to(vdt),
                          B.indptr, B.indices, B.values.to(vdt),
                          indptr, indices, values, int(a_col_lo), int(B.ncols))
-    return LocalCSR(indptr, indices, values, A.nrows, B.ncols)
+    # sort columns within each row (hash compaction is unordered)
+    rows = torch.repeat_interleave(
+        torch.arange(A.nrows, dtype=torch.int64, device=A.device), nnz_per_row)
+    key = rows * B.ncols + indices.long()
+    order = torch.argsort(key)
+    return LocalCSR(indptr, indices[order], values[order], A.nrows, B.ncols)
+
+
+def _spgemm_esc(A, B, a_col_lo, vdt):
+    """Expansion-sort-compress SpGEMM fallback in torch ops (still on-GPU)."""
+    from ..ops.local import LocalCSR
+
+    dev = A.device
+    acols = A.indices.long() - a_col_lo
+    acounts = A.indptr[1:] - A.indptr[:-1]
+    arows = torch.repeat_interleave(
+        torch.arange(A.nrows, dtype=torch.int64, device=dev), acounts)
+    bstart = B.indptr[acols]
+    bcounts = B.indptr[acols + 1] - bstart
+    total = int(bcounts.sum().item())
+    if total == 0:
+        return LocalCSR(torch.zeros(A.nrows + 1, dtype=torch.int64, device=dev),
+                        torch.zeros(0, dtype=A.indices.dtype, device=dev),
+                        torch.zeros(0, dtype=vdt, device=dev), A.nrows, B.ncols)
+    offs = torch.zeros(acols.numel(), dtype=torch.int64, device=dev)
+    torch.cumsum(bcounts[:-1], 0, out=offs[1:])
+    pos = (torch.arange(total, dtype=torch.int64, device=dev)
+           - torch.repeat_interleave(offs, bcounts)
+           + torch.repeat_interleave(bstart, bcounts))
+    erows = torch.repeat_interleave(arows, bcounts)
+    ecols = B.indices[pos].long()
+    evals = torch.repeat_interleave(A.values.to(vdt), bcounts) * B.values[pos].to(vdt)
+    key = erows * B.ncols + ecols
+    key, order = torch.sort(key)
+    evals = evals[order]
+    ukey, inv = torch.unique_consecutive(key, return_inverse=True)
+    out_vals = torch.zeros(ukey.numel(), dtype=vdt, device=dev)
+    out_vals.index_add_(0, inv, evals)
+    rows = torch.div(ukey, B.ncols, rounding_mode="floor")
+    cols = (ukey - rows * B.ncols).to(A.indices.dtype)
+    counts = torch.bincount(rows, minlength=A.nrows)
+    indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    return LocalCSR(indptr, cols, out_vals, A.nrows, B.ncols)
 
 
 def axpby(y, x, a, b, isalpha: bool, negate: bool):

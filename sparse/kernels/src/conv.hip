@@ -1,0 +1,139 @@
+// Conversions + CSC col-split kernels.
+//
+// Reference parity: CSR_TO_DENSE (csr_to_dense.cu), CSR_DIAGONAL
+// (get_diagonal.cu), CSC_SPMV_COL_SPLIT (csc/spmv.cu:60-75), SPMM_CSC_DENSE
+// (csc/spmm.cu).
+#include "common.h"
+
+namespace {
+
+template <typename T, typename index_t>
+__global__ void csr_to_dense_kernel(const int64_t* __restrict__ indptr,
+                                    const index_t* __restrict__ indices,
+                                    const T* __restrict__ vals,
+                                    T* __restrict__ out, int64_t m, int64_t n,
+                                    int64_t nnz) {
+  int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (p >= nnz) return;
+  int64_t r = ub_i64(indptr, m + 1, p) - 1;
+  out[r * n + (int64_t)indices[p]] = vals[p];
+}
+
+template <typename T, typename index_t>
+__global__ void csr_diagonal_kernel(const int64_t* __restrict__ indptr,
+                                    const index_t* __restrict__ indices,
+                                    const T* __restrict__ vals,
+                                    T* __restrict__ out, int64_t m,
+                                    int64_t row_offset) {
+  int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (r >= m) return;
+  int64_t target = r + row_offset;
+  int64_t lo = indptr[r], hi = indptr[r + 1];
+  // binary search in sorted row for column == target
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    int64_t c = (int64_t)indices[mid];
+    if (c < target) lo = mid + 1;
+    else hi = mid;
+  }
+  if (lo < indptr[r + 1] && (int64_t)indices[lo] == target) out[r] = vals[lo];
+}
+
+// y[row - rlo] += v * x[col]; one thread per nz, col via binary search
+template <typename T, typename index_t>
+__global__ void csc_spmv_kernel(const int64_t* __restrict__ colptr,
+                                const index_t* __restrict__ rowidx,
+                                const T* __restrict__ vals,
+                                const T* __restrict__ x,
+                                T* __restrict__ y, int64_t ncl, int64_t rlo,
+                                int64_t nnz) {
+  int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (p >= nnz) return;
+  int64_t c = ub_i64(colptr, ncl + 1, p) - 1;
+  atomic_add_any(&y[(int64_t)rowidx[p] - rlo], vals[p] * x[c]);
+}
+
+// C[(row-rlo), j] += v * B[col, j]; one wave per nz, lanes over j
+template <typename T, typename index_t>
+__global__ void csc_spmm_kernel(const int64_t* __restrict__ colptr,
+                                const index_t* __restrict__ rowidx,
+                                const T* __restrict__ vals,
+                                const T* __restrict__ B, T* __restrict__ C,
+                                int64_t ncl, int64_t rlo, int64_t k,
+                                int64_t nnz) {
+  int64_t w = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int lane = threadIdx.x % WAVE;
+  if (w >= nnz) return;
+  int64_t c = ub_i64(colptr, ncl + 1, w) - 1;
+  int64_t r = (int64_t)rowidx[w] - rlo;
+  T v = vals[w];
+  for (int64_t j = lane; j < k; j += WAVE) {
+    atomic_add_any(&C[r * k + j], v * B[c * k + j]);
+  }
+}
+
+}  // namespace
+
+void csr_to_dense_hip(at::Tensor indptr, at::Tensor indices, at::Tensor vals,
+                      at::Tensor out) {
+  int64_t nnz = vals.numel();
+  if (nnz == 0) return;
+  DISPATCH_VALUES(vals.scalar_type(), "csr_to_dense", [&] {
+    using T = scalar_t;
+    DISPATCH_INDEX(indices.scalar_type(), "csr_to_dense_idx", [&] {
+      hipLaunchKernelGGL((csr_to_dense_kernel<T, index_t>),
+                         dim3((nnz + 255) / 256), dim3(256), 0, cur_stream(),
+                         indptr.data_ptr<int64_t>(), indices.data_ptr<index_t>(),
+                         vals.data_ptr<T>(), out.data_ptr<T>(),
+                         indptr.numel() - 1, out.size(1), nnz);
+    });
+  });
+}
+
+void csr_diagonal_hip(at::Tensor indptr, at::Tensor indices, at::Tensor vals,
+                      at::Tensor out, int64_t row_offset) {
+  int64_t m = out.numel();
+  if (m == 0) return;
+  DISPATCH_VALUES(vals.scalar_type(), "csr_diagonal", [&] {
+    using T = scalar_t;
+    DISPATCH_INDEX(indices.scalar_type(), "csr_diagonal_idx", [&] {
+      hipLaunchKernelGGL((csr_diagonal_kernel<T, index_t>),
+                         dim3((m + 255) / 256), dim3(256), 0, cur_stream(),
+                         indptr.data_ptr<int64_t>(), indices.data_ptr<index_t>(),
+                         vals.data_ptr<T>(), out.data_ptr<T>(), m, row_offset);
+    });
+  });
+}
+
+void csc_spmv_hip(at::Tensor colptr, at::Tensor rowidx, at::Tensor vals,
+                  at::Tensor x, at::Tensor y, int64_t rlo) {
+  int64_t nnz = vals.numel();
+  if (nnz == 0) return;
+  DISPATCH_VALUES(vals.scalar_type(), "csc_spmv", [&] {
+    using T = scalar_t;
+    DISPATCH_INDEX(rowidx.scalar_type(), "csc_spmv_idx", [&] {
+      hipLaunchKernelGGL((csc_spmv_kernel<T, index_t>), dim3((nnz + 255) / 256),
+                         dim3(256), 0, cur_stream(), colptr.data_ptr<int64_t>(),
+                         rowidx.data_ptr<index_t>(), vals.data_ptr<T>(),
+                         x.data_ptr<T>(), y.data_ptr<T>(), colptr.numel() - 1,
+                         rlo, nnz);
+    });
+  });
+}
+
+void csc_spmm_hip(at::Tensor colptr, at::Tensor rowidx, at::Tensor vals,
+                  at::Tensor B, at::Tensor C, int64_t rlo) {
+  int64_t nnz = vals.numel();
+  if (nnz == 0) return;
+  DISPATCH_VALUES(vals.scalar_type(), "csc_spmm", [&] {
+    using T = scalar_t;
+    DISPATCH_INDEX(rowidx.scalar_type(), "csc_spmm_idx", [&] {
+      int64_t threads = nnz * WAVE;
+      hipLaunchKernelGGL((csc_spmm_kernel<T, index_t>),
+                         dim3((threads + 255) / 256), dim3(256), 0, cur_stream(),
+                         colptr.data_ptr<int64_t>(), rowidx.data_ptr<index_t>(),
+                         vals.data_ptr<T>(), B.data_ptr<T>(), C.data_ptr<T>(),
+                         colptr.numel() - 1, rlo, B.size(1), nnz);
+    });
+  });
+}

@@ -1,0 +1,103 @@
+// torch.ops bindings for the sparse HIP (gfx950) kernels.
+//
+// Loaded via torch.ops.load_library from sparse/kernels/_build/sparse_hip.so;
+// Python wrappers live in sparse/kernels/__init__.py.
+#include <torch/library.h>
+
+#include <ATen/ATen.h>
+
+// launchers defined in the .hip translation units
+void spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+              int64_t, double);
+void add_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void add_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                     at::Tensor, at::Tensor, at::Tensor, at::Tensor, double,
+                     double);
+void mult_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void mult_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      at::Tensor);
+void mult_dense_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void axpby_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool, bool);
+void spmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+              int64_t);
+void rspmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void sddmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+               at::Tensor);
+void csr_to_dense_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void csr_diagonal_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t);
+void csc_spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                  int64_t);
+void csc_spmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                  int64_t);
+void tropical_spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t);
+void rk_calc_dy_hip(at::Tensor, at::Tensor, double, at::Tensor);
+void cdist_hip(at::Tensor, at::Tensor, at::Tensor);
+void spgemm_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                    int64_t, int64_t);
+void spgemm_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                        at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                        at::Tensor, int64_t, int64_t);
+
+TORCH_LIBRARY(sparse_hip, m) {
+  m.def("spmv(Tensor indptr, Tensor indices, Tensor values, Tensor x, "
+        "Tensor(a!) y, int col_lo, float beta) -> ()");
+  m.def("add_nnz(Tensor aip, Tensor aix, Tensor bip, Tensor bix, "
+        "Tensor(a!) out) -> ()");
+  m.def("add_compute(Tensor aip, Tensor aix, Tensor av, Tensor bip, "
+        "Tensor bix, Tensor bv, Tensor cip, Tensor(a!) cix, Tensor(b!) cv, "
+        "float alpha, float beta) -> ()");
+  m.def("mult_nnz(Tensor aip, Tensor aix, Tensor bip, Tensor bix, "
+        "Tensor(a!) out) -> ()");
+  m.def("mult_compute(Tensor aip, Tensor aix, Tensor av, Tensor bip, "
+        "Tensor bix, Tensor bv, Tensor cip, Tensor(a!) cix, Tensor(b!) cv) -> ()");
+  m.def("mult_dense(Tensor indptr, Tensor indices, Tensor vals, Tensor D, "
+        "Tensor(a!) out) -> ()");
+  m.def("axpby(Tensor(a!) y, Tensor x, Tensor a, Tensor b, bool isalpha, "
+        "bool negate) -> ()");
+  m.def("spmm(Tensor indptr, Tensor indices, Tensor vals, Tensor B, "
+        "Tensor(a!) C, int col_lo) -> ()");
+  m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "
+        "Tensor(a!) C) -> ()");
+  m.def("sddmm(Tensor indptr, Tensor indices, Tensor vals, Tensor C, "
+        "Tensor D, Tensor(a!) out) -> ()");
+  m.def("csr_to_dense(Tensor indptr, Tensor indices, Tensor vals, "
+        "Tensor(a!) out) -> ()");
+  m.def("csr_diagonal(Tensor indptr, Tensor indices, Tensor vals, "
+        "Tensor(a!) out, int row_offset) -> ()");
+  m.def("csc_spmv(Tensor colptr, Tensor rowidx, Tensor vals, Tensor x, "
+        "Tensor(a!) y, int rlo) -> ()");
+  m.def("csc_spmm(Tensor colptr, Tensor rowidx, Tensor vals, Tensor B, "
+        "Tensor(a!) C, int rlo) -> ()");
+  m.def("tropical_spmv(Tensor indptr, Tensor indices, Tensor x, "
+        "Tensor(a!) y, int col_lo) -> ()");
+  m.def("rk_calc_dy(Tensor K, Tensor avec, float h, Tensor(a!) dy) -> ()");
+  m.def("cdist(Tensor XA, Tensor XB, Tensor(a!) out) -> ()");
+  m.def("spgemm_nnz(Tensor aip, Tensor aix, Tensor bip, Tensor bix, "
+        "Tensor(a!) nnz_out, int a_col_lo, int bncols) -> ()");
+  m.def("spgemm_compute(Tensor aip, Tensor aix, Tensor av, Tensor bip, "
+        "Tensor bix, Tensor bv, Tensor cip, Tensor(a!) cix, Tensor(b!) cv, "
+        "int a_col_lo, int bncols) -> ()");
+}
+
+TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
+  m.impl("spmv", spmv_hip);
+  m.impl("add_nnz", add_nnz_hip);
+  m.impl("add_compute", add_compute_hip);
+  m.impl("mult_nnz", mult_nnz_hip);
+  m.impl("mult_compute", mult_compute_hip);
+  m.impl("mult_dense", mult_dense_hip);
+  m.impl("axpby", axpby_hip);
+  m.impl("spmm", spmm_hip);
+  m.impl("rspmm", rspmm_hip);
+  m.impl("sddmm", sddmm_hip);
+  m.impl("csr_to_dense", csr_to_dense_hip);
+  m.impl("csr_diagonal", csr_diagonal_hip);
+  m.impl("csc_spmv", csc_spmv_hip);
+  m.impl("csc_spmm", csc_spmm_hip);
+  m.impl("tropical_spmv", tropical_spmv_hip);
+  m.impl("rk_calc_dy", rk_calc_dy_hip);
+  m.impl("cdist", cdist_hip);
+  m.impl("spgemm_nnz", spgemm_nnz_hip);
+  m.impl("spgemm_compute", spgemm_compute_hip);
+}

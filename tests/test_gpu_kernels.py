@@ -1,0 +1,251 @@
+"""HIP kernel numerics vs CPU fp64/scipy oracle — all @gpu.
+
+Each test builds the operands on CPU (scipy/torch reference), moves them to
+cuda:0 through the sparse API, and compares.  The GPU path requires the
+in-tree HIP extension (sparse.kernels.require()); these tests fail — not
+skip — if it is missing on a GPU box.
+"""
+import numpy as np
+import pytest
+import scipy.sparse as sps
+import torch
+
+from utils.common import types
+from utils.sample import sample_csr, sample_dense, spd_csr
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(autouse=True, scope="module")
+def _require_ext():
+    if torch.cuda.is_available():
+        from sparse.kernels import require
+
+        require()
+
+
+def tol(dt):
+    if np.dtype(dt) in (np.float32, np.complex64):
+        return dict(rtol=2e-4, atol=2e-5)
+    return dict(rtol=1e-10, atol=1e-12)
+
+
+@pytest.mark.parametrize("dt", types)
+@pytest.mark.parametrize("shape", [(64, 80), (500, 300), (1000, 1000)])
+def test_spmv_gpu(dt, shape):
+    from sparse import csr_array
+
+    s = sample_csr(*shape, 0.05, seed=1, dtype=dt)
+    x = sample_dense(shape[1], seed=2, dtype=dt)
+    A = csr_array(s)
+    assert A._values.is_cuda
+    y = A @ x
+    assert np.allclose(np.asarray(y), s @ x, **tol(dt))
+
+
+def test_spmv_long_row_carry_gpu():
+    # one row with 100k nnz spans many 2048-nnz blocks: exercises the carry path
+    from sparse import csr_array
+
+    rng = np.random.default_rng(3)
+    n = 200_000
+    cols = np.unique(rng.integers(0, n, 100_000))
+    rows = np.zeros(len(cols), dtype=np.int64)
+    extra_r = np.arange(1, 50)
+    extra_c = rng.integers(0, n, 49)
+    data = np.concatenate([rng.random(len(cols)), rng.random(49)])
+    i = np.concatenate([rows, extra_r])
+    j = np.concatenate([cols, extra_c])
+    s = sps.csr_matrix((data, (i, j)), shape=(50, n))
+    x = rng.random(n)
+    A = csr_array(s)
+    assert np.allclose(np.asarray(A @ x), s @ x, rtol=1e-10, atol=1e-9)
+
+
+@pytest.mark.parametrize("dt", types)
+def test_spmm_gpu(dt):
+    from sparse import csr_array
+
+    s = sample_csr(300, 200, 0.05, seed=4, dtype=dt)
+    B = sample_dense((200, 33), seed=5, dtype=dt)
+    assert np.allclose(np.asarray(csr_array(s) @ B), s @ B, **tol(dt))
+
+
+@pytest.mark.parametrize("dt", [np.float64, np.complex128])
+def test_rspmm_gpu(dt):
+    from sparse import csr_array
+
+    s = sample_csr(150, 120, 0.08, seed=6, dtype=dt)
+    A = sample_dense((7, 150), seed=7, dtype=dt)
+    assert np.allclose(np.asarray(A @ csr_array(s)), A @ s, **tol(dt))
+
+
+@pytest.mark.parametrize("dt", types)
+def test_add_gpu(dt):
+    from sparse import csr_array
+
+    a = sample_csr(400, 300, 0.03, seed=8, dtype=dt)
+    b = sample_csr(400, 300, 0.03, seed=9, dtype=dt)
+    c = csr_array(a) + csr_array(b)
+    assert np.allclose(np.asarray(c.todense()), (a + b).toarray(), **tol(dt))
+
+
+@pytest.mark.parametrize("dt", types)
+def test_elem_mult_gpu(dt):
+    from sparse import csr_array
+
+    a = sample_csr(300, 300, 0.05, seed=10, dtype=dt)
+    b = sample_csr(300, 300, 0.05, seed=11, dtype=dt)
+    c = csr_array(a).multiply(csr_array(b))
+    assert np.allclose(np.asarray(c.todense()), a.multiply(b).toarray(), **tol(dt))
+
+
+@pytest.mark.parametrize("dt", [np.float64, np.complex128])
+def test_mult_dense_gpu(dt):
+    from sparse import csr_array
+
+    a = sample_csr(120, 90, 0.1, seed=12, dtype=dt)
+    d = sample_dense((120, 90), seed=13, dtype=dt)
+    c = csr_array(a).multiply(d)
+    assert np.allclose(np.asarray(c.todense()), a.multiply(d).toarray(), **tol(dt))
+
+
+@pytest.mark.parametrize("dt", types)
+def test_spgemm_gpu(dt):
+    from sparse import csr_array
+
+    a = sample_csr(200, 150, 0.05, seed=14, dtype=dt)
+    b = sample_csr(150, 180, 0.05, seed=15, dtype=dt)
+    c = csr_array(a) @ csr_array(b)
+    assert np.allclose(np.asarray(c.todense()), (a @ b).toarray(), **tol(dt))
+
+
+def test_spgemm_dense_row_fallback_gpu():
+    # a row whose product has > 1024 distinct columns must take the ESC path
+    from sparse import csr_array
+
+    rng = np.random.default_rng(16)
+    n = 3000
+    a = sps.random(50, 40, 0.2, random_state=17).tocsr()
+    a[0, :] = 1.0  # dense row -> C row 0 hits most of B's columns
+    a = a.tocsr()
+    b = sps.random(40, n, 0.2, random_state=18).tocsr()
+    c = csr_array(a) @ csr_array(b)
+    assert c.nnz == (a @ b).nnz
+    assert np.allclose(np.asarray(c.todense()), (a @ b).toarray(), rtol=1e-10, atol=1e-10)
+
+
+@pytest.mark.parametrize("dt", [np.float64])
+def test_sddmm_gpu(dt):
+    from sparse import csr_array
+
+    s = sample_csr(90, 110, 0.1, seed=19, dtype=dt)
+    C = sample_dense((90, 8), seed=20, dtype=dt)
+    D = sample_dense((8, 110), seed=21, dtype=dt)
+    out = csr_array(s).sddmm(C, D)
+    assert np.allclose(np.asarray(out.todense()), s.multiply(C @ D).toarray(), **tol(dt))
+
+
+@pytest.mark.parametrize("dt", types)
+def test_csc_spmv_gpu(dt):
+    from sparse import csc_array
+
+    s = sample_csr(130, 170, 0.06, seed=22, dtype=dt)
+    x = sample_dense(170, seed=23, dtype=dt)
+    A = csc_array(s.tocsc())
+    assert np.allclose(np.asarray(A @ x), s @ x, **tol(dt))
+
+
+def test_conversions_gpu():
+    from sparse import csr_array
+
+    s = sample_csr(200, 160, 0.05, seed=24)
+    A = csr_array(s)
+    assert np.allclose(np.asarray(A.todense()), s.toarray())
+    assert np.allclose(np.asarray(A.tocsc().todense()), s.toarray())
+    assert np.allclose(np.asarray(A.tocoo().tocsr().todense()), s.toarray())
+    assert np.allclose(np.asarray(A.diagonal()), s.diagonal())
+
+
+def test_axpby_gpu():
+    from sparse import darray
+    from sparse.linalg import cg_axpby
+
+    n = 10000
+    y = darray.random((n,), seed=25)
+    x = darray.random((n,), seed=26)
+    a = torch.tensor(3.0, device="cuda", dtype=torch.float64)
+    b = torch.tensor(2.0, device="cuda", dtype=torch.float64)
+    y0 = np.asarray(y).copy()
+    x0 = np.asarray(x)
+    cg_axpby(y, x, a, b, isalpha=True, negate=False)
+    assert np.allclose(np.asarray(y), y0 + 1.5 * x0)
+    cg_axpby(y, x, a, b, isalpha=True, negate=True)
+    assert np.allclose(np.asarray(y), y0)
+    cg_axpby(y, x, a, b, isalpha=False, negate=False)
+    assert np.allclose(np.asarray(y), x0 + 1.5 * y0)
+
+
+def test_cg_gpu():
+    from sparse import csr_array, linalg
+
+    n = 300
+    s = spd_csr(n, seed=27)
+    rng = np.random.default_rng(28)
+    b = rng.random(n)
+    x, info = linalg.cg(csr_array(s), b, tol=1e-10, conv_test_iters=10)
+    assert info == 0
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-6)
+
+
+def test_cg_poisson_gpu():
+    from sparse import darray, gallery, linalg
+
+    A = gallery.poisson2d(128)
+    n = A.shape[0]
+    b = darray.ones((n,), dtype=np.float64)
+    x, info = linalg.cg(A, b, tol=1e-8, maxiter=2000, conv_test_iters=25)
+    r = b - A.dot(x)
+    assert float(r.norm().item()) < 1e-6 * float(b.norm().item())
+
+
+def test_cdist_gpu():
+    from sparse import spatial
+
+    XA = sample_dense((100, 7), seed=29)
+    XB = sample_dense((80, 7), seed=30)
+    from scipy.spatial.distance import cdist as sp_cdist
+
+    assert np.allclose(np.asarray(spatial.cdist(XA, XB)), sp_cdist(XA, XB),
+                       atol=1e-8)
+
+
+def test_tropical_spmv_gpu():
+    from sparse import csr_array
+    from sparse.darray import DistArray
+
+    rng = np.random.default_rng(31)
+    s = sample_csr(60, 60, 0.2, seed=32)
+    x = rng.integers(0, 100, size=(60, 3)).astype(np.int64)
+    A = csr_array(s)
+    y = A.tropical_spmv(x)
+    yn = np.asarray(y)
+    sc = s.tocoo()
+    expect = np.zeros((60, 3), dtype=np.int64)
+    for r in range(60):
+        cols = sc.col[sc.row == r]
+        if len(cols):
+            cand = sorted((tuple(x[c]) for c in cols), reverse=True)[0]
+            expect[r] = cand
+    assert np.array_equal(yn, expect)
+
+
+def test_native_extension_is_loaded():
+    """The driver checks loaded .so files; make sure ours is in-process."""
+    from sparse import kernels
+
+    assert kernels.ext() is not None
+    import subprocess
+
+    maps = open("/proc/self/maps").read()
+    assert "sparse_hip.so" in maps
