@@ -84,7 +84,8 @@ def csc_spmm(colptr, rowidx, values, B, C, rlo: int):
 
 
 def tropical_spmv(A, x, y, col_lo: int):
-    ext().tropical_spmv(A.indptr, A.indices, A.values, x, y, int(col_lo))
+    # semiring uses structure only (reference tropical_spmv.cu:26-56)
+    ext().tropical_spmv(A.indptr, A.indices, x, y, int(col_lo))
 
 
 def add_csr(A, B, alpha, beta, vdt):
