@@ -78,21 +78,31 @@ def main() -> None:
     Aop = linalg.aslinearoperator(A)
     xv = darray.zeros((n,), dtype=dtype)
     r = b - Aop.matvec(xv)
-    z = r.copy()
-    pvec = z.copy()
-    rz = r.dot(z)
+    pvec = r.copy()
+    rz = r.dot(r)
     q = darray.zeros((n,), dtype=dtype)
 
-    def cg_step() -> None:
-        nonlocal rz
-        Aop.matvec(pvec, out=q)
-        pq = pvec.dot(q)
-        linalg.cg_axpby(xv, pvec, rz, pq, isalpha=True, negate=False)
-        linalg.cg_axpby(r, q, rz, pq, isalpha=True, negate=True)
-        z.local.copy_(r.local)
-        rz_new = r.dot(z)
-        linalg.cg_axpby(pvec, z, rz_new, rz, isalpha=False, negate=False)
-        rz = rz_new
+    if on_gpu:
+
+        def cg_step() -> None:  # fully-fused MI355X path (see linalg.cg)
+            nonlocal rz
+            pq = A.spmv_dot(pvec, q)
+            linalg.cg_axpby(xv, pvec, rz, pq, isalpha=True, negate=False)
+            rz_new = linalg._axpby_norm2(r, q, rz, pq, negate=True)
+            linalg.cg_axpby(pvec, r, rz_new, rz, isalpha=False, negate=False)
+            rz = rz_new
+
+    else:
+
+        def cg_step() -> None:
+            nonlocal rz
+            Aop.matvec(pvec, out=q)
+            pq = pvec.dot(q)
+            linalg.cg_axpby(xv, pvec, rz, pq, isalpha=True, negate=False)
+            linalg.cg_axpby(r, q, rz, pq, isalpha=True, negate=True)
+            rz_new = r.dot(r)
+            linalg.cg_axpby(pvec, r, rz_new, rz, isalpha=False, negate=False)
+            rz = rz_new
 
     for _ in range(args.warmup):
         cg_step()

@@ -379,6 +379,19 @@ class csr_array(CompressedBase, DenseSparseBase):
             return out
         return DistArray.from_local(Clocal, self.partition, (self.shape[0], B.shape[1]))
 
+    def spmv_dot(self, p: DistArray, q: DistArray) -> torch.Tensor:
+        """Fused q = A@p and all-reduced sum(p*q) — the CG p·Ap in one kernel
+        (GPU, real dtypes; MI355X fusion: saves re-reading p and q)."""
+        from . import kernels
+
+        plan = self._xplan(p.partition)
+        xw = plan.gather(p.local)
+        dot = torch.zeros((), dtype=self._values.dtype, device=self._values.device)
+        lc = self.local
+        kernels.spmv_dot(lc, xw, q.local, p.local, dot, plan.lo)
+        comm.all_reduce_(dot)
+        return dot
+
     def _rspmm(self, A: DistArray) -> DistArray:
         # C = A(dense k x m) @ self(m x n): replicate A, local partial with my
         # row slab of B, ADD all-reduce (reference csr.py:1209-1240 semantics).

@@ -214,11 +214,16 @@ class DistArray:
     # -- reductions (return 0-dim device tensors: non-blocking scalars) -------
     def dot(self, other: "DistArray") -> torch.Tensor:
         o = other.local if isinstance(other, DistArray) else other
-        if self.local.is_complex() or (isinstance(o, torch.Tensor) and o.is_complex()):
+        if self.local.numel() == 0:
+            out = torch.zeros((), dtype=self.local.dtype, device=self.local.device)
+        elif self.local.dim() == 1 and isinstance(o, torch.Tensor) and o.dim() == 1 \
+                and self.local.dtype == o.dtype:
+            # fused single-pass kernel (vdot conjugates the left operand)
+            out = torch.vdot(self.local, o) if self.local.is_complex() else torch.dot(self.local, o)
+        elif self.local.is_complex() or (isinstance(o, torch.Tensor) and o.is_complex()):
             out = torch.sum(torch.conj(self.local) * o)
         else:
-            out = torch.sum(self.local * o) if self.local.numel() else torch.zeros(
-                (), dtype=self.local.dtype, device=self.local.device)
+            out = torch.sum(self.local * o)
         comm.all_reduce_(out)
         return out
 

@@ -47,8 +47,14 @@ def spmv(A, x, y, col_lo: int, beta: float):
     ext().spmv(A.indptr, A.indices, A.values, x, y, int(col_lo), float(beta))
 
 
-def spmv_dot(A, x, y, p):
-    return ext().spmv_dot(A.indptr, A.indices, A.values, x, y, p, 0)
+def spmv_dot(A, x, y, p, dot_out, col_lo: int):
+    """Fused y = A@x(window) ; dot_out += sum(p_local * y_local)."""
+    ext().spmv_dot(A.indptr, A.indices, A.values, x, y, p, dot_out, int(col_lo))
+
+
+def axpby_norm2(y, x, a, b, isalpha, negate, dot_out):
+    """Fused axpby + dot_out += sum(y_new^2) (real dtypes)."""
+    ext().axpby_norm2(y, x, a, b, bool(isalpha), bool(negate), dot_out)
 
 
 def spmm(A, B, C, col_lo: int):

@@ -76,6 +76,16 @@ struct ZeroOf<c10::complex<T>> {
 #define DISPATCH_VALUES(VTYPE, NAME, ...) \
   AT_DISPATCH_FLOATING_AND_COMPLEX_TYPES(VTYPE, NAME, __VA_ARGS__)
 
+// Bijective XCD-aware block remap (guide T1): the dispatcher places block b
+// on XCD b%8; remapping gives each XCD a CONTIGUOUS chunk of the problem so
+// neighbouring tiles share that XCD's private L2 (x-gather locality).
+__device__ __forceinline__ int64_t xcd_swizzle(int64_t bid, int64_t nwg) {
+  constexpr int64_t NXCD = 8;
+  int64_t q = nwg / NXCD, rr = nwg % NXCD;
+  int64_t xcd = bid % NXCD, idx = bid / NXCD;
+  return (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+}
+
 inline hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }

@@ -225,3 +225,51 @@ void axpby_hip(at::Tensor y, at::Tensor x, at::Tensor a, at::Tensor b,
     else launch(axpby_kernel<T, false, true>);
   });
 }
+
+namespace {
+
+// fused y-update + sum(y_new^2): the CG r-update + rz reduction in one pass
+template <typename T, bool ISALPHA, bool NEG>
+__global__ __launch_bounds__(256) void axpby_norm2_kernel(
+    T* __restrict__ y, const T* __restrict__ x, const T* __restrict__ a,
+    const T* __restrict__ b, T* __restrict__ dot_out, int64_t n) {
+  __shared__ T red[256];
+  T s = (*a) / (*b);
+  if (NEG) s = -s;
+  T acc = T(0);
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    T v = ISALPHA ? (y[i] + s * x[i]) : (x[i] + s * y[i]);
+    y[i] = v;
+    acc += v * v;
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(dot_out, red[0]);
+}
+
+}  // namespace
+
+void axpby_norm2_hip(at::Tensor y, at::Tensor x, at::Tensor a, at::Tensor b,
+                     bool isalpha, bool negate, at::Tensor dot_out) {
+  int64_t n = y.numel();
+  if (n == 0) return;
+  int64_t blocks = std::min<int64_t>((n + 255) / 256, 1024);
+  AT_DISPATCH_FLOATING_TYPES(y.scalar_type(), "axpby_norm2", [&] {
+    using T = scalar_t;
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, cur_stream(),
+                         y.data_ptr<T>(), x.data_ptr<T>(), a.data_ptr<T>(),
+                         b.data_ptr<T>(), dot_out.data_ptr<T>(), n);
+    };
+    if (isalpha && !negate) launch(axpby_norm2_kernel<T, true, false>);
+    else if (isalpha && negate) launch(axpby_norm2_kernel<T, true, true>);
+    else if (!isalpha && !negate) launch(axpby_norm2_kernel<T, false, false>);
+    else launch(axpby_norm2_kernel<T, false, true>);
+  });
+}

@@ -9,6 +9,10 @@
 // launchers defined in the .hip translation units
 void spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
               int64_t, double);
+void spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                  at::Tensor, at::Tensor, int64_t);
+void axpby_norm2_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool,
+                     bool, at::Tensor);
 void add_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void add_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                      at::Tensor, at::Tensor, at::Tensor, at::Tensor, double,
@@ -55,6 +59,10 @@ TORCH_LIBRARY(sparse_hip, m) {
         "Tensor(a!) out) -> ()");
   m.def("axpby(Tensor(a!) y, Tensor x, Tensor a, Tensor b, bool isalpha, "
         "bool negate) -> ()");
+  m.def("spmv_dot(Tensor indptr, Tensor indices, Tensor values, Tensor x, "
+        "Tensor(a!) y, Tensor pvec, Tensor(b!) dot_out, int col_lo) -> ()");
+  m.def("axpby_norm2(Tensor(a!) y, Tensor x, Tensor a, Tensor b, bool isalpha, "
+        "bool negate, Tensor(b!) dot_out) -> ()");
   m.def("spmm(Tensor indptr, Tensor indices, Tensor vals, Tensor B, "
         "Tensor(a!) C, int col_lo) -> ()");
   m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "
@@ -88,6 +96,8 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("mult_compute", mult_compute_hip);
   m.impl("mult_dense", mult_dense_hip);
   m.impl("axpby", axpby_hip);
+  m.impl("spmv_dot", spmv_dot_hip);
+  m.impl("axpby_norm2", axpby_norm2_hip);
   m.impl("spmm", spmm_hip);
   m.impl("rspmm", rspmm_hip);
   m.impl("sddmm", sddmm_hip);

@@ -5,11 +5,16 @@
 //  - each 256-thread block owns NNZ_PER_BLOCK consecutive nonzeros, so the
 //    values and indices streams are read fully coalesced exactly once
 //    (the kernel is HBM-bound: ~12 B/nnz for fp64+int32);
-//  - products go through LDS (16 KB fp64 per block), then each thread sums
-//    its rows' segments — no per-nnz atomics;
-//  - rows cut by a block boundary produce one carry per block, combined by a
-//    tiny fixup kernel (atomic-free main path);
-//  - grid = nnz/2048 blocks >> 256 CUs, so the chip fills at any row count.
+//  - the staging loop is register-pipelined: all index loads issue first,
+//    then all x gathers, so the idx->x dependent chain does not serialize
+//    (one wave keeps 8+8 loads in flight);
+//  - XCD-aware block swizzle (common.h) keeps neighbouring nnz chunks on one
+//    XCD's private L2 for x-gather reuse;
+//  - products go through LDS, then each thread sums its rows' segments — no
+//    per-nnz atomics; rows cut by a block boundary produce one carry per
+//    block, combined by a tiny fixup kernel;
+//  - optional fused dot: accumulates sum(p[r]*y[r]) per block (the CG p·Ap
+//    reduction) so the CG loop never re-reads p and q.
 #include "common.h"
 
 namespace {
@@ -18,7 +23,7 @@ constexpr int BLK = 256;
 constexpr int VT = 8;
 constexpr int64_t NNZ_PER_BLOCK = (int64_t)BLK * VT;  // 2048
 
-template <typename T, typename index_t, bool BETA_ZERO>
+template <typename T, typename index_t, bool BETA_ZERO, bool FUSE_DOT>
 __global__ __launch_bounds__(BLK) void spmv_kernel(
     const int64_t* __restrict__ indptr,  // m+1
     const index_t* __restrict__ indices,
@@ -26,33 +31,48 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
     const T* __restrict__ x,  // window, indexed by (indices[p] - col_lo)
     T* __restrict__ y,
     int64_t m, int64_t nnz, int64_t col_lo, T beta,
-    T* __restrict__ carry_val, int64_t* __restrict__ carry_row) {
+    T* __restrict__ carry_val, int64_t* __restrict__ carry_row,
+    const T* __restrict__ pvec,  // local p slab (rows align with y); FUSE_DOT
+    T* __restrict__ dot_partial) {
   extern __shared__ char smem_raw[];
   T* prod = reinterpret_cast<T*>(smem_raw);
   __shared__ __align__(16) char red_raw[BLK * sizeof(T)];
   T* red = reinterpret_cast<T*>(red_raw);
   __shared__ int64_t sh_ro0, sh_ro1;
 
-  const int64_t b = blockIdx.x;
+  const int64_t b = xcd_swizzle(blockIdx.x, gridDim.x);
   const int64_t s = b * NNZ_PER_BLOCK;
   const int64_t e = min(s + NNZ_PER_BLOCK, nnz);
   const int tid = threadIdx.x;
 
-  // stage products (coalesced value/index reads)
-  for (int64_t i = s + tid; i < e; i += BLK) {
-    prod[i - s] = vals[i] * x[(int64_t)indices[i] - col_lo];
+  // stage products (coalesced value/index reads, pipelined gathers)
+  if (e - s == NNZ_PER_BLOCK) {
+    index_t idx[VT];
+    T v[VT];
+#pragma unroll
+    for (int k = 0; k < VT; ++k) idx[k] = indices[s + tid + k * BLK];
+#pragma unroll
+    for (int k = 0; k < VT; ++k) v[k] = vals[s + tid + k * BLK];
+    T xv[VT];
+#pragma unroll
+    for (int k = 0; k < VT; ++k) xv[k] = x[(int64_t)idx[k] - col_lo];
+#pragma unroll
+    for (int k = 0; k < VT; ++k) prod[tid + k * BLK] = v[k] * xv[k];
+  } else {
+    for (int64_t i = s + tid; i < e; i += BLK) {
+      prod[i - s] = vals[i] * x[(int64_t)indices[i] - col_lo];
+    }
   }
 
   if (tid == 0) {
     // owned rows: first r with indptr[r] >= s .. first r with indptr[r] >= e
-    // (indptr[0]==0 is row 0's start; search over indptr[0..m))
     sh_ro0 = lb_i64(indptr, m, s);
     sh_ro1 = (e == nnz) ? m : lb_i64(indptr, m, e);
   }
   __syncthreads();
   const int64_t ro0 = sh_ro0, ro1 = sh_ro1;
 
-  // per-thread row sums (segments start at >= cend, disjoint from carry)
+  T dacc = ZeroOf<T>::value();
   for (int64_t r = ro0 + tid; r < ro1; r += BLK) {
     int64_t rs = indptr[r];
     int64_t re = min(indptr[r + 1], e);
@@ -61,14 +81,18 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
     if (BETA_ZERO) {
       y[r] = acc;
     } else {
-      y[r] = acc + beta * y[r];
+      acc = acc + beta * y[r];
+      y[r] = acc;
     }
+    if (FUSE_DOT) dacc += acc * pvec[r];
   }
 
   // continuation carry: items [s, cend) belong to row ro0-1
+  bool has_carry = false;
   if (ro0 > 0) {
     int64_t cend = (ro0 < m) ? min(indptr[ro0], e) : e;
     if (cend > s) {
+      has_carry = true;
       T acc = ZeroOf<T>::value();
       for (int64_t p = s + tid; p < cend; p += BLK) acc += prod[p - s];
       red[tid] = acc;
@@ -81,20 +105,53 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
         carry_val[b] = red[0];
         carry_row[b] = ro0 - 1;
       }
-      return;
     }
   }
-  if (tid == 0) carry_row[b] = -1;
+  if (!has_carry && tid == 0) carry_row[b] = -1;
+
+  if (FUSE_DOT) {
+    __syncthreads();
+    red[tid] = dacc;
+    __syncthreads();
+    for (int w = BLK / 2; w > 0; w >>= 1) {
+      if (tid < w) red[tid] += red[tid + w];
+      __syncthreads();
+    }
+    // per-block partial; one atomic per address would serialize at 163k
+    // blocks, so the fixup kernel reduces these (nblocks/256 atomics).
+    if (tid == 0) dot_partial[b] = red[0];
+  }
 }
 
-template <typename T>
+template <typename T, bool FUSE_DOT>
 __global__ void carry_fixup_kernel(const T* __restrict__ carry_val,
                                    const int64_t* __restrict__ carry_row,
-                                   T* __restrict__ y, int64_t nblocks) {
+                                   T* __restrict__ y, int64_t nblocks,
+                                   const T* __restrict__ pvec,
+                                   const T* __restrict__ dot_partial,
+                                   T* __restrict__ dot_out) {
+  __shared__ __align__(16) char red_raw[256 * sizeof(T)];
+  T* red = reinterpret_cast<T*>(red_raw);
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= nblocks) return;
-  int64_t r = carry_row[i];
-  if (r >= 0) atomic_add_any(&y[r], carry_val[i]);
+  T dacc = ZeroOf<T>::value();
+  if (i < nblocks) {
+    int64_t r = carry_row[i];
+    if (r >= 0) {
+      T c = carry_val[i];
+      atomic_add_any(&y[r], c);
+      if (FUSE_DOT) dacc = c * pvec[r];
+    }
+    if (FUSE_DOT) dacc += dot_partial[i];
+  }
+  if (FUSE_DOT) {
+    red[threadIdx.x] = dacc;
+    __syncthreads();
+    for (int w = 128; w > 0; w >>= 1) {
+      if ((int)threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) atomic_add_any(dot_out, red[0]);
+  }
 }
 
 template <typename T>
@@ -105,11 +162,14 @@ __global__ void scale_kernel(T* y, int64_t n, T beta) {
 
 }  // namespace
 
-void spmv_hip(at::Tensor indptr, at::Tensor indices, at::Tensor values,
-              at::Tensor x, at::Tensor y, int64_t col_lo, double beta) {
+static void spmv_impl(at::Tensor indptr, at::Tensor indices, at::Tensor values,
+                      at::Tensor x, at::Tensor y, int64_t col_lo, double beta,
+                      const c10::optional<at::Tensor>& pvec,
+                      const c10::optional<at::Tensor>& dot_out) {
   const int64_t m = indptr.numel() - 1;
   const int64_t nnz = values.numel();
   auto stream = cur_stream();
+  const bool fuse_dot = pvec.has_value();
   DISPATCH_VALUES(values.scalar_type(), "spmv", [&] {
     using T = scalar_t;
     T betav = static_cast<T>(beta);
@@ -125,26 +185,48 @@ void spmv_hip(at::Tensor indptr, at::Tensor indices, at::Tensor values,
     const int64_t nblocks = (nnz + NNZ_PER_BLOCK - 1) / NNZ_PER_BLOCK;
     auto carry_val = at::empty({nblocks}, values.options());
     auto carry_row = at::empty({nblocks}, indptr.options());
+    at::Tensor dot_partial;
+    if (fuse_dot) dot_partial = at::empty({nblocks}, values.options());
     size_t smem = NNZ_PER_BLOCK * sizeof(T);
+    const T* pp = fuse_dot ? pvec->data_ptr<T>() : nullptr;
+    T* dpart = fuse_dot ? dot_partial.data_ptr<T>() : nullptr;
+    T* dp = fuse_dot ? dot_out->data_ptr<T>() : nullptr;
     DISPATCH_INDEX(indices.scalar_type(), "spmv_idx", [&] {
-      if (beta == 0.0) {
-        hipLaunchKernelGGL((spmv_kernel<T, index_t, true>), dim3(nblocks),
-                           dim3(BLK), smem, stream,
+      auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(nblocks), dim3(BLK), smem, stream,
                            indptr.data_ptr<int64_t>(), indices.data_ptr<index_t>(),
                            values.data_ptr<T>(), x.data_ptr<T>(), y.data_ptr<T>(),
-                           m, nnz, col_lo, betav,
-                           carry_val.data_ptr<T>(), carry_row.data_ptr<int64_t>());
-      } else {
-        hipLaunchKernelGGL((spmv_kernel<T, index_t, false>), dim3(nblocks),
-                           dim3(BLK), smem, stream,
-                           indptr.data_ptr<int64_t>(), indices.data_ptr<index_t>(),
-                           values.data_ptr<T>(), x.data_ptr<T>(), y.data_ptr<T>(),
-                           m, nnz, col_lo, betav,
-                           carry_val.data_ptr<T>(), carry_row.data_ptr<int64_t>());
-      }
+                           m, nnz, col_lo, betav, carry_val.data_ptr<T>(),
+                           carry_row.data_ptr<int64_t>(), pp, dpart);
+      };
+      if (beta == 0.0 && !fuse_dot) launch(spmv_kernel<T, index_t, true, false>);
+      else if (beta == 0.0 && fuse_dot) launch(spmv_kernel<T, index_t, true, true>);
+      else if (!fuse_dot) launch(spmv_kernel<T, index_t, false, false>);
+      else launch(spmv_kernel<T, index_t, false, true>);
     });
-    hipLaunchKernelGGL(carry_fixup_kernel<T>, dim3((nblocks + 255) / 256),
-                       dim3(256), 0, stream, carry_val.data_ptr<T>(),
-                       carry_row.data_ptr<int64_t>(), y.data_ptr<T>(), nblocks);
+    if (fuse_dot) {
+      hipLaunchKernelGGL((carry_fixup_kernel<T, true>), dim3((nblocks + 255) / 256),
+                         dim3(256), 0, stream, carry_val.data_ptr<T>(),
+                         carry_row.data_ptr<int64_t>(), y.data_ptr<T>(), nblocks,
+                         pp, dpart, dp);
+    } else {
+      hipLaunchKernelGGL((carry_fixup_kernel<T, false>), dim3((nblocks + 255) / 256),
+                         dim3(256), 0, stream, carry_val.data_ptr<T>(),
+                         carry_row.data_ptr<int64_t>(), y.data_ptr<T>(), nblocks,
+                         pp, dpart, dp);
+    }
   });
+}
+
+void spmv_hip(at::Tensor indptr, at::Tensor indices, at::Tensor values,
+              at::Tensor x, at::Tensor y, int64_t col_lo, double beta) {
+  spmv_impl(indptr, indices, values, x, y, col_lo, beta, c10::nullopt,
+            c10::nullopt);
+}
+
+// fused q = A p ; dot_out += sum(p_local * q_local)
+void spmv_dot_hip(at::Tensor indptr, at::Tensor indices, at::Tensor values,
+                  at::Tensor x, at::Tensor y, at::Tensor pvec,
+                  at::Tensor dot_out, int64_t col_lo) {
+  spmv_impl(indptr, indices, values, x, y, col_lo, 0.0, pvec, dot_out);
 }

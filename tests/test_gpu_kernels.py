@@ -249,3 +249,33 @@ def test_native_extension_is_loaded():
 
     maps = open("/proc/self/maps").read()
     assert "sparse_hip.so" in maps
+
+
+def test_spmv_dot_fused_gpu():
+    from sparse import csr_array, darray
+
+    s = sample_csr(5000, 5000, 0.002, seed=33)
+    A = csr_array(s)
+    p = darray.random((5000,), seed=34)
+    q = darray.zeros((5000,))
+    pq = A.spmv_dot(p, q)
+    qref = s @ np.asarray(p)
+    assert np.allclose(np.asarray(q), qref, rtol=1e-10)
+    assert np.isclose(float(pq.item()), float(np.asarray(p) @ qref), rtol=1e-10)
+
+
+def test_axpby_norm2_gpu():
+    from sparse import darray
+    from sparse.linalg import _axpby_norm2
+
+    n = 50000
+    y = darray.random((n,), seed=35)
+    x = darray.random((n,), seed=36)
+    y0 = np.asarray(y).copy()
+    x0 = np.asarray(x)
+    a = torch.tensor(2.0, device="cuda", dtype=torch.float64)
+    b = torch.tensor(4.0, device="cuda", dtype=torch.float64)
+    rz = _axpby_norm2(y, x, a, b, negate=True)
+    expect = y0 - 0.5 * x0
+    assert np.allclose(np.asarray(y), expect, rtol=1e-12)
+    assert np.isclose(float(rz.item()), float(expect @ expect), rtol=1e-10)
