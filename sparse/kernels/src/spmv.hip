@@ -1,27 +1,34 @@
 // CSR SpMV, nnz-split with LDS-staged products — MI355X (gfx950) native.
 //
 // Replaces reference CSR_SPMV_ROW_SPLIT (src/sparse/array/csr/spmv.cu:25-123,
-// cuSPARSE-backed there).  Design for CDNA4:
-//  - each 256-thread block owns NNZ_PER_BLOCK consecutive nonzeros, so the
-//    values and indices streams are read fully coalesced exactly once
-//    (the kernel is HBM-bound: ~12 B/nnz for fp64+int32);
-//  - the staging loop is register-pipelined: all index loads issue first,
-//    then all x gathers, so the idx->x dependent chain does not serialize
-//    (one wave keeps 8+8 loads in flight);
-//  - XCD-aware block swizzle (common.h) keeps neighbouring nnz chunks on one
-//    XCD's private L2 for x-gather reuse;
-//  - products go through LDS, then each thread sums its rows' segments — no
-//    per-nnz atomics; rows cut by a block boundary produce one carry per
-//    block, combined by a tiny fixup kernel;
-//  - optional fused dot: accumulates sum(p[r]*y[r]) per block (the CG p·Ap
-//    reduction) so the CG loop never re-reads p and q.
+// cuSPARSE-backed there).  Design for CDNA4 (measured on MI355X):
+//  - each 256-thread block owns NNZ_PER_BLOCK consecutive nonzeros; values
+//    and indices stream through 2-element vector loads (512B-1KB per wave
+//    instruction, fully coalesced, read exactly once);
+//  - ALL global loads (values, indices, x gathers, the block's indptr
+//    segment) issue in one phase before the single barrier; the row-sum
+//    phase then runs purely out of LDS — PMC showed the original
+//    two-phase form 90% wave-parked because the sum phase had ~2 loads in
+//    flight per wave (3.3 TB/s); this restructure lifts HBM pressure;
+//  - XCD-aware block swizzle (common.h) keeps neighbouring nnz chunks on
+//    one XCD's private L2 for x-gather reuse;
+//  - rows cut by a block boundary produce one carry per block, combined by
+//    a tiny fixup kernel (atomic-free main path);
+//  - optional fused dot accumulates sum(p[r]*y[r]) per block (CG p·Ap);
+//    the fixup kernel reduces the per-block partials.
 #include "common.h"
 
 namespace {
 
 constexpr int BLK = 256;
-constexpr int VT = 8;
+constexpr int VT = 8;  // nnz per thread (4 vector pairs)
 constexpr int64_t NNZ_PER_BLOCK = (int64_t)BLK * VT;  // 2048
+constexpr int MAXROWS = (int)NNZ_PER_BLOCK;           // LDS row-offset capacity
+
+template <typename U>
+struct alignas(2 * sizeof(U)) Pair {
+  U a, b;
+};
 
 template <typename T, typename index_t, bool BETA_ZERO, bool FUSE_DOT>
 __global__ __launch_bounds__(BLK) void spmv_kernel(
@@ -35,62 +42,110 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
     const T* __restrict__ pvec,  // local p slab (rows align with y); FUSE_DOT
     T* __restrict__ dot_partial) {
   extern __shared__ char smem_raw[];
-  T* prod = reinterpret_cast<T*>(smem_raw);
+  T* prod = reinterpret_cast<T*>(smem_raw);                       // NNZ_PER_BLOCK
+  int* rowoff = reinterpret_cast<int*>(prod + NNZ_PER_BLOCK);     // MAXROWS+1
   __shared__ __align__(16) char red_raw[BLK * sizeof(T)];
   T* red = reinterpret_cast<T*>(red_raw);
-  __shared__ int64_t sh_ro0, sh_ro1;
 
   const int64_t b = xcd_swizzle(blockIdx.x, gridDim.x);
   const int64_t s = b * NNZ_PER_BLOCK;
   const int64_t e = min(s + NNZ_PER_BLOCK, nnz);
   const int tid = threadIdx.x;
 
-  // stage products (coalesced value/index reads, pipelined gathers)
-  if (e - s == NNZ_PER_BLOCK) {
-    index_t idx[VT];
-    T v[VT];
+  // owned rows (redundant per-thread binary search; uniform -> broadcast)
+  const int64_t ro0 = lb_i64(indptr, m, s);
+  const int64_t ro1 = (e == nnz) ? m : lb_i64(indptr, m, e);
+  const int64_t nrows_blk = ro1 - ro0;
+  const bool lds_rows = nrows_blk <= MAXROWS;
+
+  // ---- phase 1: issue every global load ----------------------------------
+  const bool full = (e - s) == NNZ_PER_BLOCK;
+  if (full) {
+    Pair<index_t> idx2[VT / 2];
+    Pair<T> v2[VT / 2];
 #pragma unroll
-    for (int k = 0; k < VT; ++k) idx[k] = indices[s + tid + k * BLK];
+    for (int k = 0; k < VT / 2; ++k) {
+      idx2[k] = *reinterpret_cast<const Pair<index_t>*>(
+          &indices[s + tid * 2 + (int64_t)k * (2 * BLK)]);
+    }
 #pragma unroll
-    for (int k = 0; k < VT; ++k) v[k] = vals[s + tid + k * BLK];
-    T xv[VT];
+    for (int k = 0; k < VT / 2; ++k) {
+      v2[k] = *reinterpret_cast<const Pair<T>*>(
+          &vals[s + tid * 2 + (int64_t)k * (2 * BLK)]);
+    }
+    // stage the block's indptr segment while the pair loads are in flight
+    if (lds_rows) {
+      for (int64_t i = tid; i <= nrows_blk; i += BLK) {
+        int64_t off = indptr[ro0 + i] - s;
+        rowoff[i] = (int)min(off, NNZ_PER_BLOCK);
+      }
+    }
+    T xa[VT / 2], xb[VT / 2];
 #pragma unroll
-    for (int k = 0; k < VT; ++k) xv[k] = x[(int64_t)idx[k] - col_lo];
+    for (int k = 0; k < VT / 2; ++k) {
+      xa[k] = x[(int64_t)idx2[k].a - col_lo];
+      xb[k] = x[(int64_t)idx2[k].b - col_lo];
+    }
 #pragma unroll
-    for (int k = 0; k < VT; ++k) prod[tid + k * BLK] = v[k] * xv[k];
+    for (int k = 0; k < VT / 2; ++k) {
+      Pair<T> pr{v2[k].a * xa[k], v2[k].b * xb[k]};
+      *reinterpret_cast<Pair<T>*>(&prod[tid * 2 + k * (2 * BLK)]) = pr;
+    }
   } else {
+    if (lds_rows) {
+      for (int64_t i = tid; i <= nrows_blk; i += BLK) {
+        int64_t off = indptr[ro0 + i] - s;
+        rowoff[i] = (int)min(off, NNZ_PER_BLOCK);
+      }
+    }
     for (int64_t i = s + tid; i < e; i += BLK) {
       prod[i - s] = vals[i] * x[(int64_t)indices[i] - col_lo];
     }
   }
-
-  if (tid == 0) {
-    // owned rows: first r with indptr[r] >= s .. first r with indptr[r] >= e
-    sh_ro0 = lb_i64(indptr, m, s);
-    sh_ro1 = (e == nnz) ? m : lb_i64(indptr, m, e);
-  }
   __syncthreads();
-  const int64_t ro0 = sh_ro0, ro1 = sh_ro1;
 
+  // ---- phase 2: per-thread row sums out of LDS ---------------------------
   T dacc = ZeroOf<T>::value();
-  for (int64_t r = ro0 + tid; r < ro1; r += BLK) {
-    int64_t rs = indptr[r];
-    int64_t re = min(indptr[r + 1], e);
-    T acc = ZeroOf<T>::value();
-    for (int64_t p = rs; p < re; ++p) acc += prod[p - s];
-    if (BETA_ZERO) {
-      y[r] = acc;
-    } else {
-      acc = acc + beta * y[r];
-      y[r] = acc;
+  if (lds_rows) {
+    for (int64_t r = ro0 + tid; r < ro1; r += BLK) {
+      const int i = (int)(r - ro0);
+      const int64_t rs = s + rowoff[i];
+      const int64_t re = min((int64_t)s + rowoff[i + 1], e);
+      T acc = ZeroOf<T>::value();
+      for (int64_t p = rs; p < re; ++p) acc += prod[p - s];
+      if (BETA_ZERO) {
+        y[r] = acc;
+      } else {
+        acc = acc + beta * y[r];
+        y[r] = acc;
+      }
+      if (FUSE_DOT) dacc += acc * pvec[r];
     }
-    if (FUSE_DOT) dacc += acc * pvec[r];
+  } else {
+    for (int64_t r = ro0 + tid; r < ro1; r += BLK) {
+      const int64_t rs = indptr[r];
+      const int64_t re = min(indptr[r + 1], e);
+      T acc = ZeroOf<T>::value();
+      for (int64_t p = rs; p < re; ++p) acc += prod[p - s];
+      if (BETA_ZERO) {
+        y[r] = acc;
+      } else {
+        acc = acc + beta * y[r];
+        y[r] = acc;
+      }
+      if (FUSE_DOT) dacc += acc * pvec[r];
+    }
   }
 
   // continuation carry: items [s, cend) belong to row ro0-1
   bool has_carry = false;
   if (ro0 > 0) {
-    int64_t cend = (ro0 < m) ? min(indptr[ro0], e) : e;
+    int64_t cend;
+    if (ro0 < m) {
+      cend = lds_rows ? min((int64_t)s + rowoff[0], e) : min(indptr[ro0], e);
+    } else {
+      cend = e;
+    }
     if (cend > s) {
       has_carry = true;
       T acc = ZeroOf<T>::value();
@@ -187,7 +242,7 @@ static void spmv_impl(at::Tensor indptr, at::Tensor indices, at::Tensor values,
     auto carry_row = at::empty({nblocks}, indptr.options());
     at::Tensor dot_partial;
     if (fuse_dot) dot_partial = at::empty({nblocks}, values.options());
-    size_t smem = NNZ_PER_BLOCK * sizeof(T);
+    size_t smem = NNZ_PER_BLOCK * sizeof(T) + (MAXROWS + 1) * sizeof(int);
     const T* pp = fuse_dot ? pvec->data_ptr<T>() : nullptr;
     T* dpart = fuse_dot ? dot_partial.data_ptr<T>() : nullptr;
     T* dp = fuse_dot ? dot_out->data_ptr<T>() : nullptr;
