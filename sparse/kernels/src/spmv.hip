@@ -21,13 +21,12 @@
 namespace {
 
 constexpr int BLK = 256;
-constexpr int VT = 8;  // nnz per thread (4 vector pairs)
+constexpr int VT = 8;  // nnz per thread (2 vector quads)
 constexpr int64_t NNZ_PER_BLOCK = (int64_t)BLK * VT;  // 2048
-constexpr int MAXROWS = (int)NNZ_PER_BLOCK;           // LDS row-offset capacity
 
 template <typename U>
-struct alignas(2 * sizeof(U)) Pair {
-  U a, b;
+struct alignas(4 * sizeof(U) <= 16 ? 4 * sizeof(U) : 16) Quad {
+  U v[4];
 };
 
 template <typename T, typename index_t, bool BETA_ZERO, bool FUSE_DOT>
@@ -42,8 +41,7 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
     const T* __restrict__ pvec,  // local p slab (rows align with y); FUSE_DOT
     T* __restrict__ dot_partial) {
   extern __shared__ char smem_raw[];
-  T* prod = reinterpret_cast<T*>(smem_raw);                       // NNZ_PER_BLOCK
-  int* rowoff = reinterpret_cast<int*>(prod + NNZ_PER_BLOCK);     // MAXROWS+1
+  T* prod = reinterpret_cast<T*>(smem_raw);  // NNZ_PER_BLOCK
   __shared__ __align__(16) char red_raw[BLK * sizeof(T)];
   T* red = reinterpret_cast<T*>(red_raw);
 
@@ -55,49 +53,38 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
   // owned rows (redundant per-thread binary search; uniform -> broadcast)
   const int64_t ro0 = lb_i64(indptr, m, s);
   const int64_t ro1 = (e == nnz) ? m : lb_i64(indptr, m, e);
-  const int64_t nrows_blk = ro1 - ro0;
-  const bool lds_rows = nrows_blk <= MAXROWS;
 
   // ---- phase 1: issue every global load ----------------------------------
   const bool full = (e - s) == NNZ_PER_BLOCK;
   if (full) {
-    Pair<index_t> idx2[VT / 2];
-    Pair<T> v2[VT / 2];
+    Quad<index_t> idx4[VT / 4];
+    Quad<T> v4[VT / 4];
 #pragma unroll
-    for (int k = 0; k < VT / 2; ++k) {
-      idx2[k] = *reinterpret_cast<const Pair<index_t>*>(
-          &indices[s + tid * 2 + (int64_t)k * (2 * BLK)]);
+    for (int k = 0; k < VT / 4; ++k) {
+      idx4[k] = *reinterpret_cast<const Quad<index_t>*>(
+          &indices[s + tid * 4 + (int64_t)k * (4 * BLK)]);
     }
 #pragma unroll
-    for (int k = 0; k < VT / 2; ++k) {
-      v2[k] = *reinterpret_cast<const Pair<T>*>(
-          &vals[s + tid * 2 + (int64_t)k * (2 * BLK)]);
+    for (int k = 0; k < VT / 4; ++k) {
+      v4[k] = *reinterpret_cast<const Quad<T>*>(
+          &vals[s + tid * 4 + (int64_t)k * (4 * BLK)]);
     }
-    // stage the block's indptr segment while the pair loads are in flight
-    if (lds_rows) {
-      for (int64_t i = tid; i <= nrows_blk; i += BLK) {
-        int64_t off = indptr[ro0 + i] - s;
-        rowoff[i] = (int)min(off, NNZ_PER_BLOCK);
+    T xv[VT];
+#pragma unroll
+    for (int k = 0; k < VT / 4; ++k) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        xv[k * 4 + j] = x[(int64_t)idx4[k].v[j] - col_lo];
       }
     }
-    T xa[VT / 2], xb[VT / 2];
 #pragma unroll
-    for (int k = 0; k < VT / 2; ++k) {
-      xa[k] = x[(int64_t)idx2[k].a - col_lo];
-      xb[k] = x[(int64_t)idx2[k].b - col_lo];
-    }
+    for (int k = 0; k < VT / 4; ++k) {
+      Quad<T> pr;
 #pragma unroll
-    for (int k = 0; k < VT / 2; ++k) {
-      Pair<T> pr{v2[k].a * xa[k], v2[k].b * xb[k]};
-      *reinterpret_cast<Pair<T>*>(&prod[tid * 2 + k * (2 * BLK)]) = pr;
+      for (int j = 0; j < 4; ++j) pr.v[j] = v4[k].v[j] * xv[k * 4 + j];
+      *reinterpret_cast<Quad<T>*>(&prod[tid * 4 + k * (4 * BLK)]) = pr;
     }
   } else {
-    if (lds_rows) {
-      for (int64_t i = tid; i <= nrows_blk; i += BLK) {
-        int64_t off = indptr[ro0 + i] - s;
-        rowoff[i] = (int)min(off, NNZ_PER_BLOCK);
-      }
-    }
     for (int64_t i = s + tid; i < e; i += BLK) {
       prod[i - s] = vals[i] * x[(int64_t)indices[i] - col_lo];
     }
@@ -106,35 +93,18 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
 
   // ---- phase 2: per-thread row sums out of LDS ---------------------------
   T dacc = ZeroOf<T>::value();
-  if (lds_rows) {
-    for (int64_t r = ro0 + tid; r < ro1; r += BLK) {
-      const int i = (int)(r - ro0);
-      const int64_t rs = s + rowoff[i];
-      const int64_t re = min((int64_t)s + rowoff[i + 1], e);
-      T acc = ZeroOf<T>::value();
-      for (int64_t p = rs; p < re; ++p) acc += prod[p - s];
-      if (BETA_ZERO) {
-        y[r] = acc;
-      } else {
-        acc = acc + beta * y[r];
-        y[r] = acc;
-      }
-      if (FUSE_DOT) dacc += acc * pvec[r];
+  for (int64_t r = ro0 + tid; r < ro1; r += BLK) {
+    const int64_t rs = indptr[r];
+    const int64_t re = min(indptr[r + 1], e);
+    T acc = ZeroOf<T>::value();
+    for (int64_t p = rs; p < re; ++p) acc += prod[p - s];
+    if (BETA_ZERO) {
+      y[r] = acc;
+    } else {
+      acc = acc + beta * y[r];
+      y[r] = acc;
     }
-  } else {
-    for (int64_t r = ro0 + tid; r < ro1; r += BLK) {
-      const int64_t rs = indptr[r];
-      const int64_t re = min(indptr[r + 1], e);
-      T acc = ZeroOf<T>::value();
-      for (int64_t p = rs; p < re; ++p) acc += prod[p - s];
-      if (BETA_ZERO) {
-        y[r] = acc;
-      } else {
-        acc = acc + beta * y[r];
-        y[r] = acc;
-      }
-      if (FUSE_DOT) dacc += acc * pvec[r];
-    }
+    if (FUSE_DOT) dacc += acc * pvec[r];
   }
 
   // continuation carry: items [s, cend) belong to row ro0-1
@@ -142,7 +112,7 @@ __global__ __launch_bounds__(BLK) void spmv_kernel(
   if (ro0 > 0) {
     int64_t cend;
     if (ro0 < m) {
-      cend = lds_rows ? min((int64_t)s + rowoff[0], e) : min(indptr[ro0], e);
+      cend = min(indptr[ro0], e);
     } else {
       cend = e;
     }
@@ -242,7 +212,7 @@ static void spmv_impl(at::Tensor indptr, at::Tensor indices, at::Tensor values,
     auto carry_row = at::empty({nblocks}, indptr.options());
     at::Tensor dot_partial;
     if (fuse_dot) dot_partial = at::empty({nblocks}, values.options());
-    size_t smem = NNZ_PER_BLOCK * sizeof(T) + (MAXROWS + 1) * sizeof(int);
+    size_t smem = NNZ_PER_BLOCK * sizeof(T);
     const T* pp = fuse_dot ? pvec->data_ptr<T>() : nullptr;
     T* dpart = fuse_dot ? dot_partial.data_ptr<T>() : nullptr;
     T* dp = fuse_dot ? dot_out->data_ptr<T>() : nullptr;
