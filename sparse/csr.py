@@ -173,6 +173,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         self._nnz_cache = None
         self._window_cache = None
         self._plan_cache = {}
+        self._ell_cache = None
 
     @classmethod
     def from_local(cls, indptr, indices, values, partition, shape) -> "csr_array":
@@ -196,6 +197,17 @@ class csr_array(CompressedBase, DenseSparseBase):
                 self._window_cache = (int(self._indices.min().item()),
                                       int(self._indices.max().item()) + 1)
         return self._window_cache
+
+    def _ell(self):
+        """Cached padded-ELL mirror (GPU fast SpMV; kernels.build_ell)."""
+        if not self._values.is_cuda or self._ell_cache == "no":
+            return None
+        if self._ell_cache is None:
+            from . import kernels
+
+            kernels.require()
+            self._ell_cache = kernels.build_ell(self.local) or "no"
+        return None if self._ell_cache == "no" else self._ell_cache
 
     def _xplan(self, xpart: RowPartition) -> WindowGatherPlan:
         key = ("x", xpart.starts)
@@ -227,6 +239,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         off = sum(counts[: comm.rank()])
         mine = np.asarray(v)[off: off + counts[comm.rank()]]
         self._values = torch.as_tensor(mine, device=self._values.device).to(self._values.dtype)
+        self._ell_cache = None
 
     @property
     def indices(self) -> np.ndarray:
@@ -354,10 +367,18 @@ class csr_array(CompressedBase, DenseSparseBase):
         plan = self._xplan(x.partition)
         vdt = self._out_dtype(x.local.dtype)
         xw = plan.gather(x.local.to(vdt))
-        lc = self.local
-        if lc.values.dtype != vdt:
-            lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt), lc.nrows, lc.ncols)
-        ylocal = ops.spmv(lc, xw, col_lo=plan.lo)
+        ell = self._ell() if self._values.dtype == vdt else None
+        if ell is not None:
+            from . import kernels
+
+            ylocal = torch.empty(self.partition.count(comm.rank()), dtype=vdt,
+                                 device=self._values.device)
+            kernels.ell_spmv(ell, xw, ylocal, plan.lo)
+        else:
+            lc = self.local
+            if lc.values.dtype != vdt:
+                lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt), lc.nrows, lc.ncols)
+            ylocal = ops.spmv(lc, xw, col_lo=plan.lo)
         part = self.partition
         if out is not None:
             out.local.copy_(ylocal.to(out.local.dtype))
@@ -386,9 +407,13 @@ class csr_array(CompressedBase, DenseSparseBase):
 
         plan = self._xplan(p.partition)
         xw = plan.gather(p.local)
-        dot = torch.zeros((), dtype=self._values.dtype, device=self._values.device)
-        lc = self.local
-        kernels.spmv_dot(lc, xw, q.local, p.local, dot, plan.lo)
+        ell = self._ell()
+        if ell is not None:
+            dot = kernels.ell_spmv_dot(ell, xw, q.local, p.local, plan.lo)
+        else:
+            dot = torch.zeros((), dtype=self._values.dtype, device=self._values.device)
+            lc = self.local
+            kernels.spmv_dot(lc, xw, q.local, p.local, dot, plan.lo)
         comm.all_reduce_(dot)
         return dot
 

@@ -202,3 +202,49 @@ def rk_calc_dy(K, a, h, dy):
 
 def cdist(XA, XB, out):
     ext().cdist(XA, XB, out)
+
+
+# -- ELL fast path ------------------------------------------------------------
+class EllMirror:
+    """Column-major padded-ELL copy of a row-uniform CSR slab (fast SpMV)."""
+
+    __slots__ = ("eidx", "evals", "W", "m")
+
+    def __init__(self, eidx, evals, W, m):
+        self.eidx = eidx
+        self.evals = evals
+        self.W = W
+        self.m = m
+
+
+def build_ell(A):
+    """Build the ELL mirror of a LocalCSR, or None when unprofitable
+    (W too large or padding blowup > 1.6x the CSR bytes)."""
+    m = A.nrows
+    if m == 0 or A.nnz == 0:
+        return None
+    counts = A.indptr[1:] - A.indptr[:-1]
+    W = int(counts.max().item())
+    if W == 0 or W > 48:
+        return None
+    mp = (m + 1) // 2 * 2
+    if W * mp > 1.6 * A.nnz + 4096:
+        return None
+    eidx = torch.empty(W * mp, dtype=A.indices.dtype, device=A.device)
+    evals = torch.empty(W * mp, dtype=A.values.dtype, device=A.device)
+    pad_idx = int(A.indices[0].item())
+    ext().build_ell(A.indptr, A.indices, A.values, eidx, evals, W, pad_idx)
+    return EllMirror(eidx, evals, W, m)
+
+
+def ell_spmv(ell: EllMirror, x, y, col_lo: int):
+    ext().ell_spmv(ell.eidx, ell.evals, x, y, ell.W, ell.m, int(col_lo))
+
+
+def ell_spmv_dot(ell: EllMirror, x, y, p, col_lo: int):
+    mp = ell.evals.numel() // ell.W
+    nblocks = (mp // 2 + 255) // 256
+    partial = torch.empty(nblocks, dtype=ell.evals.dtype, device=ell.evals.device)
+    ext().ell_spmv_dot(ell.eidx, ell.evals, x, y, p, partial, ell.W, ell.m,
+                       int(col_lo))
+    return partial.sum()

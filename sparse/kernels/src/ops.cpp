@@ -13,6 +13,12 @@ void spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                   at::Tensor, at::Tensor, int64_t);
 void axpby_norm2_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool,
                      bool, at::Tensor);
+void build_ell_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                   int64_t, int64_t);
+void ell_spmv_plain_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                        int64_t, int64_t, int64_t);
+void ell_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      at::Tensor, at::Tensor, int64_t, int64_t, int64_t);
 void add_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void add_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                      at::Tensor, at::Tensor, at::Tensor, at::Tensor, double,
@@ -63,6 +69,12 @@ TORCH_LIBRARY(sparse_hip, m) {
         "Tensor(a!) y, Tensor pvec, Tensor(b!) dot_out, int col_lo) -> ()");
   m.def("axpby_norm2(Tensor(a!) y, Tensor x, Tensor a, Tensor b, bool isalpha, "
         "bool negate, Tensor(b!) dot_out) -> ()");
+  m.def("build_ell(Tensor indptr, Tensor indices, Tensor values, "
+        "Tensor(a!) eidx, Tensor(b!) evals, int W, int pad_idx) -> ()");
+  m.def("ell_spmv(Tensor eidx, Tensor evals, Tensor x, Tensor(a!) y, int W, "
+        "int m, int col_lo) -> ()");
+  m.def("ell_spmv_dot(Tensor eidx, Tensor evals, Tensor x, Tensor(a!) y, "
+        "Tensor pvec, Tensor(b!) dot_partial, int W, int m, int col_lo) -> ()");
   m.def("spmm(Tensor indptr, Tensor indices, Tensor vals, Tensor B, "
         "Tensor(a!) C, int col_lo) -> ()");
   m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "
@@ -98,6 +110,9 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("axpby", axpby_hip);
   m.impl("spmv_dot", spmv_dot_hip);
   m.impl("axpby_norm2", axpby_norm2_hip);
+  m.impl("build_ell", build_ell_hip);
+  m.impl("ell_spmv", ell_spmv_plain_hip);
+  m.impl("ell_spmv_dot", ell_spmv_dot_hip);
   m.impl("spmm", spmm_hip);
   m.impl("rspmm", rspmm_hip);
   m.impl("sddmm", sddmm_hip);
