@@ -1,0 +1,123 @@
+"""SPMD worker for the distributed (gloo, world_size>1) test battery.
+
+Launched by test_distributed.py via torchrun; every rank runs the same
+checks against a replicated scipy oracle and asserts locally — any failure
+exits nonzero.
+"""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import scipy.sparse as sps
+import torch
+import torch.distributed as dist
+
+
+def main():
+    dist.init_process_group("gloo")
+    import sparse
+    from sparse import coo_array, csc_array, csr_array, darray, gallery, linalg
+    from sparse.parallel import comm
+
+    ws = dist.get_world_size()
+    rank = dist.get_rank()
+    assert ws > 1
+
+    rng = np.random.default_rng(7)
+
+    # ---- construction + properties are rank-invariant ----------------------
+    s = sps.random(37, 29, 0.2, random_state=1, format="csr")
+    s.sort_indices()
+    A = csr_array(s)
+    assert A.nnz == s.nnz
+    assert np.allclose(np.asarray(A.todense()), s.toarray())
+    assert np.array_equal(A.indptr, s.indptr.astype(np.int64))
+    assert np.allclose(A.data, s.data)
+
+    # ---- SpMV with window gather -------------------------------------------
+    x = rng.random(29)
+    y = A @ x
+    assert np.allclose(np.asarray(y), s @ x), "dist spmv"
+
+    # ---- SpMM ---------------------------------------------------------------
+    B = rng.random((29, 5))
+    assert np.allclose(np.asarray(A @ B), s @ B), "dist spmm"
+
+    # ---- rspmm --------------------------------------------------------------
+    Ad = rng.random((3, 37))
+    assert np.allclose(np.asarray(Ad @ A), Ad @ s), "dist rspmm"
+
+    # ---- elementwise (includes repartition alignment) ----------------------
+    s2 = sps.random(37, 29, 0.25, random_state=2, format="csr")
+    A2 = csr_array(s2)
+    assert np.allclose(np.asarray((A + A2).todense()), (s + s2).toarray()), "dist add"
+    assert np.allclose(np.asarray(A.multiply(A2).todense()),
+                       s.multiply(s2).toarray()), "dist mult"
+
+    # ---- SpGEMM (gather_csr_rows path) --------------------------------------
+    s3 = sps.random(29, 23, 0.3, random_state=3, format="csr")
+    A3 = csr_array(s3)
+    C = A @ A3
+    assert np.allclose(np.asarray(C.todense()), (s @ s3).toarray()), "dist spgemm"
+
+    # ---- conversions (shuffle-to-owner paths) -------------------------------
+    assert np.allclose(np.asarray(A.tocsc().todense()), s.toarray()), "dist tocsc"
+    assert np.allclose(np.asarray(A.tocsc().tocsr().todense()), s.toarray()), "dist csc->csr"
+    co = A.tocoo()
+    assert np.allclose(np.asarray(co.tocsr().todense()), s.toarray()), "dist coo->csr"
+    assert np.allclose(np.asarray(A.T.todense()), s.T.toarray()), "dist T view"
+
+    # ---- CSC spmv (col-split + reduce-scatter) ------------------------------
+    Ac = csc_array(s.tocsc())
+    assert np.allclose(np.asarray(Ac @ x), s @ x), "dist csc spmv"
+
+    # ---- COO construction from triples --------------------------------------
+    sc = s.tocoo()
+    C2 = coo_array((sc.data, (sc.row, sc.col)), shape=sc.shape)
+    assert np.allclose(np.asarray(C2.todense()), s.toarray()), "dist coo ctor"
+
+    # ---- balance() ----------------------------------------------------------
+    skew = sps.random(40, 40, 0.05, random_state=4, format="csr")
+    skew = (skew + sps.csr_matrix((np.ones(30), (np.zeros(30, dtype=int),
+                                                 np.arange(30))), (40, 40))).tocsr()
+    Ask = csr_array(skew)
+    Ask.balance()
+    assert np.allclose(np.asarray(Ask.todense()), skew.toarray()), "balance"
+    xb = rng.random(40)
+    assert np.allclose(np.asarray(Ask @ xb), skew @ xb), "balanced spmv"
+
+    # ---- distributed CG on Poisson ------------------------------------------
+    P = gallery.poisson2d(24)  # 576 unknowns
+    b = darray.ones((P.shape[0],))
+    sol, info = linalg.cg(P, b, tol=1e-10, maxiter=3000, conv_test_iters=20)
+    r = b - P.dot(sol)
+    assert float(r.norm().item()) < 1e-7 * float(b.norm().item()), "dist cg"
+
+    # ---- gmres + eigsh quick ------------------------------------------------
+    sq = (sps.random(30, 30, 0.3, random_state=5) + 30 * sps.eye(30)).tocsr()
+    bq = rng.random(30)
+    xs, info = linalg.gmres(csr_array(sq), bq, tol=1e-10)
+    assert np.allclose(sq @ np.asarray(xs), bq, atol=1e-5), "dist gmres"
+
+    # ---- module fns ---------------------------------------------------------
+    E = sparse.eye(50)
+    ones = np.ones(50)
+    assert np.allclose(np.asarray(E @ ones), ones), "dist eye"
+    K = sparse.kron(A, csr_array(sps.eye(2).tocsr()))
+    assert np.allclose(np.asarray(K.todense()), sps.kron(s, sps.eye(2)).toarray()), "dist kron"
+
+    # ---- darray reductions --------------------------------------------------
+    v = darray.random((101,), seed=9)
+    w = darray.random((101,), seed=10)
+    assert np.isclose(float(v.dot(w).item()), np.asarray(v) @ np.asarray(w)), "dist dot"
+    assert np.isclose(float(v.norm().item()), np.linalg.norm(np.asarray(v))), "dist norm"
+
+    if rank == 0:
+        print("DIST_ALL_OK")
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
