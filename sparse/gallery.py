@@ -94,3 +94,45 @@ def banded(n: int, ndiags: int = 11, dtype=np.float64) -> csr_array:
         vals.append(one * (1.0 if k else float(ndiags)))
         valid.append((c >= 0) & (c < n))
     return _assemble(rows, cols, vals, valid, (n, n), part, dtype)
+
+
+def interpolation2d(nx: int, ny: Optional[int] = None, dtype=np.float64):
+    """Bilinear prolongation P (N_f x N_c) for GMG on a 2-D grid with
+    n_f = 2*n_c + 1 per axis (vertex-centered, Dirichlet).  Built directly
+    per row slab (distributed); R = P.T (Galerkin).
+
+    Capability parity: reference examples/gmg.py linear_operator
+    (gmg.py:303-...) — there built as an explicit sparse matrix too.
+    """
+    ny = nx if ny is None else ny
+    assert nx % 2 == 1 and ny % 2 == 1, "GMG grids need odd dims (2k+1)"
+    nxc, nyc = (nx - 1) // 2, (ny - 1) // 2
+    Nf, Nc = nx * ny, nxc * nyc
+    part = RowPartition.equal(Nf, comm.world_size())
+    r = comm.rank()
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    ix = rows % nx
+    iy = rows // nx
+    # 1-D weight pairs: odd index i=2I+1 -> [(I,1)]; even i=2I -> [(I-1,.5),(I,.5)]
+    def wpairs(i, n_c):
+        odd = (i % 2) == 1
+        I = torch.where(odd, (i - 1) // 2, i // 2)
+        # entries (c0,w0),(c1,w1); invalid marked by c<0 or c>=n_c
+        c0 = torch.where(odd, I, I - 1)
+        w0 = torch.where(odd, torch.ones_like(i, dtype=torch.float64),
+                         torch.full_like(i, 0.5, dtype=torch.float64))
+        c1 = torch.where(odd, torch.full_like(I, -1), I)
+        w1 = torch.full_like(i, 0.5, dtype=torch.float64)
+        v0 = (c0 >= 0) & (c0 < n_c)
+        v1 = (c1 >= 0) & (c1 < n_c) & (~odd)
+        return (c0, w0, v0), (c1, w1, v1)
+    (cx0, wx0, vx0), (cx1, wx1, vx1) = wpairs(ix, nxc)
+    (cy0, wy0, vy0), (cy1, wy1, vy1) = wpairs(iy, nyc)
+    cols_list, vals_list, valid_list = [], [], []
+    # combos in ascending (cy, cx) order: (y0,x0),(y0,x1),(y1,x0),(y1,x1)
+    for (cy, wy, vy) in ((cy0, wy0, vy0), (cy1, wy1, vy1)):
+        for (cx, wx, vx) in ((cx0, wx0, vx0), (cx1, wx1, vx1)):
+            cols_list.append(cy.clamp(min=0) * nxc + cx.clamp(min=0))
+            vals_list.append(wy * wx)
+            valid_list.append(vy & vx)
+    return _assemble(rows, cols_list, vals_list, valid_list, (Nf, Nc), part, dtype)
