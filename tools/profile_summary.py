@@ -23,8 +23,8 @@ def main(path):
     print(f"| kernel | calls | total ms | avg us | % | vgpr | lds B |")
     print(f"|---|---|---|---|---|---|---|")
     for r in rows:
-        name = r[0].split("(")[0].replace("void ", "").replace(
-            "(anonymous namespace)::", "")[:70]
+        name = r[0].replace("void ", "").replace("(anonymous namespace)::", "")
+        name = name.split("(")[0].split("<")[0][:70] or name[:70]
         print(f"| {name} | {r[1]} | {r[2]:.2f} | {r[3]:.1f} | "
               f"{100*r[2]/total:.1f} | {r[4]} | {r[6]} |")
     print(f"\nTotal GPU kernel time: {total:.2f} ms")
