@@ -34,9 +34,12 @@ def _rms_norm(local_sq_sum: torch.Tensor, n: int) -> float:
     return math.sqrt(max(float(t.item()), 0.0) / n)
 
 
+def _sq(t: torch.Tensor) -> torch.Tensor:
+    return torch.sum(torch.abs(t) ** 2) if t.is_complex() else torch.sum(t * t)
+
+
 def _err_norm(e_local: torch.Tensor, scale_local: torch.Tensor, n: int) -> float:
-    s = torch.sum((e_local / scale_local) ** 2)
-    return _rms_norm(s, n)
+    return _rms_norm(_sq(e_local / scale_local), n)
 
 
 def _combine(K: torch.Tensor, coeffs: np.ndarray, h: float) -> torch.Tensor:
@@ -95,12 +98,12 @@ class RungeKutta:
         """scipy's select_initial_step (reference integrate.py:127)."""
         y, f = self.y, self.f
         scale = self._scale_local(y.local)
-        d0 = _rms_norm(torch.sum((y.local / scale) ** 2), self.n)
-        d1 = _rms_norm(torch.sum((f.local / scale) ** 2), self.n)
+        d0 = _rms_norm(_sq(y.local / scale), self.n)
+        d1 = _rms_norm(_sq(f.local / scale), self.n)
         h0 = 1e-6 if d0 < 1e-5 or d1 < 1e-5 else 0.01 * d0 / d1
         y1 = y + f * (h0 * self.direction)
         f1 = asdistarray(self.fun(self.t + h0 * self.direction, y1))
-        d2 = _rms_norm(torch.sum(((f1.local - f.local) / scale) ** 2), self.n) / h0
+        d2 = _rms_norm(_sq((f1.local - f.local) / scale), self.n) / h0
         if d1 <= 1e-15 and d2 <= 1e-15:
             h1 = max(1e-6, h0 * 1e-3)
         else:
@@ -258,8 +261,8 @@ class DOP853(RungeKutta):
 
 def _dop853_err_norm(self_obj, est, scale, n):
     err5, err3, h = est
-    s5 = torch.sum((err5 / scale) ** 2)
-    s3 = torch.sum((err3 / scale) ** 2)
+    s5 = _sq(err5 / scale)
+    s3 = _sq(err3 / scale)
     t = torch.stack([s5, s3])
     comm.all_reduce_(t)
     e5 = float(t[0].item()) / n
