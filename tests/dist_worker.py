@@ -114,9 +114,37 @@ def main():
     assert np.isclose(float(v.dot(w).item()), np.asarray(v) @ np.asarray(w)), "dist dot"
     assert np.isclose(float(v.norm().item()), np.linalg.norm(np.asarray(v))), "dist norm"
 
+    extra_samplesort_check()
+
     if rank == 0:
         print("DIST_ALL_OK")
     dist.destroy_process_group()
+
+
+def extra_samplesort_check():
+    """Appended check: distributed samplesort (keys + payload)."""
+    import torch
+    from sparse.parallel.sort import samplesort
+
+    rank = dist.get_rank()
+    rng2 = np.random.default_rng(100 + rank)
+    k = torch.as_tensor(rng2.integers(0, 1000, 257))
+    v = torch.as_tensor(rng2.random(257))
+    sk, sv = samplesort(k, v)
+    # global concatenation must be globally sorted & a permutation
+    from sparse.parallel import comm as _c
+
+    counts = torch.zeros(dist.get_world_size(), dtype=torch.int64)
+    counts[rank] = sk.numel()
+    _c.all_reduce_(counts)
+    gk = _c.all_gather_rows(sk, [int(c) for c in counts]).numpy()
+    gv = _c.all_gather_rows(sv, [int(c) for c in counts]).numpy()
+    assert np.all(np.diff(gk) >= 0), "samplesort order"
+    # keys follow values
+    allk = _c.all_gather_rows(k, [257] * dist.get_world_size()).numpy()
+    allv = _c.all_gather_rows(v, [257] * dist.get_world_size()).numpy()
+    assert sorted(allk.tolist()) == gk.tolist(), "samplesort permutation"
+    assert np.isclose(gv.sum(), allv.sum()), "payload preserved"
 
 
 if __name__ == "__main__":
