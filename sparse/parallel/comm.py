@@ -12,7 +12,7 @@ RCCL reductions do not support complex).
 """
 from __future__ import annotations
 
-from typing import List, Sequence
+from typing import List, Optional, Sequence
 
 import torch
 import torch.distributed as dist
@@ -78,10 +78,13 @@ def bcast_(t: torch.Tensor, src: int = 0, group=None) -> None:
     dist.broadcast(torch.view_as_real(t) if t.is_complex() else t, src=src, group=group)
 
 
-def all_to_all_v(send: List[torch.Tensor], group=None) -> List[torch.Tensor]:
+def all_to_all_v(send: List[torch.Tensor], group=None,
+                 recv_counts: Optional[List[int]] = None) -> List[torch.Tensor]:
     """Exchange send[r] -> rank r (1-D tensors); returns recv list indexed by
-    source.  Sizes exchanged via equal-size all_gather.  Zero-size guards
-    mirror the reference's NCCL hang workaround (sort.cu:259-263).
+    source.  Sizes exchanged via equal-size all_gather unless the caller
+    already knows recv_counts (ELEMENT counts per source; cached plans pass
+    them to skip the extra collective).  Zero-size guards mirror the
+    reference's NCCL hang workaround (sort.cu:259-263).
     """
     ws = world_size(group)
     me = rank(group)
@@ -93,11 +96,14 @@ def all_to_all_v(send: List[torch.Tensor], group=None) -> List[torch.Tensor]:
     wire = [(_as_real(s.contiguous())) for s in send]
     wdtype = wire[0].dtype
     backend = dist.get_backend(group)
-    cdev = device if backend == "nccl" else torch.device("cpu")
-    counts = torch.tensor([int(s.numel()) for s in wire], dtype=torch.int64, device=cdev)
-    all_counts = [torch.zeros(ws, dtype=torch.int64, device=cdev) for _ in range(ws)]
-    dist.all_gather(all_counts, counts, group=group)
-    recv_counts = [int(all_counts[src][me].item()) for src in range(ws)]
+    if recv_counts is None:
+        cdev = device if backend == "nccl" else torch.device("cpu")
+        counts = torch.tensor([int(s.numel()) for s in wire], dtype=torch.int64, device=cdev)
+        all_counts = [torch.zeros(ws, dtype=torch.int64, device=cdev) for _ in range(ws)]
+        dist.all_gather(all_counts, counts, group=group)
+        recv_counts = [int(all_counts[src][me].item()) for src in range(ws)]
+    elif cplx:
+        recv_counts = [2 * c for c in recv_counts]
     recv = [torch.empty(c, dtype=wdtype, device=device) for c in recv_counts]
     if backend == "nccl":
         dist.all_to_all(recv, wire, group=group)

@@ -71,7 +71,8 @@ class WindowGatherPlan:
         for p in range(ws):
             a, b = self.send_ranges[p]
             send.append(xlocal[a:b].reshape(-1))
-        recv = comm.all_to_all_v(send, group=self.group)
+        rc = [c * k for c in self.recv_counts]
+        recv = comm.all_to_all_v(send, group=self.group, recv_counts=rc)
         out = torch.cat([r for r in recv], dim=0)
         return out.reshape(self.hi - self.lo, *tail) if tail else out
 
@@ -99,13 +100,17 @@ class ReduceScatterPlan:
         # send partial pieces to owners: piece for peer p is the overlap of my
         # window with p's slab — the same recv_counts layout, reversed.
         xpart = self.fwd.xpart
+        tail_k = 1
+        for t in partial.shape[1:]:
+            tail_k *= t
         send = []
         off = 0
         for p in range(ws):
             c = self.fwd.recv_counts[p]
             send.append(partial[off: off + c].reshape(-1))
             off += c
-        recv = comm.all_to_all_v(send, group=self.fwd.group)
+        rc = [(b - a) * tail_k for a, b in self.fwd.send_ranges]
+        recv = comm.all_to_all_v(send, group=self.fwd.group, recv_counts=rc)
         s0 = xpart.start(me)
         tail = ylocal.shape[1:]
         for p in range(ws):
