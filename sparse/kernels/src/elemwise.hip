@@ -112,18 +112,40 @@ __global__ void mult_dense_kernel(const int64_t* __restrict__ indptr,
   out[p] = vals[p] * D[r * n + (int64_t)indices[p]];
 }
 
-// fused y = y ± (a/b) x  (isalpha) | y = x ± (a/b) y ; a,b 0-dim device
+// fused y = y ± (a/b) x  (isalpha) | y = x ± (a/b) y ; a,b 0-dim device.
+// 2-element vector loads/stores (16B for fp64): 8B accesses run at ~0.6x
+// the 16B rate on gfx950.
+template <typename T>
+struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) VPair {
+  T a, b;
+};
+
 template <typename T, bool ISALPHA, bool NEG>
-__global__ void axpby_kernel(T* __restrict__ y, const T* __restrict__ x,
-                             const T* __restrict__ a, const T* __restrict__ b,
-                             int64_t n) {
+__global__ __launch_bounds__(256) void axpby_kernel(
+    T* __restrict__ y, const T* __restrict__ x, const T* __restrict__ a,
+    const T* __restrict__ b, int64_t n) {
   T s = (*a) / (*b);
   if (NEG) s = -s;
+  const int64_t half = n / 2;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    if (ISALPHA) y[i] = y[i] + s * x[i];
-    else y[i] = x[i] + s * y[i];
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  auto* y2 = reinterpret_cast<VPair<T>*>(y);
+  auto* x2 = reinterpret_cast<const VPair<T>*>(x);
+  for (; i < half; i += stride) {
+    VPair<T> yv = y2[i];
+    const VPair<T> xv = x2[i];
+    if (ISALPHA) {
+      yv.a = yv.a + s * xv.a;
+      yv.b = yv.b + s * xv.b;
+    } else {
+      yv.a = xv.a + s * yv.a;
+      yv.b = xv.b + s * yv.b;
+    }
+    y2[i] = yv;
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    int64_t t = n - 1;
+    y[t] = ISALPHA ? (y[t] + s * x[t]) : (x[t] + s * y[t]);
   }
 }
 
@@ -211,7 +233,7 @@ void axpby_hip(at::Tensor y, at::Tensor x, at::Tensor a, at::Tensor b,
                bool isalpha, bool negate) {
   int64_t n = y.numel();
   if (n == 0) return;
-  int64_t blocks = std::min<int64_t>((n + 255) / 256, 65535);
+  int64_t blocks = std::min<int64_t>((n / 2 + 255) / 256 + 1, 32768);
   DISPATCH_VALUES(y.scalar_type(), "axpby", [&] {
     using T = scalar_t;
     auto launch = [&](auto kern) {
@@ -237,11 +259,28 @@ __global__ __launch_bounds__(256) void axpby_norm2_kernel(
   T s = (*a) / (*b);
   if (NEG) s = -s;
   T acc = T(0);
+  const int64_t half = n / 2;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    T v = ISALPHA ? (y[i] + s * x[i]) : (x[i] + s * y[i]);
-    y[i] = v;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  auto* y2 = reinterpret_cast<VPair<T>*>(y);
+  auto* x2 = reinterpret_cast<const VPair<T>*>(x);
+  for (; i < half; i += stride) {
+    VPair<T> yv = y2[i];
+    const VPair<T> xv = x2[i];
+    if (ISALPHA) {
+      yv.a = yv.a + s * xv.a;
+      yv.b = yv.b + s * xv.b;
+    } else {
+      yv.a = xv.a + s * yv.a;
+      yv.b = xv.b + s * yv.b;
+    }
+    y2[i] = yv;
+    acc += yv.a * yv.a + yv.b * yv.b;
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    int64_t t = n - 1;
+    T v = ISALPHA ? (y[t] + s * x[t]) : (x[t] + s * y[t]);
+    y[t] = v;
     acc += v * v;
   }
   red[threadIdx.x] = acc;
