@@ -127,28 +127,75 @@ def elem_mult_csr(A, B, vdt):
 
 
 def spgemm_csr(A, B, a_col_lo, vdt):
+    """Two-phase size-binned Gustavson SpGEMM.
+
+    Rows are binned by product upper bound (sum of touched B-row sizes) into
+    LDS hash sizes 64/256/1024; rows with ub > 512 go through the vectorized
+    expand-sort-reduce path and are scattered into the shared output."""
     from ..ops.local import LocalCSR
 
-    nnz_per_row = torch.empty(A.nrows, dtype=torch.int64, device=A.device)
-    ext().spgemm_nnz(A.indptr, A.indices, B.indptr, B.indices, nnz_per_row,
-                     int(a_col_lo), int(B.ncols))
-    if bool((nnz_per_row < 0).any().item()):
-        # rows denser than the LDS hash: vectorized expand-sort-reduce
-        return _spgemm_esc(A, B, a_col_lo, vdt)
-    indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=A.device)
-    torch.cumsum(nnz_per_row, 0, out=indptr[1:])
+    dev = A.device
+    m = A.nrows
+    if m == 0 or A.nnz == 0:
+        return LocalCSR(torch.zeros(m + 1, dtype=torch.int64, device=dev),
+                        torch.zeros(0, dtype=A.indices.dtype, device=dev),
+                        torch.zeros(0, dtype=vdt, device=dev), m, B.ncols)
+    acounts = A.indptr[1:] - A.indptr[:-1]
+    acols = A.indices.long() - a_col_lo
+    bcounts = B.indptr[1:] - B.indptr[:-1]
+    rows_of_nnz = torch.repeat_interleave(
+        torch.arange(m, dtype=torch.int64, device=dev), acounts)
+    ub = torch.zeros(m, dtype=torch.int64, device=dev)
+    ub.index_add_(0, rows_of_nnz, bcounts[acols])
+    counts = torch.zeros(m, dtype=torch.int64, device=dev)
+    bins = [((ub <= 32), 64), ((ub > 32) & (ub <= 128), 256),
+            ((ub > 128) & (ub <= 512), 1024)]
+    rowlists = [mask.nonzero(as_tuple=False).flatten() for mask, _ in bins]
+    Av = A.values.to(vdt)
+    Bv = B.values.to(vdt)
+    for rl, (_, H) in zip(rowlists, bins):
+        if rl.numel():
+            ext().spgemm_nnz(A.indptr, A.indices, B.indptr, B.indices, rl,
+                             counts, int(a_col_lo), H)
+    esc_rows = (ub > 512).nonzero(as_tuple=False).flatten()
+    esc_sub = None
+    if esc_rows.numel():
+        sc = acounts[esc_rows]
+        sub_ip = torch.zeros(esc_rows.numel() + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(sc, 0, out=sub_ip[1:])
+        tot = int(sub_ip[-1].item())
+        pos = (torch.arange(tot, dtype=torch.int64, device=dev)
+               - torch.repeat_interleave(sub_ip[:-1], sc)
+               + torch.repeat_interleave(A.indptr[esc_rows], sc))
+        subA = LocalCSR(sub_ip, A.indices[pos], Av[pos], esc_rows.numel(), A.ncols)
+        esc_sub = _spgemm_esc(subA, B, a_col_lo, vdt)
+        counts[esc_rows] = esc_sub.indptr[1:] - esc_sub.indptr[:-1]
+    indptr = torch.zeros(m + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=indptr[1:])
     nnz = int(indptr[-1].item())
-    indices = torch.empty(nnz, dtype=A.indices.dtype, device=A.device)
-    values = torch.empty(nnz, dtype=vdt, device=A.device)
-    ext().spgemm_compute(A.indptr, A.indices, A.values.to(vdt),
-                         B.indptr, B.indices, B.values.to(vdt),
-                         indptr, indices, values, int(a_col_lo), int(B.ncols))
+    indices = torch.empty(nnz, dtype=A.indices.dtype, device=dev)
+    values = torch.empty(nnz, dtype=vdt, device=dev)
+    for rl, (_, H) in zip(rowlists, bins):
+        if rl.numel():
+            ext().spgemm_compute(A.indptr, A.indices, Av, B.indptr, B.indices,
+                                 Bv, rl, indptr, indices, values,
+                                 int(a_col_lo), H)
+    if esc_sub is not None and esc_sub.nnz:
+        ec = counts[esc_rows]
+        eoff = torch.zeros(esc_rows.numel(), dtype=torch.int64, device=dev)
+        torch.cumsum(ec[:-1], 0, out=eoff[1:])
+        tot = int(ec.sum().item())
+        dst = (torch.arange(tot, dtype=torch.int64, device=dev)
+               - torch.repeat_interleave(eoff, ec)
+               + torch.repeat_interleave(indptr[esc_rows], ec))
+        indices[dst] = esc_sub.indices
+        values[dst] = esc_sub.values
     # sort columns within each row (hash compaction is unordered)
     rows = torch.repeat_interleave(
-        torch.arange(A.nrows, dtype=torch.int64, device=A.device), nnz_per_row)
+        torch.arange(m, dtype=torch.int64, device=dev), counts)
     key = rows * B.ncols + indices.long()
     order = torch.argsort(key)
-    return LocalCSR(indptr, indices[order], values[order], A.nrows, B.ncols)
+    return LocalCSR(indptr, indices[order], values[order], m, B.ncols)
 
 
 def _spgemm_esc(A, B, a_col_lo, vdt):

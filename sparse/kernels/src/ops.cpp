@@ -44,10 +44,10 @@ void tropical_spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t);
 void rk_calc_dy_hip(at::Tensor, at::Tensor, double, at::Tensor);
 void cdist_hip(at::Tensor, at::Tensor, at::Tensor);
 void spgemm_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                    int64_t, int64_t);
+                    at::Tensor, int64_t, int64_t);
 void spgemm_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                         at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                        at::Tensor, int64_t, int64_t);
+                        at::Tensor, at::Tensor, int64_t, int64_t);
 
 TORCH_LIBRARY(sparse_hip, m) {
   m.def("spmv(Tensor indptr, Tensor indices, Tensor values, Tensor x, "
@@ -94,10 +94,10 @@ TORCH_LIBRARY(sparse_hip, m) {
   m.def("rk_calc_dy(Tensor K, Tensor avec, float h, Tensor(a!) dy) -> ()");
   m.def("cdist(Tensor XA, Tensor XB, Tensor(a!) out) -> ()");
   m.def("spgemm_nnz(Tensor aip, Tensor aix, Tensor bip, Tensor bix, "
-        "Tensor(a!) nnz_out, int a_col_lo, int bncols) -> ()");
+        "Tensor rowlist, Tensor(a!) nnz_out, int a_col_lo, int hash_size) -> ()");
   m.def("spgemm_compute(Tensor aip, Tensor aix, Tensor av, Tensor bip, "
-        "Tensor bix, Tensor bv, Tensor cip, Tensor(a!) cix, Tensor(b!) cv, "
-        "int a_col_lo, int bncols) -> ()");
+        "Tensor bix, Tensor bv, Tensor rowlist, Tensor cip, Tensor(a!) cix, "
+        "Tensor(b!) cv, int a_col_lo, int hash_size) -> ()");
 }
 
 TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {

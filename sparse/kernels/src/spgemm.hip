@@ -1,50 +1,48 @@
-// SpGEMM: two-phase Gustavson with per-wave LDS hash accumulators.
+// SpGEMM: two-phase Gustavson with per-wave LDS hash accumulators,
+// size-binned.
 //
 // Reference parity: SPGEMM_CSR_CSR_CSR_GPU (spgemm_csr_csr_csr.cu:33-272,
 // cuSPARSE-backed there) and the CPU Gustavson semantics
 // (spgemm_csr_csr_csr.cc:27-85).  MI355X design: one wave per C row, an LDS
-// open-addressing hash of HASH entries per wave (keys + fp accumulators),
-// no global scratch; rows whose distinct-column count exceeds the table are
-// flagged (-1) and the Python wrapper computes them with a vectorized
-// expand-sort-reduce fallback.  Output columns are compacted unsorted and
-// sorted per-row by the wrapper.
+// open-addressing hash (keys + fp accumulators) per wave.  Rows are BINNED
+// by their product upper bound (sum of B-row sizes) into hash sizes
+// 64/256/1024 so short stencil rows don't pay a 1024-entry table init;
+// rows with ub > 512 fall to the wrapper's vectorized expand-sort-reduce.
+// Output columns are compacted unsorted and sorted per-row by the wrapper.
 #include "common.h"
 
 namespace {
 
-constexpr int HASH = 1024;          // entries per wave (power of 2)
 constexpr int WAVES_PER_BLOCK = 4;  // 256 threads
-constexpr int CAP = HASH - 64;      // bail threshold
 
-__device__ __forceinline__ uint32_t hash_col(int64_t c) {
-  return (uint32_t)(((uint64_t)c * 2654435761u) & (HASH - 1));
+__device__ __forceinline__ uint32_t hash_mul(int64_t c) {
+  return (uint32_t)((uint64_t)c * 2654435761u);
 }
 
-template <typename index_t>
+template <typename index_t, int HASH>
 __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
     const int64_t* __restrict__ aip, const index_t* __restrict__ aix,
     const int64_t* __restrict__ bip, const index_t* __restrict__ bix,
-    int64_t* __restrict__ nnz_out, int64_t m, int64_t a_col_lo) {
+    const int64_t* __restrict__ rowlist, int64_t nrows,
+    int64_t* __restrict__ nnz_out, int64_t a_col_lo) {
   __shared__ int64_t keys[WAVES_PER_BLOCK][HASH];
   __shared__ int counts[WAVES_PER_BLOCK];
-  __shared__ int overflow[WAVES_PER_BLOCK];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const int64_t r = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
-  if (r >= m) return;
+  const int64_t li = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (li >= nrows) return;
+  const int64_t r = rowlist[li];
   for (int i = lane; i < HASH; i += WAVE) keys[wid][i] = -1;
-  if (lane == 0) { counts[wid] = 0; overflow[wid] = 0; }
+  if (lane == 0) counts[wid] = 0;
   __builtin_amdgcn_wave_barrier();
   __threadfence_block();
   const int64_t as = aip[r], ae = aip[r + 1];
   for (int64_t p = as + lane; p < ae; p += WAVE) {
-    if (overflow[wid]) break;
     const int64_t brow = (int64_t)aix[p] - a_col_lo;
     const int64_t bs = bip[brow], be = bip[brow + 1];
     for (int64_t q = bs; q < be; ++q) {
       int64_t c = (int64_t)bix[q];
-      uint32_t h = hash_col(c);
-      int probes = 0;
+      uint32_t h = hash_mul(c) & (HASH - 1);
       while (true) {
         int64_t old = atomicCAS((unsigned long long*)&keys[wid][h],
                                 (unsigned long long)(-1ll),
@@ -52,33 +50,31 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
         if (old == -1ll) { atomicAdd(&counts[wid], 1); break; }
         if (old == c) break;
         h = (h + 1) & (HASH - 1);
-        if (++probes > HASH || counts[wid] >= CAP) { overflow[wid] = 1; break; }
       }
-      if (overflow[wid]) break;
     }
   }
   __builtin_amdgcn_wave_barrier();
   __threadfence_block();
-  if (lane == 0) nnz_out[r] = overflow[wid] ? -1 : (int64_t)counts[wid];
+  if (lane == 0) nnz_out[r] = (int64_t)counts[wid];
 }
 
-template <typename T, typename index_t>
+template <typename T, typename index_t, int HASH>
 __global__ __launch_bounds__(256) void spgemm_compute_kernel(
     const int64_t* __restrict__ aip, const index_t* __restrict__ aix,
     const T* __restrict__ av, const int64_t* __restrict__ bip,
     const index_t* __restrict__ bix, const T* __restrict__ bv,
+    const int64_t* __restrict__ rowlist, int64_t nrows,
     const int64_t* __restrict__ cip, index_t* __restrict__ cix,
-    T* __restrict__ cv, int64_t m, int64_t a_col_lo) {
+    T* __restrict__ cv, int64_t a_col_lo) {
   __shared__ int64_t keys[WAVES_PER_BLOCK][HASH];
   __shared__ __align__(16) char accs_raw[WAVES_PER_BLOCK * HASH * sizeof(T)];
   auto accs = reinterpret_cast<T(*)[HASH]>(accs_raw);
   __shared__ int slots[WAVES_PER_BLOCK];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const int64_t r = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
-  if (r >= m) return;
-  // flagged rows (nnz unknown) are produced by the fallback path instead
-  if (cip[r + 1] - cip[r] < 0) return;
+  const int64_t li = (int64_t)blockIdx.x * WAVES_PER_BLOCK + wid;
+  if (li >= nrows) return;
+  const int64_t r = rowlist[li];
   for (int i = lane; i < HASH; i += WAVE) {
     keys[wid][i] = -1;
     accs[wid][i] = ZeroOf<T>::value();
@@ -93,7 +89,7 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
     const int64_t bs = bip[brow], be = bip[brow + 1];
     for (int64_t q = bs; q < be; ++q) {
       int64_t c = (int64_t)bix[q];
-      uint32_t h = hash_col(c);
+      uint32_t h = hash_mul(c) & (HASH - 1);
       while (true) {
         int64_t old = atomicCAS((unsigned long long*)&keys[wid][h],
                                 (unsigned long long)(-1ll),
@@ -108,7 +104,6 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
   }
   __builtin_amdgcn_wave_barrier();
   __threadfence_block();
-  // compact (unsorted; wrapper sorts per row)
   const int64_t base = cip[r];
   for (int i = lane; i < HASH; i += WAVE) {
     if (keys[wid][i] != -1) {
@@ -122,36 +117,47 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
 }  // namespace
 
 void spgemm_nnz_hip(at::Tensor aip, at::Tensor aix, at::Tensor bip,
-                    at::Tensor bix, at::Tensor nnz_out, int64_t a_col_lo,
-                    int64_t bncols) {
-  int64_t m = nnz_out.numel();
-  if (m == 0) return;
+                    at::Tensor bix, at::Tensor rowlist, at::Tensor nnz_out,
+                    int64_t a_col_lo, int64_t hash_size) {
+  int64_t nrows = rowlist.numel();
+  if (nrows == 0) return;
+  int64_t nb = (nrows + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
   DISPATCH_INDEX(aix.scalar_type(), "spgemm_nnz", [&] {
-    hipLaunchKernelGGL((spgemm_nnz_kernel<index_t>),
-                       dim3((m + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK),
-                       dim3(256), 0, cur_stream(), aip.data_ptr<int64_t>(),
-                       aix.data_ptr<index_t>(), bip.data_ptr<int64_t>(),
-                       bix.data_ptr<index_t>(), nnz_out.data_ptr<int64_t>(), m,
-                       a_col_lo);
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(nb), dim3(256), 0, cur_stream(),
+                         aip.data_ptr<int64_t>(), aix.data_ptr<index_t>(),
+                         bip.data_ptr<int64_t>(), bix.data_ptr<index_t>(),
+                         rowlist.data_ptr<int64_t>(), nrows,
+                         nnz_out.data_ptr<int64_t>(), a_col_lo);
+    };
+    if (hash_size <= 64) launch(spgemm_nnz_kernel<index_t, 64>);
+    else if (hash_size <= 256) launch(spgemm_nnz_kernel<index_t, 256>);
+    else launch(spgemm_nnz_kernel<index_t, 1024>);
   });
 }
 
 void spgemm_compute_hip(at::Tensor aip, at::Tensor aix, at::Tensor av,
                         at::Tensor bip, at::Tensor bix, at::Tensor bv,
-                        at::Tensor cip, at::Tensor cix, at::Tensor cv,
-                        int64_t a_col_lo, int64_t bncols) {
-  int64_t m = aip.numel() - 1;
-  if (m == 0) return;
+                        at::Tensor rowlist, at::Tensor cip, at::Tensor cix,
+                        at::Tensor cv, int64_t a_col_lo, int64_t hash_size) {
+  int64_t nrows = rowlist.numel();
+  if (nrows == 0) return;
+  int64_t nb = (nrows + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
   DISPATCH_VALUES(cv.scalar_type(), "spgemm_compute", [&] {
     using T = scalar_t;
     DISPATCH_INDEX(aix.scalar_type(), "spgemm_compute_idx", [&] {
-      hipLaunchKernelGGL((spgemm_compute_kernel<T, index_t>),
-                         dim3((m + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK),
-                         dim3(256), 0, cur_stream(), aip.data_ptr<int64_t>(),
-                         aix.data_ptr<index_t>(), av.data_ptr<T>(),
-                         bip.data_ptr<int64_t>(), bix.data_ptr<index_t>(),
-                         bv.data_ptr<T>(), cip.data_ptr<int64_t>(),
-                         cix.data_ptr<index_t>(), cv.data_ptr<T>(), m, a_col_lo);
+      auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(nb), dim3(256), 0, cur_stream(),
+                           aip.data_ptr<int64_t>(), aix.data_ptr<index_t>(),
+                           av.data_ptr<T>(), bip.data_ptr<int64_t>(),
+                           bix.data_ptr<index_t>(), bv.data_ptr<T>(),
+                           rowlist.data_ptr<int64_t>(), nrows,
+                           cip.data_ptr<int64_t>(), cix.data_ptr<index_t>(),
+                           cv.data_ptr<T>(), a_col_lo);
+      };
+      if (hash_size <= 64) launch(spgemm_compute_kernel<T, index_t, 64>);
+      else if (hash_size <= 256) launch(spgemm_compute_kernel<T, index_t, 256>);
+      else launch(spgemm_compute_kernel<T, index_t, 1024>);
     });
   });
 }
