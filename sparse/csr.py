@@ -366,15 +366,16 @@ class csr_array(CompressedBase, DenseSparseBase):
             raise ValueError(f"dimension mismatch {self.shape} @ {x.shape}")
         plan = self._xplan(x.partition)
         vdt = self._out_dtype(x.local.dtype)
-        xw = plan.gather(x.local.to(vdt))
         ell = self._ell() if self._values.dtype == vdt else None
         if ell is not None:
             from . import kernels
 
+            pieces = plan.gather_halos(x.local.to(vdt))
             ylocal = torch.empty(self.partition.count(comm.rank()), dtype=vdt,
                                  device=self._values.device)
-            kernels.ell_spmv(ell, xw, ylocal, plan.lo)
+            kernels.ell_spmv(ell, pieces, ylocal, plan.lo)
         else:
+            xw = plan.gather(x.local.to(vdt))
             lc = self.local
             if lc.values.dtype != vdt:
                 lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt), lc.nrows, lc.ncols)
@@ -406,11 +407,12 @@ class csr_array(CompressedBase, DenseSparseBase):
         from . import kernels
 
         plan = self._xplan(p.partition)
-        xw = plan.gather(p.local)
         ell = self._ell()
         if ell is not None:
-            dot = kernels.ell_spmv_dot(ell, xw, q.local, p.local, plan.lo)
+            pieces = plan.gather_halos(p.local)
+            dot = kernels.ell_spmv_dot(ell, pieces, q.local, p.local, plan.lo)
         else:
+            xw = plan.gather(p.local)
             dot = torch.zeros((), dtype=self._values.dtype, device=self._values.device)
             lc = self.local
             kernels.spmv_dot(lc, xw, q.local, p.local, dot, plan.lo)

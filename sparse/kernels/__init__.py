@@ -284,14 +284,18 @@ def build_ell(A):
     return EllMirror(eidx, evals, W, m)
 
 
-def ell_spmv(ell: EllMirror, x, y, col_lo: int):
-    ext().ell_spmv(ell.eidx, ell.evals, x, y, ell.W, ell.m, int(col_lo))
+def ell_spmv(ell: EllMirror, pieces, y, col_lo: int):
+    hlo, own, hhi = pieces
+    ext().ell_spmv(ell.eidx, ell.evals, hlo.contiguous(), own.contiguous(),
+                   hhi.contiguous(), y, ell.W, ell.m, int(col_lo))
 
 
-def ell_spmv_dot(ell: EllMirror, x, y, p, col_lo: int):
+def ell_spmv_dot(ell: EllMirror, pieces, y, p, col_lo: int):
+    hlo, own, hhi = pieces
     mp = ell.evals.numel() // ell.W
     nblocks = (mp // 2 + 255) // 256
     partial = torch.empty(nblocks, dtype=ell.evals.dtype, device=ell.evals.device)
-    ext().ell_spmv_dot(ell.eidx, ell.evals, x, y, p, partial, ell.W, ell.m,
+    ext().ell_spmv_dot(ell.eidx, ell.evals, hlo.contiguous(), own.contiguous(),
+                       hhi.contiguous(), y, p, partial, ell.W, ell.m,
                        int(col_lo))
     return partial.sum()

@@ -16,9 +16,10 @@ void axpby_norm2_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool,
 void build_ell_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                    int64_t, int64_t);
 void ell_spmv_plain_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                        int64_t, int64_t, int64_t);
+                        at::Tensor, at::Tensor, int64_t, int64_t, int64_t);
 void ell_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                      at::Tensor, at::Tensor, int64_t, int64_t, int64_t);
+                      at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      int64_t, int64_t, int64_t);
 void add_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void add_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                      at::Tensor, at::Tensor, at::Tensor, at::Tensor, double,
@@ -71,10 +72,11 @@ TORCH_LIBRARY(sparse_hip, m) {
         "bool negate, Tensor(b!) dot_out) -> ()");
   m.def("build_ell(Tensor indptr, Tensor indices, Tensor values, "
         "Tensor(a!) eidx, Tensor(b!) evals, int W, int pad_idx) -> ()");
-  m.def("ell_spmv(Tensor eidx, Tensor evals, Tensor x, Tensor(a!) y, int W, "
-        "int m, int col_lo) -> ()");
-  m.def("ell_spmv_dot(Tensor eidx, Tensor evals, Tensor x, Tensor(a!) y, "
-        "Tensor pvec, Tensor(b!) dot_partial, int W, int m, int col_lo) -> ()");
+  m.def("ell_spmv(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
+        "Tensor hhi, Tensor(a!) y, int W, int m, int col_lo) -> ()");
+  m.def("ell_spmv_dot(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
+        "Tensor hhi, Tensor(a!) y, Tensor pvec, Tensor(b!) dot_partial, "
+        "int W, int m, int col_lo) -> ()");
   m.def("spmm(Tensor indptr, Tensor indices, Tensor vals, Tensor B, "
         "Tensor(a!) C, int col_lo) -> ()");
   m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "

@@ -295,12 +295,25 @@ __global__ void build_ell_kernel(const int64_t* __restrict__ indptr,
   }
 }
 
+// window-relative x lookup over (halo_lo | own slab | halo_hi) — the own
+// piece is the rank's x slab used IN PLACE (no per-SpMV self-copy).
+template <typename T>
+__device__ __forceinline__ T xpiece(int64_t idx, const T* __restrict__ hlo,
+                                    int64_t nlo, const T* __restrict__ own,
+                                    int64_t nown, const T* __restrict__ hhi) {
+  if (idx < nlo) return hlo[idx];
+  idx -= nlo;
+  if (idx < nown) return own[idx];
+  return hhi[idx - nown];
+}
+
 template <typename T, typename index_t, bool FUSE_DOT>
 __global__ __launch_bounds__(BLK) void ell_spmv_kernel(
     const index_t* __restrict__ eidx, const T* __restrict__ evals,
-    const T* __restrict__ x, T* __restrict__ y,
+    const T* __restrict__ hlo, const T* __restrict__ own,
+    const T* __restrict__ hhi, T* __restrict__ y,
     const T* __restrict__ pvec, T* __restrict__ dot_partial,
-    int64_t m, int64_t mp, int W, int64_t col_lo) {
+    int64_t m, int64_t mp, int W, int64_t col_lo, int64_t nlo, int64_t nown) {
   __shared__ __align__(16) char red_raw[BLK * sizeof(T)];
   T* red = reinterpret_cast<T*>(red_raw);
   const int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
@@ -314,8 +327,8 @@ __global__ __launch_bounds__(BLK) void ell_spmv_kernel(
       struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
       const IP ii = *reinterpret_cast<const IP*>(&eidx[base]);
       const TP vv = *reinterpret_cast<const TP*>(&evals[base]);
-      a0 += vv.a * x[(int64_t)ii.a - col_lo];
-      a1 += vv.b * x[(int64_t)ii.b - col_lo];
+      a0 += vv.a * xpiece((int64_t)ii.a - col_lo, hlo, nlo, own, nown, hhi);
+      a1 += vv.b * xpiece((int64_t)ii.b - col_lo, hlo, nlo, own, nown, hhi);
     }
     if (r0 + 1 < m) {
       struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
@@ -358,39 +371,49 @@ void build_ell_hip(at::Tensor indptr, at::Tensor indices, at::Tensor values,
   });
 }
 
-void ell_spmv_hip(at::Tensor eidx, at::Tensor evals, at::Tensor x, at::Tensor y,
+void ell_spmv_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
+                  at::Tensor own, at::Tensor hhi, at::Tensor y,
                   int64_t W, int64_t m, int64_t col_lo,
                   const c10::optional<at::Tensor>& pvec,
                   const c10::optional<at::Tensor>& dot_partial) {
   const int64_t mp = evals.numel() / W;
   const bool fuse = pvec.has_value();
   const int64_t nblocks = (mp / 2 + BLK - 1) / BLK;
+  const int64_t nlo = hlo.numel();
+  const int64_t nown = own.numel();
   DISPATCH_VALUES(evals.scalar_type(), "ell_spmv", [&] {
     using T = scalar_t;
     DISPATCH_INDEX(eidx.scalar_type(), "ell_spmv_idx", [&] {
+      const T* hlo_p = nlo ? hlo.data_ptr<T>() : own.data_ptr<T>();
+      const T* hhi_p = hhi.numel() ? hhi.data_ptr<T>() : own.data_ptr<T>();
       if (fuse) {
         hipLaunchKernelGGL((ell_spmv_kernel<T, index_t, true>), dim3(nblocks),
                            dim3(BLK), 0, cur_stream(), eidx.data_ptr<index_t>(),
-                           evals.data_ptr<T>(), x.data_ptr<T>(), y.data_ptr<T>(),
+                           evals.data_ptr<T>(), hlo_p, own.data_ptr<T>(), hhi_p,
+                           y.data_ptr<T>(),
                            pvec->data_ptr<T>(), dot_partial->data_ptr<T>(),
-                           m, mp, (int)W, col_lo);
+                           m, mp, (int)W, col_lo, nlo, nown);
       } else {
         hipLaunchKernelGGL((ell_spmv_kernel<T, index_t, false>), dim3(nblocks),
                            dim3(BLK), 0, cur_stream(), eidx.data_ptr<index_t>(),
-                           evals.data_ptr<T>(), x.data_ptr<T>(), y.data_ptr<T>(),
-                           nullptr, nullptr, m, mp, (int)W, col_lo);
+                           evals.data_ptr<T>(), hlo_p, own.data_ptr<T>(), hhi_p,
+                           y.data_ptr<T>(),
+                           nullptr, nullptr, m, mp, (int)W, col_lo, nlo, nown);
       }
     });
   });
 }
 
-void ell_spmv_plain_hip(at::Tensor eidx, at::Tensor evals, at::Tensor x,
-                        at::Tensor y, int64_t W, int64_t m, int64_t col_lo) {
-  ell_spmv_hip(eidx, evals, x, y, W, m, col_lo, c10::nullopt, c10::nullopt);
+void ell_spmv_plain_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
+                        at::Tensor own, at::Tensor hhi, at::Tensor y,
+                        int64_t W, int64_t m, int64_t col_lo) {
+  ell_spmv_hip(eidx, evals, hlo, own, hhi, y, W, m, col_lo, c10::nullopt,
+               c10::nullopt);
 }
 
-void ell_spmv_dot_hip(at::Tensor eidx, at::Tensor evals, at::Tensor x,
-                      at::Tensor y, at::Tensor pvec, at::Tensor dot_partial,
+void ell_spmv_dot_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
+                      at::Tensor own, at::Tensor hhi, at::Tensor y,
+                      at::Tensor pvec, at::Tensor dot_partial,
                       int64_t W, int64_t m, int64_t col_lo) {
-  ell_spmv_hip(eidx, evals, x, y, W, m, col_lo, pvec, dot_partial);
+  ell_spmv_hip(eidx, evals, hlo, own, hhi, y, W, m, col_lo, pvec, dot_partial);
 }

@@ -55,6 +55,30 @@ class WindowGatherPlan:
             self.recv_counts.append(max(0, b - a))
         assert sum(self.recv_counts) == self.hi - self.lo
 
+    def gather_halos(self, xlocal: torch.Tensor):
+        """Exchange ONLY the halo pieces; my own slab portion is used in
+        place (no self-copy through the collective).  Returns
+        (halo_lo, own_view, halo_hi)."""
+        ws = comm.world_size(self.group)
+        me = comm.rank(self.group)
+        xs, xe = self.xpart.start(me), self.xpart.stop(me)
+        own_a, own_b = max(self.lo, xs), min(self.hi, xe)
+        if ws == 1:
+            return xlocal[:0], xlocal[self.lo: self.hi], xlocal[:0]
+        send = []
+        for p in range(ws):
+            if p == me:
+                send.append(xlocal[:0])
+                continue
+            a, b = self.send_ranges[p]
+            send.append(xlocal[a:b].reshape(-1))
+        rc = [0 if p == me else self.recv_counts[p] for p in range(ws)]
+        recv = comm.all_to_all_v(send, group=self.group, recv_counts=rc)
+        halo_lo = torch.cat([recv[p] for p in range(me)]) if me > 0 else xlocal[:0]
+        halo_hi = torch.cat([recv[p] for p in range(me + 1, ws)]) if me < ws - 1 else xlocal[:0]
+        own = xlocal[own_a - xs: own_b - xs]
+        return halo_lo, own, halo_hi
+
     def gather(self, xlocal: torch.Tensor) -> torch.Tensor:
         """Return the window x[lo:hi) (dim 0 slices; works for 1-D and 2-D)."""
         ws = comm.world_size(self.group)

@@ -115,10 +115,31 @@ def main():
     assert np.isclose(float(v.norm().item()), np.linalg.norm(np.asarray(v))), "dist norm"
 
     extra_samplesort_check()
+    extra_halo_check()
 
     if rank == 0:
         print("DIST_ALL_OK")
     dist.destroy_process_group()
+
+
+def extra_halo_check():
+    """gather_halos pieces must reassemble to the gathered window."""
+    import torch
+    from sparse.parallel.gather import WindowGatherPlan
+    from sparse.parallel.partition import RowPartition
+
+    ws = dist.get_world_size()
+    rank = dist.get_rank()
+    n = 101
+    part = RowPartition.equal(n, ws)
+    x = torch.arange(part.start(rank), part.stop(rank), dtype=torch.float64)
+    for lo, hi in [(0, n), (3, n - 5), (part.start(rank), part.stop(rank)),
+                   (max(0, part.start(rank) - 7), min(n, part.stop(rank) + 9))]:
+        plan = WindowGatherPlan(lo, hi, part)
+        full = plan.gather(x)
+        hlo, own, hhi = plan.gather_halos(x)
+        re = torch.cat([hlo, own, hhi])
+        assert torch.equal(re, full), (rank, lo, hi)
 
 
 def extra_samplesort_check():

@@ -279,3 +279,31 @@ def test_axpby_norm2_gpu():
     expect = y0 - 0.5 * x0
     assert np.allclose(np.asarray(y), expect, rtol=1e-12)
     assert np.isclose(float(rz.item()), float(expect @ expect), rtol=1e-10)
+
+
+def test_ell_spmv_three_piece_split():
+    """Exercise the halo (3-piece x) addressing path of the ELL kernel with
+    an artificial split — the multi-GPU layout with ws=1 data."""
+    from sparse import csr_array, darray, kernels
+
+    s = sample_csr(4000, 4000, 0.003, seed=40)
+    A = csr_array(s)
+    x = darray.random((4000,), seed=41)
+    ell = A._ell()
+    assert ell is not None
+    plan = A._xplan(x.partition)
+    xw = plan.gather(x.local)
+    ref = torch.empty(4000, dtype=torch.float64, device="cuda")
+    kernels.ell_spmv(ell, (xw[:0], xw, xw[:0]), ref, plan.lo)
+    n = xw.numel()
+    for cut1, cut2 in [(0, n), (100, n - 137), (1, 2), (n // 2, n // 2)]:
+        out = torch.empty(4000, dtype=torch.float64, device="cuda")
+        pieces = (xw[:cut1].clone(), xw[cut1:cut2].clone(), xw[cut2:].clone())
+        kernels.ell_spmv(ell, pieces, out, plan.lo)
+        assert torch.allclose(out, ref), (cut1, cut2)
+    # fused dot variant
+    q = torch.empty(4000, dtype=torch.float64, device="cuda")
+    d = kernels.ell_spmv_dot(ell, (xw[:50].clone(), xw[50:n - 60].clone(),
+                                   xw[n - 60:].clone()), q, x.local, plan.lo)
+    assert torch.allclose(q, ref)
+    assert np.isclose(float(d.item()), float(torch.dot(x.local, ref).item()))
