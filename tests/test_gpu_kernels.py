@@ -284,10 +284,9 @@ def test_axpby_norm2_gpu():
 def test_ell_spmv_three_piece_split():
     """Exercise the halo (3-piece x) addressing path of the ELL kernel with
     an artificial split — the multi-GPU layout with ws=1 data."""
-    from sparse import csr_array, darray, kernels
+    from sparse import darray, gallery, kernels
 
-    s = sample_csr(4000, 4000, 0.003, seed=40)
-    A = csr_array(s)
+    A = gallery.banded(4000, ndiags=7)
     x = darray.random((4000,), seed=41)
     ell = A._ell()
     assert ell is not None
