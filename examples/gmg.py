@@ -25,6 +25,7 @@ from benchmark import parse_common_args
 
 parser = argparse.ArgumentParser()
 parser.add_argument("-N", type=int, default=255, help="grid edge (odd, 2^k-1)")
+parser.add_argument("-dim", type=int, default=2, choices=[2, 3])
 parser.add_argument("-levels", type=int, default=None)
 parser.add_argument("-maxiter", type=int, default=200)
 parser.add_argument("-tol", type=float, default=1e-8)
@@ -59,18 +60,20 @@ class Level:
 
 
 class GMG:
-    """V-cycle preconditioner."""
+    """V-cycle preconditioner (2-D bilinear or 3-D trilinear transfers)."""
 
-    def __init__(self, A, nx, levels=None, smooth_iters=2, coarse_threshold=1024):
+    def __init__(self, A, nx, levels=None, smooth_iters=2, coarse_threshold=1024,
+                 dim=2):
         self.levels = []
         cur_nx = nx
         cur = A
         maxl = levels or 64
+        interp = gallery.interpolation2d if dim == 2 else gallery.interpolation3d
         while True:
             self.levels.append(Level(cur))
             if len(self.levels) >= maxl or cur.shape[0] <= coarse_threshold or cur_nx < 7:
                 break
-            P = gallery.interpolation2d(cur_nx)
+            P = interp(cur_nx)
             Ac = (P.T @ cur) @ P  # Galerkin triple product (distributed SpGEMM)
             self.levels[-1].Pdown = P
             cur = Ac
@@ -120,17 +123,21 @@ def main():
     N = args.N
     assert N % 2 == 1, "N must be odd (2^k - 1)"
     h = 1.0 / (N + 1)
-    A = gallery.poisson2d(N, scale=1.0 / (h * h))
+    if args.dim == 2:
+        A = gallery.poisson2d(N, scale=1.0 / (h * h))
+    else:
+        A = gallery.poisson3d(N, scale=1.0 / (h * h))
     n = A.shape[0]
     ii = darray.arange(n).astype(np.float64)
     xl = (ii.local % N).to(torch.float64) * h
-    yl = (ii.local // N).to(torch.float64) * h
+    yl = ((ii.local // N) % N).to(torch.float64) * h
     b = darray.DistArray.from_local(
         torch.sin(math.pi * xl) * torch.sin(math.pi * yl), ii.partition, (n,))
 
     t0 = timer
     t0.start()
-    mg = GMG(A, N, levels=args.levels, smooth_iters=args.smooth_iters)
+    mg = GMG(A, N, levels=args.levels, smooth_iters=args.smooth_iters,
+             dim=args.dim)
     setup_ms = t0.stop()
 
     it_count = [0]

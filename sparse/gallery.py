@@ -136,3 +136,46 @@ def interpolation2d(nx: int, ny: Optional[int] = None, dtype=np.float64):
             vals_list.append(wy * wx)
             valid_list.append(vy & vx)
     return _assemble(rows, cols_list, vals_list, valid_list, (Nf, Nc), part, dtype)
+
+
+def interpolation3d(nx: int, ny: Optional[int] = None, nz: Optional[int] = None,
+                    dtype=np.float64):
+    """Trilinear prolongation P (N_f x N_c) for 3-D GMG (n_f = 2*n_c + 1 per
+    axis, vertex-centered Dirichlet) — the 512^3 7-pt benchmark operator's
+    grid transfer.  Built per row slab like interpolation2d."""
+    ny = nx if ny is None else ny
+    nz = nx if nz is None else nz
+    assert nx % 2 == 1 and ny % 2 == 1 and nz % 2 == 1
+    nxc, nyc, nzc = (nx - 1) // 2, (ny - 1) // 2, (nz - 1) // 2
+    Nf, Nc = nx * ny * nz, nxc * nyc * nzc
+    part = RowPartition.equal(Nf, comm.world_size())
+    r = comm.rank()
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    ix = rows % nx
+    iy = (rows // nx) % ny
+    iz = rows // (nx * ny)
+
+    def wpairs(i, n_c):
+        odd = (i % 2) == 1
+        I = torch.where(odd, (i - 1) // 2, i // 2)
+        c0 = torch.where(odd, I, I - 1)
+        w0 = torch.where(odd, torch.ones_like(i, dtype=torch.float64),
+                         torch.full_like(i, 0.5, dtype=torch.float64))
+        c1 = torch.where(odd, torch.full_like(I, -1), I)
+        w1 = torch.full_like(i, 0.5, dtype=torch.float64)
+        v0 = (c0 >= 0) & (c0 < n_c)
+        v1 = (c1 >= 0) & (c1 < n_c) & (~odd)
+        return (c0, w0, v0), (c1, w1, v1)
+
+    px = wpairs(ix, nxc)
+    py = wpairs(iy, nyc)
+    pz = wpairs(iz, nzc)
+    cols_list, vals_list, valid_list = [], [], []
+    for (cz, wz, vz) in pz:
+        for (cy, wy, vy) in py:
+            for (cx, wx, vx) in px:
+                cols_list.append((cz.clamp(min=0) * nyc + cy.clamp(min=0)) * nxc
+                                 + cx.clamp(min=0))
+                vals_list.append(wz * wy * wx)
+                valid_list.append(vz & vy & vx)
+    return _assemble(rows, cols_list, vals_list, valid_list, (Nf, Nc), part, dtype)

@@ -53,3 +53,10 @@ def test_spgemm_microbenchmark():
 def test_spectral_norm():
     out = run("spectral_norm.py")
     assert "OK" in out
+
+
+def test_gmg_3d():
+    out = run("gmg.py", "-N", "31", "-dim", "3", "-maxiter", "60")
+    assert "info=0" in out
+    iters = int(out.split("iters=")[1].split()[0])
+    assert iters <= 15, out
