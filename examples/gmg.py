@@ -74,8 +74,10 @@ class GMG:
             if len(self.levels) >= maxl or cur.shape[0] <= coarse_threshold or cur_nx < 7:
                 break
             P = interp(cur_nx)
-            Ac = (P.T @ cur) @ P  # Galerkin triple product (distributed SpGEMM)
+            R = P.T
+            Ac = (R @ cur) @ P  # Galerkin triple product (distributed SpGEMM)
             self.levels[-1].Pdown = P
+            self.levels[-1].Rdown = R  # cached: keeps gather plans warm
             cur = Ac
             cur_nx = (cur_nx - 1) // 2
         # replicated coarse solve (machine-scoping equivalent)
@@ -84,10 +86,13 @@ class GMG:
         self.smooth_iters = smooth_iters
 
     def _smooth(self, lvl, x, b, iters):
-        # weighted Jacobi: x += omega * dinv * (b - A x)
+        # weighted Jacobi: x += omega * dinv * (b - A x), fused in-place
+        import torch as _t
+
         for _ in range(iters):
-            r = b - lvl.A.dot(x)
-            x += r * lvl.dinv * lvl.omega
+            r = lvl.A.dot(x)
+            r.local.sub_(b.local).neg_()  # r = b - A x, in place
+            x.local.addcmul_(r.local, lvl.dinv.local, value=lvl.omega)
         return x
 
     def _vcycle(self, li, b):
@@ -96,11 +101,15 @@ class GMG:
             bg = np.asarray(b)
             xg = self.coarse_inv @ bg
             return darray.asdistarray(xg)
-        x = b * lvl.dinv * lvl.omega  # pre-smooth from zero
+        import torch as _t
+
+        x = b * lvl.dinv
+        x.local.mul_(lvl.omega)  # pre-smooth from zero
         x = self._smooth(lvl, x, b, self.smooth_iters - 1)
-        r = b - lvl.A.dot(x)
+        r = lvl.A.dot(x)
+        r.local.sub_(b.local).neg_()
         P = lvl.Pdown
-        rc = P.T.dot(r)  # restriction (CSC col-split SpMV w/ reduction)
+        rc = lvl.Rdown.dot(r)  # restriction (CSC col-split SpMV w/ reduction)
         xc = self._vcycle(li + 1, rc)
         x += P.dot(xc)
         x = self._smooth(lvl, x, b, self.smooth_iters)

@@ -190,12 +190,8 @@ def spgemm_csr(A, B, a_col_lo, vdt):
                + torch.repeat_interleave(indptr[esc_rows], ec))
         indices[dst] = esc_sub.indices
         values[dst] = esc_sub.values
-    # sort columns within each row (hash compaction is unordered)
-    rows = torch.repeat_interleave(
-        torch.arange(m, dtype=torch.int64, device=dev), counts)
-    key = rows * B.ncols + indices.long()
-    order = torch.argsort(key)
-    return LocalCSR(indptr, indices[order], values[order], m, B.ncols)
+    # rows are emitted sorted (in-kernel LDS bitonic compaction)
+    return LocalCSR(indptr, indices, values, m, B.ncols)
 
 
 def _spgemm_esc(A, B, a_col_lo, vdt):

@@ -69,6 +69,11 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
   __shared__ int64_t keys[WAVES_PER_BLOCK][HASH];
   __shared__ __align__(16) char accs_raw[WAVES_PER_BLOCK * HASH * sizeof(T)];
   auto accs = reinterpret_cast<T(*)[HASH]>(accs_raw);
+  // compacted (col, val) pairs; count <= HASH/2 by the ub binning
+  __shared__ int64_t ckeys_s[WAVES_PER_BLOCK][HASH / 2 + 1];
+  __shared__ __align__(16) char cvals_raw[WAVES_PER_BLOCK * (HASH / 2 + 1) * sizeof(T)];
+  auto ckeys = ckeys_s;
+  auto cvals = reinterpret_cast<T(*)[HASH / 2 + 1]>(cvals_raw);
   __shared__ int slots[WAVES_PER_BLOCK];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
@@ -104,13 +109,50 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
   }
   __builtin_amdgcn_wave_barrier();
   __threadfence_block();
+  // compact into [0, count) then LDS bitonic sort by column — emitting the
+  // row SORTED skips the wrapper's global argsort (2 full passes over C).
   const int64_t base = cip[r];
+  const int count = (int)(cip[r + 1] - base);
   for (int i = lane; i < HASH; i += WAVE) {
     if (keys[wid][i] != -1) {
       int slot = atomicAdd(&slots[wid], 1);
-      cix[base + slot] = (index_t)keys[wid][i];
-      cv[base + slot] = accs[wid][i];
+      ckeys[wid][slot] = keys[wid][i];
+      cvals[wid][slot] = accs[wid][i];
     }
+  }
+  __builtin_amdgcn_wave_barrier();
+  __threadfence_block();
+  // pad to the next power of two with +inf sentinels
+  int np2 = 1;
+  while (np2 < count) np2 <<= 1;
+  for (int i = count + lane; i < np2; i += WAVE) {
+    ckeys[wid][i] = INT64_MAX;
+  }
+  __builtin_amdgcn_wave_barrier();
+  __threadfence_block();
+  for (int k = 2; k <= np2; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int i = lane; i < np2; i += WAVE) {
+        int ixj = i ^ j;
+        if (ixj > i) {
+          bool up = (i & k) == 0;
+          int64_t ki = ckeys[wid][i], kj = ckeys[wid][ixj];
+          if ((ki > kj) == up) {
+            ckeys[wid][i] = kj;
+            ckeys[wid][ixj] = ki;
+            T tv = cvals[wid][i];
+            cvals[wid][i] = cvals[wid][ixj];
+            cvals[wid][ixj] = tv;
+          }
+        }
+      }
+      __builtin_amdgcn_wave_barrier();
+      __threadfence_block();
+    }
+  }
+  for (int i = lane; i < count; i += WAVE) {
+    cix[base + i] = (index_t)ckeys[wid][i];
+    cv[base + i] = cvals[wid][i];
   }
 }
 
