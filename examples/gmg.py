@@ -80,10 +80,15 @@ class GMG:
             self.levels[-1].Rdown = R  # cached: keeps gather plans warm
             cur = Ac
             cur_nx = (cur_nx - 1) // 2
-        # replicated coarse solve (machine-scoping equivalent)
+        # replicated coarse solve (machine-scoping equivalent); kept on the
+        # GPU so the V-cycle never touches the host (hipGraph-capturable)
         coarse = self.levels[-1].A.to_scipy_sparse_csr().toarray()
-        self.coarse_inv = np.linalg.pinv(coarse)
+        dev = self.levels[0].A._values.device
+        self.coarse_inv_t = torch.as_tensor(np.linalg.pinv(coarse), device=dev,
+                                            dtype=self.levels[0].A._values.dtype)
         self.smooth_iters = smooth_iters
+        self._graph = None
+        self._graph_tried = False
 
     def _smooth(self, lvl, x, b, iters):
         # weighted Jacobi: x += omega * dinv * (b - A x), fused in-place
@@ -96,10 +101,12 @@ class GMG:
         return x
 
     def _vcycle(self, li, b):
+        from sparse.parallel import comm as _comm
+
         lvl = self.levels[li]
         if li == len(self.levels) - 1:
-            bg = np.asarray(b)
-            xg = self.coarse_inv @ bg
+            bg = b.gather() if _comm.world_size() > 1 else b.local
+            xg = self.coarse_inv_t @ bg
             return darray.asdistarray(xg)
         import torch as _t
 
@@ -120,8 +127,47 @@ class GMG:
         return linalg.LinearOperator(
             (n, n), matvec=lambda r, out=None: self._matvec(r, out))
 
+    def _try_capture(self, r):
+        """hipGraph-capture the V-cycle (the MI355X replacement for the
+        reference's Legion tracing): one graph replay instead of ~300
+        eager launches per preconditioner application.  ws=1 only — RCCL
+        collectives stay outside graphs for now."""
+        self._graph_tried = True
+        from sparse.parallel import comm as _comm
+
+        if not r.local.is_cuda or _comm.world_size() > 1 or                 os.environ.get("SPARSE_NO_HIPGRAPH"):
+            return
+        try:
+            self._gin = r.local.clone()
+            rin = darray.DistArray.from_local(self._gin, r.partition, r.gshape)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(2):
+                    self._vcycle(0, rin)
+            torch.cuda.current_stream().wait_stream(side)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                z = self._vcycle(0, rin)
+            self._graph = g
+            self._gout = z.local
+        except Exception as e:  # capture unsupported -> stay eager
+            print(f"[gmg] hipGraph capture unavailable ({e}); eager V-cycles")
+            self._graph = None
+
     def _matvec(self, r, out=None):
-        z = self._vcycle(0, darray.asdistarray(r))
+        r = darray.asdistarray(r)
+        if not self._graph_tried:
+            self._try_capture(r)
+        if self._graph is not None:
+            self._gin.copy_(r.local)
+            self._graph.replay()
+            if out is not None:
+                out.local.copy_(self._gout)
+                return out
+            return darray.DistArray.from_local(self._gout.clone(), r.partition,
+                                               r.gshape)
+        z = self._vcycle(0, r)
         if out is not None:
             out.local.copy_(z.local.to(out.local.dtype))
             return out
