@@ -39,7 +39,9 @@ __global__ void csr_diagonal_kernel(const int64_t* __restrict__ indptr,
   if (lo < indptr[r + 1] && (int64_t)indices[lo] == target) out[r] = vals[lo];
 }
 
-// y[row - rlo] += v * x[col]; one thread per nz, col via binary search
+// y[row - rlo] += v * x[col]; one thread per COLUMN (no per-nnz search —
+// the scatter targets absorb the atomics; a 1B-nnz GMG restriction was
+// bound by the old per-nnz binary search)
 template <typename T, typename index_t>
 __global__ void csc_spmv_kernel(const int64_t* __restrict__ colptr,
                                 const index_t* __restrict__ rowidx,
@@ -47,13 +49,16 @@ __global__ void csc_spmv_kernel(const int64_t* __restrict__ colptr,
                                 const T* __restrict__ x,
                                 T* __restrict__ y, int64_t ncl, int64_t rlo,
                                 int64_t nnz) {
-  int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (p >= nnz) return;
-  int64_t c = ub_i64(colptr, ncl + 1, p) - 1;
-  atomic_add_any(&y[(int64_t)rowidx[p] - rlo], vals[p] * x[c]);
+  int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= ncl) return;
+  const T xv = x[c];
+  const int64_t e = colptr[c + 1];
+  for (int64_t p = colptr[c]; p < e; ++p) {
+    atomic_add_any(&y[(int64_t)rowidx[p] - rlo], vals[p] * xv);
+  }
 }
 
-// C[(row-rlo), j] += v * B[col, j]; one wave per nz, lanes over j
+// C[(row-rlo), j] += v * B[col, j]; one wave per column, lanes over j
 template <typename T, typename index_t>
 __global__ void csc_spmm_kernel(const int64_t* __restrict__ colptr,
                                 const index_t* __restrict__ rowidx,
@@ -61,14 +66,16 @@ __global__ void csc_spmm_kernel(const int64_t* __restrict__ colptr,
                                 const T* __restrict__ B, T* __restrict__ C,
                                 int64_t ncl, int64_t rlo, int64_t k,
                                 int64_t nnz) {
-  int64_t w = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t c = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   int lane = threadIdx.x % WAVE;
-  if (w >= nnz) return;
-  int64_t c = ub_i64(colptr, ncl + 1, w) - 1;
-  int64_t r = (int64_t)rowidx[w] - rlo;
-  T v = vals[w];
-  for (int64_t j = lane; j < k; j += WAVE) {
-    atomic_add_any(&C[r * k + j], v * B[c * k + j]);
+  if (c >= ncl) return;
+  const int64_t e = colptr[c + 1];
+  for (int64_t p = colptr[c]; p < e; ++p) {
+    int64_t r = (int64_t)rowidx[p] - rlo;
+    T v = vals[p];
+    for (int64_t j = lane; j < k; j += WAVE) {
+      atomic_add_any(&C[r * k + j], v * B[c * k + j]);
+    }
   }
 }
 
@@ -112,7 +119,8 @@ void csc_spmv_hip(at::Tensor colptr, at::Tensor rowidx, at::Tensor vals,
   DISPATCH_VALUES(vals.scalar_type(), "csc_spmv", [&] {
     using T = scalar_t;
     DISPATCH_INDEX(rowidx.scalar_type(), "csc_spmv_idx", [&] {
-      hipLaunchKernelGGL((csc_spmv_kernel<T, index_t>), dim3((nnz + 255) / 256),
+      int64_t ncl = colptr.numel() - 1;
+      hipLaunchKernelGGL((csc_spmv_kernel<T, index_t>), dim3((ncl + 256) / 256),
                          dim3(256), 0, cur_stream(), colptr.data_ptr<int64_t>(),
                          rowidx.data_ptr<index_t>(), vals.data_ptr<T>(),
                          x.data_ptr<T>(), y.data_ptr<T>(), colptr.numel() - 1,
@@ -128,9 +136,9 @@ void csc_spmm_hip(at::Tensor colptr, at::Tensor rowidx, at::Tensor vals,
   DISPATCH_VALUES(vals.scalar_type(), "csc_spmm", [&] {
     using T = scalar_t;
     DISPATCH_INDEX(rowidx.scalar_type(), "csc_spmm_idx", [&] {
-      int64_t threads = nnz * WAVE;
+      int64_t threads = (colptr.numel() - 1) * WAVE;
       hipLaunchKernelGGL((csc_spmm_kernel<T, index_t>),
-                         dim3((threads + 255) / 256), dim3(256), 0, cur_stream(),
+                         dim3((threads + 256) / 256), dim3(256), 0, cur_stream(),
                          colptr.data_ptr<int64_t>(), rowidx.data_ptr<index_t>(),
                          vals.data_ptr<T>(), B.data_ptr<T>(), C.data_ptr<T>(),
                          colptr.numel() - 1, rlo, B.size(1), nnz);
