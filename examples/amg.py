@@ -134,27 +134,88 @@ def build_hierarchy(A: csr_array, theta: float, max_coarse: int):
             "omega": omega,
         })
         cur = Ac
+    import torch
+
     coarse = cur.to_scipy_sparse_csr().toarray()
-    levels.append({"A": cur, "coarse_inv": np.linalg.pinv(coarse)})
+    levels.append({"A": cur, "coarse_inv": torch.as_tensor(
+        np.linalg.pinv(coarse), device=cur._values.device,
+        dtype=cur._values.dtype)})
     return levels
 
 
 def vcycle(levels, li, b):
+    from sparse.parallel import comm as _comm
+
     lvl = levels[li]
     if "coarse_inv" in lvl:
-        return darray.asdistarray(lvl["coarse_inv"] @ np.asarray(b))
+        bg = b.gather() if _comm.world_size() > 1 else b.local
+        return darray.asdistarray(lvl["coarse_inv"] @ bg)
     A, dinv, omega = lvl["A"], lvl["dinv"], lvl["omega"]
-    x = b * dinv * omega
-    r = b - A.dot(x)
-    x += r * dinv * omega
-    r = b - A.dot(x)
+    x = b * dinv
+    x.local.mul_(omega)
+    for _ in range(1):
+        r = A.dot(x)
+        r.local.sub_(b.local).neg_()
+        x.local.addcmul_(r.local, dinv.local, value=omega)
+    r = A.dot(x)
+    r.local.sub_(b.local).neg_()
     rc = lvl["R"].dot(r)
     xc = vcycle(levels, li + 1, rc)
     x += lvl["P"].dot(xc)
     for _ in range(2):
-        r = b - A.dot(x)
-        x += r * dinv * omega
+        r = A.dot(x)
+        r.local.sub_(b.local).neg_()
+        x.local.addcmul_(r.local, dinv.local, value=omega)
     return x
+
+
+class _GraphedVcycle:
+    """hipGraph-captured V-cycle (same recipe as gmg.py)."""
+
+    def __init__(self, levels):
+        self.levels = levels
+        self.graph = None
+        self.tried = False
+
+    def __call__(self, r, out=None):
+        import torch
+
+        r = darray.asdistarray(r)
+        from sparse.parallel import comm as _comm
+
+        if not self.tried:
+            self.tried = True
+            if r.local.is_cuda and _comm.world_size() == 1 and not os.environ.get(
+                    "SPARSE_NO_HIPGRAPH"):
+                try:
+                    self.gin = r.local.clone()
+                    rin = darray.DistArray.from_local(self.gin, r.partition, r.gshape)
+                    side = torch.cuda.Stream()
+                    side.wait_stream(torch.cuda.current_stream())
+                    with torch.cuda.stream(side):
+                        for _ in range(2):
+                            vcycle(self.levels, 0, rin)
+                    torch.cuda.current_stream().wait_stream(side)
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        z = vcycle(self.levels, 0, rin)
+                    self.graph = g
+                    self.gout = z.local
+                except Exception as e:
+                    print(f"[amg] hipGraph capture unavailable ({e})")
+                    self.graph = None
+        if self.graph is not None:
+            self.gin.copy_(r.local)
+            self.graph.replay()
+            if out is not None:
+                out.local.copy_(self.gout)
+                return out
+            return darray.DistArray.from_local(self.gout.clone(), r.partition, r.gshape)
+        z = vcycle(self.levels, 0, r)
+        if out is not None:
+            out.local.copy_(z.local.to(out.local.dtype))
+            return out
+        return z
 
 
 def main():
@@ -169,7 +230,9 @@ def main():
     setup_ms = timer.stop()
     op_complexity = sum(l["A"].nnz for l in levels) / levels[0]["A"].nnz
 
-    M = linalg.LinearOperator((n, n), matvec=lambda r, out=None: _apply(levels, r, out))
+    gv = _GraphedVcycle(levels)
+    M = linalg.LinearOperator((n, n), matvec=gv)
+    gv(b)  # warm + capture outside the timer
 
     it_count = [0]
     timer.start()
@@ -183,14 +246,6 @@ def main():
               f"setup={setup_ms:.1f}ms solve={solve_ms:.1f}ms iters={it_count[0]} "
               f"({it_count[0] / max(solve_ms, 1e-9) * 1000.0:.2f} iters/s) "
               f"residual={float(r.norm().item()):.3e} info={info}")
-
-
-def _apply(levels, r, out=None):
-    z = vcycle(levels, 0, darray.asdistarray(r))
-    if out is not None:
-        out.local.copy_(z.local.to(out.local.dtype))
-        return out
-    return z
 
 
 if __name__ == "__main__":

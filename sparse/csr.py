@@ -165,6 +165,21 @@ class csr_array(CompressedBase, DenseSparseBase):
 
     # -- internal -------------------------------------------------------------
     def _init_from_local(self, indptr, indices, values, partition, shape):
+        import os as _os
+
+        if _os.environ.get("SPARSE_BOUNDS_CHECKS"):
+            # debug-mode accessor checks (reference Legion_BOUNDS_CHECKS,
+            # legate_sparse_cpp.cmake:62)
+            assert indptr.numel() == partition.count(comm.rank()) + 1, \
+                "indptr length != local rows + 1"
+            assert int(indptr[0].item()) == 0, "indptr[0] != 0"
+            assert bool((indptr[1:] >= indptr[:-1]).all().item()), \
+                "indptr not monotone"
+            assert int(indptr[-1].item()) == indices.numel() == values.numel(), \
+                "nnz mismatch between indptr/indices/values"
+            if indices.numel():
+                assert int(indices.min().item()) >= 0 and \
+                    int(indices.max().item()) < shape[1], "column out of bounds"
         self._indptr = indptr
         self._indices = indices
         self._values = values
