@@ -13,6 +13,8 @@ from __future__ import annotations
 
 import argparse
 import json
+import os
+import sys
 import time
 
 import numpy as np
@@ -82,15 +84,38 @@ def main() -> None:
     rz = r.dot(r)
     q = darray.zeros((n,), dtype=dtype)
 
+    graph = None
     if on_gpu:
+        rz_buf = rz.clone()
 
-        def cg_step() -> None:  # fully-fused MI355X path (see linalg.cg)
-            nonlocal rz
+        def cg_step_gpu() -> None:  # fully-fused MI355X path (see linalg.cg)
+            # rz_buf is a persistent 0-dim buffer so the step is
+            # hipGraph-capturable (fixed addresses across replays)
             pq = A.spmv_dot(pvec, q)
-            linalg.cg_axpby(xv, pvec, rz, pq, isalpha=True, negate=False)
-            rz_new = linalg._axpby_norm2(r, q, rz, pq, negate=True)
-            linalg.cg_axpby(pvec, r, rz_new, rz, isalpha=False, negate=False)
-            rz = rz_new
+            linalg.cg_axpby(xv, pvec, rz_buf, pq, isalpha=True, negate=False)
+            rz_new = linalg._axpby_norm2(r, q, rz_buf, pq, negate=True)
+            linalg.cg_axpby(pvec, r, rz_new, rz_buf, isalpha=False, negate=False)
+            rz_buf.copy_(rz_new)
+
+        cg_step = cg_step_gpu
+        if args.gpus == 1 and not os.environ.get("SPARSE_NO_HIPGRAPH"):
+            # capture one CG iteration as a single hipGraph (launch-overhead
+            # free replay); multi-GPU stays eager (RCCL outside graphs)
+            try:
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(3):
+                        cg_step_gpu()
+                torch.cuda.current_stream().wait_stream(side)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    cg_step_gpu()
+                graph = g
+                cg_step = graph.replay
+            except Exception as e:
+                print(f"# hipGraph capture unavailable ({e}); eager steps",
+                      file=sys.stderr)
 
     else:
 
