@@ -34,12 +34,30 @@ def _as_real(t: torch.Tensor) -> torch.Tensor:
     return torch.view_as_real(t).flatten() if t.is_complex() else t
 
 
+def _nccl(group=None) -> bool:
+    return initialized() and dist.get_backend(group) == "nccl"
+
+
+def _to_wire_device(t: torch.Tensor, group=None) -> torch.Tensor:
+    """NCCL moves only CUDA tensors; host scalars ride via a device copy."""
+    if _nccl(group) and not t.is_cuda:
+        return t.cuda()
+    return t
+
+
 def all_reduce_(t: torch.Tensor, op: str = "sum", group=None, async_op: bool = False):
     """In-place all-reduce; no-op at world size 1. Complex -> viewed as real
     (valid for sum/min/max-on-abs is NOT handled — sum only for complex)."""
     if not initialized() or world_size(group) == 1:
         return None
     ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
+    if _nccl(group) and not t.is_cuda:
+        # host tensor under RCCL: reduce a device copy, write back (sync)
+        tt = _as_real(t) if t.is_complex() else t
+        dev = tt.cuda()
+        dist.all_reduce(dev, op=ops[op], group=group)
+        tt.copy_(dev.cpu())
+        return None
     if t.is_complex():
         assert op == "sum", "complex all-reduce supports sum only"
         return dist.all_reduce(torch.view_as_real(t), op=ops[op], group=group, async_op=async_op)
@@ -58,7 +76,11 @@ def all_gather_rows(local: torch.Tensor, counts: Sequence[int], group=None) -> t
     local_c = local.contiguous()
     cplx = local_c.is_complex()
     wire = torch.view_as_real(local_c) if cplx else local_c
+    host = not wire.is_cuda
+    wire = _to_wire_device(wire, group)
     maxc = max(counts)
+    if maxc == 0:
+        return local
     tail = wire.shape[1:]
     pad = torch.empty((maxc, *tail), dtype=wire.dtype, device=wire.device)
     if wire.shape[0] > 0:
@@ -67,6 +89,8 @@ def all_gather_rows(local: torch.Tensor, counts: Sequence[int], group=None) -> t
     dist.all_gather(outs, pad, group=group)
     pieces = [outs[r][: counts[r]] for r in range(ws)]
     out = torch.cat(pieces, dim=0)
+    if host and out.is_cuda:
+        out = out.cpu()
     if cplx:
         out = torch.view_as_complex(out)
     return out
@@ -96,17 +120,33 @@ def all_to_all_v(send: List[torch.Tensor], group=None,
     wire = [(_as_real(s.contiguous())) for s in send]
     wdtype = wire[0].dtype
     backend = dist.get_backend(group)
+    host = not wire[0].is_cuda
+    if backend == "nccl" and host:
+        wire = [w.cuda() for w in wire]
     if recv_counts is None:
-        cdev = device if backend == "nccl" else torch.device("cpu")
+        cdev = wire[0].device if backend == "nccl" else torch.device("cpu")
         counts = torch.tensor([int(s.numel()) for s in wire], dtype=torch.int64, device=cdev)
         all_counts = [torch.zeros(ws, dtype=torch.int64, device=cdev) for _ in range(ws)]
         dist.all_gather(all_counts, counts, group=group)
         recv_counts = [int(all_counts[src][me].item()) for src in range(ws)]
     elif cplx:
         recv_counts = [2 * c for c in recv_counts]
-    recv = [torch.empty(c, dtype=wdtype, device=device) for c in recv_counts]
+    recv = [torch.empty(c, dtype=wdtype, device=wire[0].device)
+            for c in recv_counts]
     if backend == "nccl":
-        dist.all_to_all(recv, wire, group=group)
+        # batched p2p of only the NONZERO pairs (xGMI neighbor exchange;
+        # zero-size guards per the reference's NCCL hang workaround,
+        # sort.cu:259-263)
+        p2p = []
+        for peer in range(ws):
+            if peer != me and wire[peer].numel() > 0:
+                p2p.append(dist.P2POp(dist.isend, wire[peer], peer, group=group))
+            if peer != me and recv[peer].numel() > 0:
+                p2p.append(dist.P2POp(dist.irecv, recv[peer], peer, group=group))
+        if p2p:
+            for req in dist.batch_isend_irecv(p2p):
+                req.wait()
+        recv[me].copy_(wire[me])
     else:
         reqs = []
         for peer in range(ws):
@@ -118,6 +158,8 @@ def all_to_all_v(send: List[torch.Tensor], group=None,
                 dist.recv(recv[peer], src=peer, group=group)
         for r in reqs:
             r.wait()
+    if host and backend == "nccl":
+        recv = [r.cpu() for r in recv]
     if cplx:
         recv = [torch.view_as_complex(r.view(-1, 2)) for r in recv]
     return recv
