@@ -306,3 +306,47 @@ def test_ell_spmv_three_piece_split():
                                    xw[n - 60:].clone()), q, x.local, plan.lo)
     assert torch.allclose(q, ref)
     assert np.isclose(float(d.item()), float(torch.dot(x.local, ref).item()))
+
+
+@pytest.mark.parametrize("dt", [np.float32, np.complex128])
+def test_ell_spmv_dtypes_gpu(dt):
+    """ELL fast path across dtypes (banded => ELL-eligible)."""
+    from sparse import csr_array, gallery
+
+    A = gallery.banded(3000, ndiags=9, dtype=np.float64)
+    if np.dtype(dt) != np.float64:
+        A = A.astype(dt)
+    x = sample_dense(3000, seed=50, dtype=dt)
+    assert A._ell() is not None
+    y = A @ x
+    sref = A.to_scipy_sparse_csr()
+    assert np.allclose(np.asarray(y), sref @ x, rtol=1e-4 if dt == np.float32 else 1e-10)
+
+
+def test_solve_ivp_gpu():
+    """solve_ivp on GPU DistArrays (exercises the rk_calc_dy kernel)."""
+    from sparse import asdistarray, integrate
+
+    y0 = np.linspace(1.0, 2.0, 5000)
+
+    def f(t, y):
+        return y * (-0.3)
+
+    res = integrate.solve_ivp(f, (0, 2.0), y0, method="RK45", rtol=1e-8,
+                              atol=1e-10)
+    assert res.success
+    assert np.allclose(res.y[:, -1], y0 * np.exp(-0.6), rtol=1e-6)
+
+
+def test_profiling_ranges_gpu():
+    from sparse import csr_array, profiling
+
+    s = sample_csr(200, 200, 0.05, seed=51)
+    A = csr_array(s)
+    x = sample_dense(200, seed=52)
+    with profiling.profile() as prof:
+        from sparse import linalg
+
+        linalg.cg(A.T @ A + csr_array(5 * np.eye(200)), x, tol=1e-6, maxiter=50)
+    keys = [e.key for e in prof.key_averages()]
+    assert any("sparse::" in k for k in keys), keys[:10]
