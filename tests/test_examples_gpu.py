@@ -1,0 +1,36 @@
+"""End-to-end example runs on the GPU (small sizes; @gpu)."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+EX = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "examples")
+
+
+def run(script, *a, timeout=300):
+    r = subprocess.run([sys.executable, os.path.join(EX, script), *a],
+                       capture_output=True, timeout=timeout)
+    assert r.returncode == 0, r.stdout.decode()[-1200:] + r.stderr.decode()[-1200:]
+    return r.stdout.decode()
+
+
+def test_pde_gpu():
+    out = run("pde.py", "-nx", "512", "-ny", "512", "-throughput", "-max_iter", "100")
+    assert "Solve finished: 100 iterations" in out
+
+
+def test_gmg_gpu():
+    out = run("gmg.py", "-N", "255", "-maxiter", "50")
+    assert "info=0" in out
+
+
+def test_amg_gpu():
+    out = run("amg.py", "-n", "65536", "-maxiter", "100")
+    assert "info=0" in out
+
+
+def test_dot_micro_gpu():
+    out = run("dot_microbenchmark.py", "-n", "1000000", "-iters", "20", "-warmup", "3")
+    assert "SpMVs" in out
