@@ -133,31 +133,19 @@ def all_to_all_v(send: List[torch.Tensor], group=None,
         recv_counts = [2 * c for c in recv_counts]
     recv = [torch.empty(c, dtype=wdtype, device=wire[0].device)
             for c in recv_counts]
-    if backend == "nccl":
-        # batched p2p of only the NONZERO pairs (xGMI neighbor exchange;
-        # zero-size guards per the reference's NCCL hang workaround,
-        # sort.cu:259-263)
-        p2p = []
-        for peer in range(ws):
-            if peer != me and wire[peer].numel() > 0:
-                p2p.append(dist.P2POp(dist.isend, wire[peer], peer, group=group))
-            if peer != me and recv[peer].numel() > 0:
-                p2p.append(dist.P2POp(dist.irecv, recv[peer], peer, group=group))
-        if p2p:
-            for req in dist.batch_isend_irecv(p2p):
-                req.wait()
-        recv[me].copy_(wire[me])
-    else:
-        reqs = []
-        for peer in range(ws):
-            if peer != me and wire[peer].numel() > 0:
-                reqs.append(dist.isend(wire[peer], dst=peer, group=group))
-        recv[me].copy_(wire[me])
-        for peer in range(ws):
-            if peer != me and recv[peer].numel() > 0:
-                dist.recv(recv[peer], src=peer, group=group)
-        for r in reqs:
-            r.wait()
+    # batched p2p of only the NONZERO pairs — one code path for gloo (CI)
+    # and RCCL (xGMI neighbor exchange); zero-size guards per the
+    # reference's NCCL hang workaround, sort.cu:259-263.
+    p2p = []
+    for peer in range(ws):
+        if peer != me and wire[peer].numel() > 0:
+            p2p.append(dist.P2POp(dist.isend, wire[peer], peer, group=group))
+        if peer != me and recv[peer].numel() > 0:
+            p2p.append(dist.P2POp(dist.irecv, recv[peer], peer, group=group))
+    if p2p:
+        for req in dist.batch_isend_irecv(p2p):
+            req.wait()
+    recv[me].copy_(wire[me])
     if host and backend == "nccl":
         recv = [r.cpu() for r in recv]
     if cplx:
