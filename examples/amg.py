@@ -153,19 +153,14 @@ def vcycle(levels, li, b):
     A, dinv, omega = lvl["A"], lvl["dinv"], lvl["omega"]
     x = b * dinv
     x.local.mul_(omega)
-    for _ in range(1):
-        r = A.dot(x)
-        r.local.sub_(b.local).neg_()
-        x.local.addcmul_(r.local, dinv.local, value=omega)
+    x = A.jacobi_smooth(x, b, dinv, omega)
     r = A.dot(x)
     r.local.sub_(b.local).neg_()
     rc = lvl["R"].dot(r)
     xc = vcycle(levels, li + 1, rc)
     x += lvl["P"].dot(xc)
     for _ in range(2):
-        r = A.dot(x)
-        r.local.sub_(b.local).neg_()
-        x.local.addcmul_(r.local, dinv.local, value=omega)
+        x = A.jacobi_smooth(x, b, dinv, omega)
     return x
 
 

@@ -91,13 +91,9 @@ class GMG:
         self._graph_tried = False
 
     def _smooth(self, lvl, x, b, iters):
-        # weighted Jacobi: x += omega * dinv * (b - A x), fused in-place
-        import torch as _t
-
+        # fused weighted-Jacobi sweeps (single ELL kernel each)
         for _ in range(iters):
-            r = lvl.A.dot(x)
-            r.local.sub_(b.local).neg_()  # r = b - A x, in place
-            x.local.addcmul_(r.local, lvl.dinv.local, value=lvl.omega)
+            x = lvl.A.jacobi_smooth(x, b, lvl.dinv, lvl.omega)
         return x
 
     def _vcycle(self, li, b):

@@ -416,6 +416,29 @@ class csr_array(CompressedBase, DenseSparseBase):
             return out
         return DistArray.from_local(Clocal, self.partition, (self.shape[0], B.shape[1]))
 
+    def jacobi_smooth(self, x: DistArray, b: DistArray, dinv: DistArray,
+                      omega: float, out: Optional[DistArray] = None) -> DistArray:
+        """One fused weighted-Jacobi sweep x' = x + omega*dinv*(b - A x)
+        (the GMG/AMG smoother; reference WeightedJacobi gmg.py:247-285).
+        Single kernel on the ELL fast path; generic fallback otherwise."""
+        ell = self._ell()
+        if out is None:
+            out = DistArray.from_local(torch.empty_like(x.local), x.partition,
+                                       x.gshape)
+        if ell is not None:
+            from . import kernels
+
+            plan = self._xplan(x.partition)
+            pieces = plan.gather_halos(x.local)
+            kernels.ell_jacobi(ell, pieces, x.local, b.local, dinv.local,
+                               omega, out.local, plan.lo)
+            return out
+        r = self._spmv(x)
+        r.local.sub_(b.local).neg_()  # r = b - A x
+        out.local.copy_(x.local)
+        out.local.addcmul_(r.local, dinv.local, value=omega)
+        return out
+
     def spmv_dot(self, p: DistArray, q: DistArray) -> torch.Tensor:
         """Fused q = A@p and all-reduced sum(p*q) — the CG p·Ap in one kernel
         (GPU, real dtypes; MI355X fusion: saves re-reading p and q)."""

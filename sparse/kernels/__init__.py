@@ -286,6 +286,14 @@ def ell_spmv(ell: EllMirror, pieces, y, col_lo: int):
                    hhi.contiguous(), y, ell.W, ell.m, int(col_lo))
 
 
+def ell_jacobi(ell: EllMirror, pieces, xloc, b, dinv, omega, xout, col_lo: int):
+    """Fused weighted-Jacobi sweep: xout = x + omega*dinv*(b - A x)."""
+    hlo, own, hhi = pieces
+    ext().ell_jacobi(ell.eidx, ell.evals, hlo.contiguous(), own.contiguous(),
+                     hhi.contiguous(), xloc, b, dinv, xout, ell.W, ell.m,
+                     int(col_lo), float(omega))
+
+
 def ell_spmv_dot(ell: EllMirror, pieces, y, p, col_lo: int):
     hlo, own, hhi = pieces
     mp = ell.evals.numel() // ell.W

@@ -417,3 +417,69 @@ void ell_spmv_dot_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
                       int64_t W, int64_t m, int64_t col_lo) {
   ell_spmv_hip(eidx, evals, hlo, own, hhi, y, W, m, col_lo, pvec, dot_partial);
 }
+
+// ---------------------------------------------------------------------------
+// Fused weighted-Jacobi sweep on the ELL mirror:
+//   x_out[r] = x[r] + omega * dinv[r] * (b[r] - (A x)[r])
+// One pass over A + 4 vector streams instead of spmv + 2 elementwise
+// kernels (the GMG/AMG smoother, reference WeightedJacobi gmg.py:247-285).
+namespace {
+
+template <typename T, typename index_t>
+__global__ __launch_bounds__(BLK) void ell_jacobi_kernel(
+    const index_t* __restrict__ eidx, const T* __restrict__ evals,
+    const T* __restrict__ hlo, const T* __restrict__ own,
+    const T* __restrict__ hhi, const T* __restrict__ xloc,
+    const T* __restrict__ b, const T* __restrict__ dinv,
+    T* __restrict__ xout, int64_t m, int64_t mp, int W, int64_t col_lo,
+    int64_t nlo, int64_t nown, T omega) {
+  const int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
+  const int64_t r0 = 2 * t;
+  if (r0 >= mp) return;
+  T a0 = ZeroOf<T>::value(), a1 = ZeroOf<T>::value();
+  for (int k = 0; k < W; ++k) {
+    const int64_t base = (int64_t)k * mp + r0;
+    struct alignas(2 * sizeof(index_t) <= 16 ? 2 * sizeof(index_t) : 16) IP { index_t a, b; };
+    struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
+    const IP ii = *reinterpret_cast<const IP*>(&eidx[base]);
+    const TP vv = *reinterpret_cast<const TP*>(&evals[base]);
+    a0 += vv.a * xpiece((int64_t)ii.a - col_lo, hlo, nlo, own, nown, hhi);
+    a1 += vv.b * xpiece((int64_t)ii.b - col_lo, hlo, nlo, own, nown, hhi);
+  }
+  if (r0 + 1 < m) {
+    struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
+    const TP xv = *reinterpret_cast<const TP*>(&xloc[r0]);
+    const TP bv = *reinterpret_cast<const TP*>(&b[r0]);
+    const TP dv = *reinterpret_cast<const TP*>(&dinv[r0]);
+    TP out{xv.a + omega * dv.a * (bv.a - a0),
+           xv.b + omega * dv.b * (bv.b - a1)};
+    *reinterpret_cast<TP*>(&xout[r0]) = out;
+  } else if (r0 < m) {
+    xout[r0] = xloc[r0] + omega * dinv[r0] * (b[r0] - a0);
+  }
+}
+
+}  // namespace
+
+void ell_jacobi_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
+                    at::Tensor own, at::Tensor hhi, at::Tensor xloc,
+                    at::Tensor b, at::Tensor dinv, at::Tensor xout,
+                    int64_t W, int64_t m, int64_t col_lo, double omega) {
+  const int64_t mp = evals.numel() / W;
+  const int64_t nblocks = (mp / 2 + BLK - 1) / BLK;
+  const int64_t nlo = hlo.numel();
+  const int64_t nown = own.numel();
+  DISPATCH_VALUES(evals.scalar_type(), "ell_jacobi", [&] {
+    using T = scalar_t;
+    DISPATCH_INDEX(eidx.scalar_type(), "ell_jacobi_idx", [&] {
+      const T* hlo_p = nlo ? hlo.data_ptr<T>() : own.data_ptr<T>();
+      const T* hhi_p = hhi.numel() ? hhi.data_ptr<T>() : own.data_ptr<T>();
+      hipLaunchKernelGGL((ell_jacobi_kernel<T, index_t>), dim3(nblocks),
+                         dim3(BLK), 0, cur_stream(), eidx.data_ptr<index_t>(),
+                         evals.data_ptr<T>(), hlo_p, own.data_ptr<T>(), hhi_p,
+                         xloc.data_ptr<T>(), b.data_ptr<T>(),
+                         dinv.data_ptr<T>(), xout.data_ptr<T>(), m, mp, (int)W,
+                         col_lo, nlo, nown, static_cast<T>(omega));
+    });
+  });
+}

@@ -138,3 +138,19 @@ def test_empty_and_shape_ctor():
     A = csr_array((4, 5))
     assert A.nnz == 0 and A.shape == (4, 5)
     assert np.allclose(np.asarray(A.todense()), np.zeros((4, 5)))
+
+
+def test_jacobi_smooth_cpu():
+    import sparse
+    from sparse import darray, gallery
+
+    A = gallery.poisson2d(16)
+    n = A.shape[0]
+    x = darray.random((n,), seed=60)
+    b = darray.random((n,), seed=61)
+    d = A.diagonal()
+    dinv = darray.DistArray.from_local(1.0 / d.local, d.partition, d.shape)
+    out = A.jacobi_smooth(x, b, dinv, 0.7)
+    r = b - A.dot(x)
+    expect = np.asarray(x) + 0.7 * np.asarray(dinv) * np.asarray(r)
+    assert np.allclose(np.asarray(out), expect, rtol=1e-12)

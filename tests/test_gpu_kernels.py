@@ -350,3 +350,18 @@ def test_profiling_ranges_gpu():
         linalg.cg(A.T @ A + csr_array(5 * np.eye(200)), x, tol=1e-6, maxiter=50)
     keys = [e.key for e in prof.key_averages()]
     assert any("sparse::" in k for k in keys), keys[:10]
+
+
+def test_jacobi_smooth_gpu():
+    from sparse import darray, gallery
+
+    A = gallery.poisson2d(64)
+    n = A.shape[0]
+    x = darray.random((n,), seed=60)
+    b = darray.random((n,), seed=61)
+    d = A.diagonal()
+    dinv = darray.DistArray.from_local(1.0 / d.local, d.partition, d.shape)
+    out = A.jacobi_smooth(x, b, dinv, 0.7)
+    r = b - A.dot(x)
+    expect = np.asarray(x) + 0.7 * np.asarray(dinv) * np.asarray(r)
+    assert np.allclose(np.asarray(out), expect, rtol=1e-12)
