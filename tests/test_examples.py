@@ -60,3 +60,10 @@ def test_gmg_3d():
     assert "info=0" in out
     iters = int(out.split("iters=")[1].split()[0])
     assert iters <= 15, out
+
+
+def test_quantum_mis():
+    out = run("quantum_mis.py", "-l", "3", "-T", "2.0")
+    assert "approx ratio" in out
+    ratio = float(out.split("approx ratio = ")[1].split()[0])
+    assert 0.3 < ratio <= 1.0
