@@ -213,6 +213,51 @@ __global__ __launch_bounds__(BLK) void v23_nnz(const int64_t* __restrict__ indpt
   if (!has_carry && tid == 0) carry_row[b] = -1;
 }
 
+// ---- V5t: ELL with compile-time W (fully unrolled k-loop) -----------------
+template <int WCT>
+__global__ __launch_bounds__(BLK) void v5_ell_ct(const int* __restrict__ eidx,
+                                                 const double* __restrict__ evals,
+                                                 const double* __restrict__ x,
+                                                 double* __restrict__ y,
+                                                 int64_t mp) {
+  int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
+  int64_t r0 = 2 * t;
+  if (r0 >= mp) return;
+  double a0 = 0, a1 = 0;
+#pragma unroll
+  for (int k = 0; k < WCT; ++k) {
+    const double2 v = *reinterpret_cast<const double2*>(&evals[(int64_t)k * mp + r0]);
+    const int2 ii = *reinterpret_cast<const int2*>(&eidx[(int64_t)k * mp + r0]);
+    a0 += v.x * x[ii.x];
+    a1 += v.y * x[ii.y];
+  }
+  *reinterpret_cast<double2*>(&y[r0]) = double2{a0, a1};
+}
+
+// ---- V7: ELL 4 rows/thread ------------------------------------------------
+__global__ __launch_bounds__(BLK) void v7_ell4(const int* __restrict__ eidx,
+                                               const double* __restrict__ evals,
+                                               const double* __restrict__ x,
+                                               double* __restrict__ y,
+                                               int64_t mp, int W) {
+  int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
+  int64_t r0 = 4 * t;
+  if (r0 >= mp) return;
+  double a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+  for (int k = 0; k < W; ++k) {
+    const int64_t base = (int64_t)k * mp + r0;
+    const double2 va = *reinterpret_cast<const double2*>(&evals[base]);
+    const double2 vb = *reinterpret_cast<const double2*>(&evals[base + 2]);
+    const int4 ii = *reinterpret_cast<const int4*>(&eidx[base]);
+    a0 += va.x * x[ii.x];
+    a1 += va.y * x[ii.y];
+    a2 += vb.x * x[ii.z];
+    a3 += vb.y * x[ii.w];
+  }
+  *reinterpret_cast<double2*>(&y[r0]) = double2{a0, a1};
+  *reinterpret_cast<double2*>(&y[r0 + 2]) = double2{a2, a3};
+}
+
 // ---- V5: column-major padded ELL, 2 rows/thread, 16B loads ---------------
 template <bool FUSE_DOT>
 __global__ __launch_bounds__(BLK) void v5_ell(const int* __restrict__ eidx,
@@ -277,7 +322,7 @@ int main(int argc, char** argv) {
   int64_t nb8 = (nnz + 2047) / 2048, nb16 = (nnz + 4095) / 4096;
   // build ELL mirror on host (W=5)
   int W = 5;
-  int64_t mp = (N + 1) & ~1ll;
+  int64_t mp = (N + 3) & ~3ll;
   int* eidx; double* evals;
   CHECK(hipMalloc(&eidx, W * mp * sizeof(int)));
   CHECK(hipMalloc(&evals, W * mp * sizeof(double)));
@@ -311,7 +356,8 @@ int main(int argc, char** argv) {
   std::vector<V> vs = {{"v0_stream", 0}, {"v0_gather", 1}, {"v1_row", 2},
                        {"v1_row_swz", 3}, {"v2_scalar_swz", 4},
                        {"v3_quad_swz", 5}, {"v3_quad_noswz", 6},
-                       {"v4_quad_vt16_swz", 7}, {"v5_ell", 8}};
+                       {"v4_quad_vt16_swz", 7}, {"v5_ell", 8},
+                       {"v6_ell_ctW", 9}, {"v7_ell4", 10}};
   const int ROUNDS = 7, REPS = 3;
   std::vector<std::vector<float>> times(vs.size());
   hipEvent_t t0, t1;
@@ -353,6 +399,12 @@ int main(int argc, char** argv) {
             break;
           case 8:
             hipLaunchKernelGGL((v5_ell<false>), dim3((mp / 2 + BLK - 1) / BLK), dim3(BLK), 0, 0, eidx, evals, x, y, nullptr, nullptr, mp, W);
+            break;
+          case 9:
+            hipLaunchKernelGGL((v5_ell_ct<5>), dim3((mp / 2 + BLK - 1) / BLK), dim3(BLK), 0, 0, eidx, evals, x, y, mp);
+            break;
+          case 10:
+            hipLaunchKernelGGL(v7_ell4, dim3((mp / 4 + BLK - 1) / BLK), dim3(BLK), 0, 0, eidx, evals, x, y, mp, W);
             break;
         }
       }
