@@ -46,15 +46,55 @@ def _neighbor_masks(graph) -> List[int]:
     return nbr
 
 
+def _popcount64(a: np.ndarray) -> np.ndarray:
+    a = a.astype(np.uint64, copy=True)
+    out = np.zeros(a.shape, dtype=np.int64)
+    while a.any():
+        out += (a & np.uint64(1)).astype(np.int64)
+        a >>= np.uint64(1)
+    return out
+
+
 def enumerate_independent_sets(graph, k: int, prevk_sets=None, prevk_queues=None):
     """Independence sets of size k from those of size k-1 (BFS expansion,
-    reference quantum.cc:27-...).  Returns (sets, queues): bitmask lists.
+    reference quantum.cc:27-... with its IntSet bitsets).  Returns
+    (sets, queues) bitmask sequences — numpy-vectorized int64 masks for
+    graphs with <= 63 nodes, python arbitrary-width ints beyond.
 
     queues[i] = candidate nodes with index greater than every member of
     sets[i] and not adjacent to it (the canonical-extension frontier)."""
     n = graph.number_of_nodes()
+    if n > 63:
+        return _enumerate_py(graph, k, prevk_sets, prevk_queues)
+    nbr = np.array(_neighbor_masks(graph), dtype=np.int64)
+    if k == 1:
+        sets = np.array([1 << v for v in range(n)], dtype=np.int64)
+        above = np.array([(((1 << n) - 1) >> (v + 1)) << (v + 1) for v in range(n)],
+                         dtype=np.int64)
+        queues = above & ~nbr
+        return sets, queues
+    assert prevk_sets is not None and prevk_queues is not None
+    S = np.asarray(prevk_sets, dtype=np.int64)
+    Q = np.asarray(prevk_queues, dtype=np.int64)
+    out_s, out_q = [], []
+    # vectorized per candidate node v: every parent whose queue has bit v
+    for v in range(n):
+        bit = np.int64(1 << v)
+        rows = (Q & bit) != 0
+        if not rows.any():
+            continue
+        lowmask = np.int64(((1 << (v + 1)) - 1))
+        out_s.append(S[rows] | bit)
+        out_q.append(Q[rows] & ~lowmask & ~nbr[v])
+    if not out_s:
+        return np.zeros(0, dtype=np.int64), np.zeros(0, dtype=np.int64)
+    return np.concatenate(out_s), np.concatenate(out_q)
+
+
+def _enumerate_py(graph, k, prevk_sets, prevk_queues):
+    """Arbitrary-width fallback (> 63 nodes)."""
+    n = graph.number_of_nodes()
     nbr = _neighbor_masks(graph)
-    full = (1 << n) - 1
     if k == 1:
         sets = [1 << v for v in range(n)]
         queues = []
@@ -72,16 +112,16 @@ def enumerate_independent_sets(graph, k: int, prevk_sets=None, prevk_queues=None
         while q:
             v = (q & -q).bit_length() - 1
             q &= q - 1
-            newS = S | (1 << v)
-            newQ = Q & ~((1 << (v + 1)) - 1) & ~nbr[v]
-            sets.append(newS)
-            queues.append(newQ)
+            sets.append(S | (1 << v))
+            queues.append(Q & ~((1 << (v + 1)) - 1) & ~nbr[v])
     return sets, queues
 
 
 def sets_to_sizes(queues, graph) -> np.ndarray:
     """Popcount of each candidate queue (reference SETS_TO_SIZES)."""
-    return np.array([bin(q).count("1") for q in queues], dtype=np.int64)
+    if isinstance(queues, np.ndarray):
+        return _popcount64(queues)
+    return np.array([bin(int(q)).count("1") for q in queues], dtype=np.int64)
 
 
 def independence_polynomial(graph) -> List[int]:
@@ -92,10 +132,13 @@ def independence_polynomial(graph) -> List[int]:
     for k in range(1, graph.number_of_nodes() + 1):
         sets, nbrs = enumerate_independent_sets(graph, k, prevk_sets=sets,
                                                 prevk_queues=nbrs)
-        if not sets:
+        if len(sets) == 0:
             break
         ip.append(len(sets))
-        if all(q == 0 for q in nbrs):
+        if isinstance(nbrs, np.ndarray):
+            if not nbrs.any():
+                break
+        elif all(q == 0 for q in nbrs):
             break
     return ip
 
@@ -122,32 +165,39 @@ class HamiltonianDriver:
         for k in range(1, n + 1):
             sets, nbrs = enumerate_independent_sets(graph, k, prevk_sets=sets,
                                                     prevk_queues=nbrs)
-            if not sets:
+            if len(sets) == 0:
                 break
             self.ip.append(len(sets))
-            all_sets.append(list(sets))
-            if all(q == 0 for q in nbrs):
+            all_sets.append(sets)
+            if isinstance(nbrs, np.ndarray):
+                if not nbrs.any():
+                    break
+            elif all(q == 0 for q in nbrs):
                 break
         self.nstates = sum(self.ip)
-        # ascending ids: group k starts at offsets[k]
+        # ascending ids: group k starts at offsets[k]; within each group ids
+        # follow VALUE-sorted mask order so subset lookup is a searchsorted
         offsets = np.concatenate([[0], np.cumsum(self.ip)])
-        id_of = {}
+        groups = []
         for k, group in enumerate(all_sets):
-            for i, S in enumerate(group):
-                id_of[S] = offsets[k] + i
+            g = np.asarray(group, dtype=np.int64) if not isinstance(
+                group, np.ndarray) else group
+            groups.append(np.sort(g))
         rows_l, cols_l = [], []
-        for k in range(1, len(all_sets)):
-            for S in all_sets[k]:
-                sid = id_of[S]
-                q = S
-                while q:
-                    v = (q & -q).bit_length() - 1
-                    q &= q - 1
-                    tid = id_of[S & ~(1 << v)]
-                    rows_l.append(sid)
-                    cols_l.append(tid)
-        rows = np.asarray(rows_l, dtype=np.int64)
-        cols = np.asarray(cols_l, dtype=np.int64)
+        for k in range(1, len(groups)):
+            Sk = groups[k]
+            sid = offsets[k] + np.arange(len(Sk), dtype=np.int64)
+            # peel the k set bits of every mask, vectorized per position
+            rem = Sk.copy()
+            for _ in range(k):
+                low = rem & -rem
+                Tm = Sk & ~low  # subset with that member removed
+                tid = offsets[k - 1] + np.searchsorted(groups[k - 1], Tm)
+                rows_l.append(sid)
+                cols_l.append(tid)
+                rem = rem & ~low
+        rows = np.concatenate(rows_l)
+        cols = np.concatenate(cols_l)
         # reference state ordering: largest sets first, empty set last
         rows = self.nstates - 1 - rows
         cols = self.nstates - 1 - cols
