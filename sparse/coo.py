@@ -14,6 +14,9 @@ import numbers
 import numpy as np
 import torch
 
+import scipy.sparse as _sps
+
+from .coverage import clone_scipy_arr_kind
 from .base import CompressedBase
 from .darray import DistArray, asdistarray
 from .ops import local as ops
@@ -29,6 +32,7 @@ from .types import (
 )
 
 
+@clone_scipy_arr_kind(_sps.coo_matrix)
 class coo_array(CompressedBase):
     _format = "coo"
 

@@ -15,6 +15,9 @@ import numbers
 import numpy as np
 import torch
 
+import scipy.sparse as _sps
+
+from .coverage import clone_scipy_arr_kind
 from .base import CompressedBase, DenseSparseBase
 from .darray import DistArray, asdistarray
 from .ops import local as ops
@@ -31,6 +34,7 @@ from .types import (
 )
 
 
+@clone_scipy_arr_kind(_sps.csc_matrix)
 class csc_array(CompressedBase, DenseSparseBase):
     _format = "csc"
 

@@ -15,6 +15,9 @@ from typing import Optional
 import numpy as np
 import torch
 
+import scipy.sparse as _sps
+
+from .coverage import clone_scipy_arr_kind
 from .base import CompressedBase, DenseSparseBase
 from .darray import DistArray, asdistarray
 from .ops import local as ops
@@ -36,6 +39,7 @@ def _default_partition(m: int) -> RowPartition:
     return RowPartition.equal(m, comm.world_size())
 
 
+@clone_scipy_arr_kind(_sps.csr_matrix)
 class csr_array(CompressedBase, DenseSparseBase):
     _format = "csr"
 
@@ -102,10 +106,9 @@ class csr_array(CompressedBase, DenseSparseBase):
                 data, indices, indptr = arg
                 if isinstance(indptr, torch.Tensor) and shape is not None and (
                     indptr.numel() == _default_partition(shape[0]).count(comm.rank()) + 1
-                    and (comm.world_size() > 1 or True)
                     and indptr.numel() != shape[0] + 1
                 ):
-                    # already-local tensors (internal fast path)
+                    # already-local torch tensors (internal fast path)
                     part = _default_partition(shape[0])
                     self._init_from_local(indptr.to(torch.int64), indices, data, part, shape)
                     return
@@ -129,9 +132,6 @@ class csr_array(CompressedBase, DenseSparseBase):
                     vals, part, gshape)
                 return
             raise NotImplementedError(f"cannot construct csr_array from tuple of len {len(arg)}")
-
-        if isinstance(arg, (coo_like := tuple())):
-            pass
 
         from .coo import coo_array
         from .csc import csc_array

@@ -16,12 +16,16 @@ import numbers
 import numpy as np
 import torch
 
+import scipy.sparse as _sps
+
+from .coverage import clone_scipy_arr_kind
 from .base import CompressedBase
 from .parallel import comm
 from .runtime import runtime
 from .types import promote_value_dtype, to_numpy_dtype, to_torch_dtype
 
 
+@clone_scipy_arr_kind(_sps.dia_matrix)
 class dia_array(CompressedBase):
     _format = "dia"
 
