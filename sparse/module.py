@@ -202,11 +202,15 @@ def random(m, n, density=0.01, format="coo", dtype=np.float64, random_state=None
         random_state = 42
     rng = (np.random.default_rng(random_state)
            if isinstance(random_state, (int, np.integer)) else random_state)
-    # sample without replacement in flat index space
-    if nnz > 0:
+    # sample without replacement in flat index space; for huge matrices
+    # draw with replacement and deduplicate (memory-safe at n*m >> 1e8)
+    if nnz <= 0:
+        flat = np.zeros(0, dtype=np.int64)
+    elif m * n <= 100_000_000:
         flat = rng.choice(m * n, size=nnz, replace=False)
     else:
-        flat = np.zeros(0, dtype=np.int64)
+        flat = np.unique(rng.integers(0, m * n, size=int(nnz * 1.05) + 16))
+        flat = rng.permutation(flat)[:nnz]
     rows = flat // n
     cols = flat % n
     if data_rvs is None:

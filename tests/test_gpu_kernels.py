@@ -365,3 +365,18 @@ def test_jacobi_smooth_gpu():
     r = b - A.dot(x)
     expect = np.asarray(x) + 0.7 * np.asarray(dinv) * np.asarray(r)
     assert np.allclose(np.asarray(out), expect, rtol=1e-12)
+
+
+def test_cg_complex_gpu():
+    """Hermitian positive-definite complex CG (eager GPU path: complex axpby
+    kernels + vdot)."""
+    from sparse import csr_array, linalg
+
+    n = 120
+    s = sample_csr(n, n, 0.2, seed=70, dtype=np.complex128)
+    H = (s + s.conj().T + 2 * n * sps.eye(n)).tocsr()
+    rng = np.random.default_rng(71)
+    b = rng.random(n) + 1j * rng.random(n)
+    x, info = linalg.cg(csr_array(H), b, tol=1e-10, maxiter=500, conv_test_iters=5)
+    assert info == 0
+    assert np.allclose(H @ np.asarray(x), b, atol=1e-6)
