@@ -124,7 +124,7 @@ def build_hierarchy(A: csr_array, theta: float, max_coarse: int):
         DinvA = csr_array.from_local(lc.indptr, lc.indices, scaled_vals,
                                      cur.partition, cur.shape)
         P = (DinvA @ Td) + Td  # (I - omega D^-1 A) T
-        R = P.T
+        R = P.T.tocsr()  # materialized row-partitioned (ELL restriction)
         AP = cur @ P
         Ac = R @ AP  # Galerkin product (distributed SpGEMM chain)
         d = cur.diagonal()

@@ -74,10 +74,12 @@ class GMG:
             if len(self.levels) >= maxl or cur.shape[0] <= coarse_threshold or cur_nx < 7:
                 break
             P = interp(cur_nx)
-            R = P.T
+            # materialize R row-partitioned: restriction becomes a row-split
+            # ELL SpMV (no atomic scatter) at the cost of one transposed copy
+            R = P.T.tocsr()
             Ac = (R @ cur) @ P  # Galerkin triple product (distributed SpGEMM)
             self.levels[-1].Pdown = P
-            self.levels[-1].Rdown = R  # cached: keeps gather plans warm
+            self.levels[-1].Rdown = R
             cur = Ac
             cur_nx = (cur_nx - 1) // 2
         # replicated coarse solve (machine-scoping equivalent); kept on the
