@@ -28,6 +28,8 @@ def main() -> None:
     p.add_argument("--warmup", type=int, default=15)
     p.add_argument("--nx", type=int, default=16384)
     p.add_argument("--dtype", default="fp64", choices=["fp64", "fp32"])
+    p.add_argument("--weak", action="store_true",
+                   help="scale nx by sqrt(world) (reference weak-scaling protocol)")
     args = p.parse_args()
 
     import sparse
@@ -44,6 +46,12 @@ def main() -> None:
     dtype = np.float64 if args.dtype == "fp64" else np.float32
 
     nx = args.nx
+    if args.weak:
+        import math
+
+        from sparse.parallel import comm as _c
+
+        nx = int(round(args.nx * math.sqrt(max(1, _c.world_size()))))
     n = nx * nx
     t0 = time.time()
     A = gallery.poisson2d(nx, dtype=dtype)
@@ -154,7 +162,7 @@ def main() -> None:
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
-            "scaling": "strong",
+            "scaling": "weak" if args.weak else "strong",
             "vs_baseline": iters_per_sec / 75.9,
             "dtype": args.dtype,
             "data": "synthetic",

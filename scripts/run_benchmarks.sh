@@ -1,0 +1,27 @@
+#!/bin/bash
+# Reproduce the numbers in profiles/RESULTS_r01.md on an MI355X box.
+# Usage: bash scripts/run_benchmarks.sh [quick]
+set -x
+QUICK=${1:-}
+ITERS=300; STEPS=60
+if [ "$QUICK" = "quick" ]; then ITERS=50; STEPS=20; fi
+
+# headline: CG on 5-pt Poisson (BASELINE.json config)
+python bench.py --steps $STEPS --warmup 10 --nx 16384
+
+# reference's own headline config (75.9 it/s on V100)
+python examples/pde.py -nx 6000 -ny 6000 -throughput -max_iter $ITERS
+
+# SpMV microbenchmark (347.7 it/s on V100)
+python examples/dot_microbenchmark.py -n 10000000 -iters $ITERS
+
+# GMG (37.2 it/s at N=4500 on V100)
+python examples/gmg.py -N 4095 -maxiter 100
+python examples/gmg.py -N 511 -dim 3 -maxiter 60
+
+# AMG + Galerkin SpGEMM
+python examples/amg.py -n 1048576 -maxiter 200
+python examples/spgemm_microbenchmark.py -nx 2047 -iters 10
+
+# quantum MIS demo
+python examples/quantum_mis.py -l 4 -T 3.0
