@@ -11,6 +11,8 @@ void spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
               int64_t, double);
 void spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                   at::Tensor, at::Tensor, int64_t);
+void csr_row_spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      at::Tensor, int64_t);
 void axpby_norm2_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool,
                      bool, at::Tensor);
 void build_ell_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
@@ -71,6 +73,8 @@ TORCH_LIBRARY(sparse_hip, m) {
         "bool negate) -> ()");
   m.def("spmv_dot(Tensor indptr, Tensor indices, Tensor values, Tensor x, "
         "Tensor(a!) y, Tensor pvec, Tensor(b!) dot_out, int col_lo) -> ()");
+  m.def("csr_row_spmv(Tensor indptr, Tensor indices, Tensor values, Tensor x, "
+        "Tensor(a!) y, int col_lo) -> ()");
   m.def("axpby_norm2(Tensor(a!) y, Tensor x, Tensor a, Tensor b, bool isalpha, "
         "bool negate, Tensor(b!) dot_out) -> ()");
   m.def("build_ell(Tensor indptr, Tensor indices, Tensor values, "
@@ -117,6 +121,7 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("mult_dense", mult_dense_hip);
   m.impl("axpby", axpby_hip);
   m.impl("spmv_dot", spmv_dot_hip);
+  m.impl("csr_row_spmv", csr_row_spmv_hip);
   m.impl("axpby_norm2", axpby_norm2_hip);
   m.impl("build_ell", build_ell_hip);
   m.impl("ell_spmv", ell_spmv_plain_hip);
