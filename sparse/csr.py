@@ -189,6 +189,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         self._window_cache = None
         self._plan_cache = {}
         self._ell_cache = None
+        self._maxrow_cache = None
 
     @classmethod
     def from_local(cls, indptr, indices, values, partition, shape) -> "csr_array":
@@ -200,7 +201,14 @@ class csr_array(CompressedBase, DenseSparseBase):
     @property
     def local(self) -> ops.LocalCSR:
         return ops.LocalCSR(self._indptr, self._indices, self._values,
-                            self.partition.count(comm.rank()), self.shape[1])
+                            self.partition.count(comm.rank()), self.shape[1],
+                            self._maxrow_cache)
+
+    def _max_row_nnz(self) -> int:
+        if self._maxrow_cache is None:
+            d = self._indptr[1:] - self._indptr[:-1]
+            self._maxrow_cache = int(d.max().item()) if d.numel() else 0
+        return self._maxrow_cache
 
     def _col_window(self):
         """[lo,hi) min/max column window of the local slab (the MinMaxImage
@@ -391,9 +399,11 @@ class csr_array(CompressedBase, DenseSparseBase):
             kernels.ell_spmv(ell, pieces, ylocal, plan.lo)
         else:
             xw = plan.gather(x.local.to(vdt))
+            self._max_row_nnz()
             lc = self.local
             if lc.values.dtype != vdt:
-                lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt), lc.nrows, lc.ncols)
+                lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt),
+                                  lc.nrows, lc.ncols, lc.max_row_nnz)
             ylocal = ops.spmv(lc, xw, col_lo=plan.lo)
         part = self.partition
         if out is not None:

@@ -44,12 +44,13 @@ def require():
 
 # -- high-level wrappers ------------------------------------------------------
 def spmv(A, x, y, col_lo: int, beta: float):
-    if beta == 0.0 and A.nrows > 0 and A.nnz > 0:
-        # short-row heuristic: thread-per-row beats the nnz-split kernel
-        # when rows are short (measured; tools/spmv_bench.hip)
-        if A.nnz <= 24 * A.nrows:
-            ext().csr_row_spmv(A.indptr, A.indices, A.values, x, y, int(col_lo))
-            return
+    # short-row heuristic: thread-per-row beats the nnz-split kernel on
+    # short uniform rows (measured; tools/spmv_bench.hip); a long row would
+    # serialize one thread, so require a known max row length
+    if (beta == 0.0 and A.nnz > 0 and A.max_row_nnz is not None
+            and A.max_row_nnz <= 64):
+        ext().csr_row_spmv(A.indptr, A.indices, A.values, x, y, int(col_lo))
+        return
     ext().spmv(A.indptr, A.indices, A.values, x, y, int(col_lo), float(beta))
 
 

@@ -53,6 +53,7 @@ class LocalCSR:
     values: torch.Tensor
     nrows: int
     ncols: int  # global number of columns
+    max_row_nnz: Optional[int] = None  # cached by csr_array (kernel dispatch)
 
     @property
     def nnz(self) -> int:
