@@ -48,7 +48,7 @@ def spmv(A, x, y, col_lo: int, beta: float):
     # short uniform rows (measured; tools/spmv_bench.hip); a long row would
     # serialize one thread, so require a known max row length
     if (beta == 0.0 and A.nnz > 0 and A.max_row_nnz is not None
-            and A.max_row_nnz <= 64):
+            and A.max_row_nnz <= 64 and not os.environ.get("SPARSE_NO_ROWKERNEL")):
         ext().csr_row_spmv(A.indptr, A.indices, A.values, x, y, int(col_lo))
         return
     ext().spmv(A.indptr, A.indices, A.values, x, y, int(col_lo), float(beta))
