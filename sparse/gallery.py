@@ -24,16 +24,31 @@ __all__ = ["poisson2d", "poisson3d", "banded"]
 def _assemble(rows_i, cols_list, vals_list, valid_list, shape, part, dtype):
     rt = runtime()
     mloc = rows_i.numel()
+    nd = len(cols_list)
     idt = index_dtype_for(shape)
-    C = torch.stack(cols_list, dim=1)
-    V = torch.stack(vals_list, dim=1).to(to_torch_dtype(dtype))
-    M = torch.stack(valid_list, dim=1)
-    counts = M.sum(dim=1)
-    indptr = torch.zeros(mloc + 1, dtype=torch.int64, device=C.device)
+    vdt = to_torch_dtype(dtype)
+    dev = cols_list[0].device if nd else rows_i.device
+    counts = torch.zeros(mloc, dtype=torch.int64, device=dev)
+    for mvec in valid_list:
+        counts += mvec.to(torch.int64)
+    indptr = torch.zeros(mloc + 1, dtype=torch.int64, device=dev)
     torch.cumsum(counts, 0, out=indptr[1:])
-    flat = M.reshape(-1)
-    indices = C.reshape(-1)[flat].to(idt)
-    values = V.reshape(-1)[flat]
+    nnz = int(indptr[-1].item()) if mloc else 0
+    indices = torch.empty(nnz, dtype=idt, device=dev)
+    values = torch.empty(nnz, dtype=vdt, device=dev)
+    # chunk rows so each flat masked select stays < 2^31 elements (torch's
+    # boolean indexing overflows beyond that) and peak memory stays bounded
+    chunk = max(1, min(mloc, (1 << 30) // max(nd, 1)))
+    for a in range(0, mloc, chunk):
+        b = min(a + chunk, mloc)
+        Cc = torch.stack([c[a:b] for c in cols_list], dim=1)
+        Vc = torch.stack([v[a:b] for v in vals_list], dim=1).to(vdt)
+        Mc = torch.stack([mv[a:b] for mv in valid_list], dim=1)
+        flat = Mc.reshape(-1)
+        s0 = int(indptr[a].item())
+        e0 = int(indptr[b].item())
+        indices[s0:e0] = Cc.reshape(-1)[flat].to(idt)
+        values[s0:e0] = Vc.reshape(-1)[flat]
     return csr_array.from_local(indptr.to(rt.device), indices.to(rt.device),
                                 values.to(rt.device), part, shape)
 
