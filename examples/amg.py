@@ -81,16 +81,91 @@ def aggregate_host(C: sps.csr_matrix) -> np.ndarray:
     return agg
 
 
+def aggregate_device(C: csr_array, seed: int = 17) -> np.ndarray:
+    """Luby-style MIS aggregation on the DEVICE using the tropical
+    (max, lexicographic) semiring SpMV — the distributed counterpart of the
+    reference's maximal_independent_set + mis_aggregate (amg.py:199-283).
+
+    Round structure: unaggregated local priority maxima become aggregate
+    roots; their unaggregated neighbors join them; repeat.  Returns the
+    aggregate id per node (gathered numpy)."""
+    import torch
+
+    n = C.shape[0]
+    rng = np.random.default_rng(seed)
+    prio_g = rng.permutation(n).astype(np.int64) + 1  # distinct, >= 1
+    prio = darray.asdistarray(prio_g)
+    ids = darray.arange(n)
+    agg = darray.asdistarray(np.full(n, -1, dtype=np.int64))
+    root_of = darray.asdistarray(np.full(n, -1, dtype=np.int64))
+    def trop(fields):
+        return C.tropical_spmv(darray.DistArray.from_local(
+            fields, agg.partition, (n, 2))).local
+
+    for _round in range(64):
+        live = agg.local < 0
+        if not bool(comm_any(live)):
+            break
+        # MIS(2) roots (reference amg.py:199-238 uses distance-2): a live
+        # node whose priority is the maximum over its 2-hop neighborhood.
+        # The self-inclusive 1-hop tropical max applied twice gives the
+        # 2-hop max.
+        f0 = torch.where(live, prio.local, torch.zeros_like(prio.local))
+        nb1 = trop(torch.stack([f0, ids.local], dim=1))
+        nb2 = trop(nb1)
+        roots = live & (f0 == nb2[:, 0]) & (f0 > 0)
+        agg.local[roots] = ids.local[roots]
+        # join pass 1: live neighbors of a root join it
+        rf = torch.where(roots, prio.local, torch.zeros_like(prio.local))
+        j1 = trop(torch.stack([rf, ids.local], dim=1))
+        live = agg.local < 0
+        join = live & (j1[:, 0] > 0)
+        agg.local[join] = j1[join, 1]
+        # join pass 2: remaining live nodes adopt the aggregate of their
+        # highest-priority aggregated neighbor (distance-2 attachment)
+        af = torch.where(agg.local >= 0, prio.local,
+                         torch.zeros_like(prio.local))
+        j2 = trop(torch.stack([af, agg.local], dim=1))
+        live = agg.local < 0
+        join2 = live & (j2[:, 0] > 0)
+        agg.local[join2] = j2[join2, 1]
+    # leftovers (isolated): own aggregate
+    left = agg.local < 0
+    agg.local[left] = ids.local[left]
+    # compact ids
+    g = agg.gather().cpu().numpy()
+    _, compact = np.unique(g, return_inverse=True)
+    return compact
+
+
+def comm_any(mask) -> bool:
+    import torch
+
+    t = mask.any().to(torch.int32)
+    from sparse.parallel import comm as _c
+
+    _c.all_reduce_(t, op="max")
+    return bool(t.item())
+
+
 def build_hierarchy(A: csr_array, theta: float, max_coarse: int):
     """Returns list of levels: dicts with A, P, R, dinv, omega."""
     levels = []
     cur = A
     while cur.shape[0] > max_coarse and len(levels) < 20:
         # ---- setup graph on gathered structure (host) ----------------------
-        S = cur.to_scipy_sparse_csr()
-        C = strength_mask_host(S, theta)
-        agg = aggregate_host(C)
+        if os.environ.get("SPARSE_AMG_HOST_AGG"):
+            S = cur.to_scipy_sparse_csr()
+            agg = aggregate_host(strength_mask_host(S, theta))
+        else:
+            Cs = cur  # theta=0: full connectivity strength graph
+            if theta > 0:
+                S = cur.to_scipy_sparse_csr()
+                Cs = csr_array(strength_mask_host(S, theta))
+            agg = aggregate_device(Cs)
         nc = int(agg.max()) + 1
+        if nc >= 0.9 * cur.shape[0]:  # aggregation stalled
+            break
         if nc >= cur.shape[0]:
             break
         # tentative prolongator: T[i, agg[i]] = 1, column-normalized

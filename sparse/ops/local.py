@@ -272,19 +272,21 @@ def tropical_spmv(A: LocalCSR, x: torch.Tensor, col_lo: int = 0) -> torch.Tensor
         y = torch.zeros((A.nrows, x.shape[1]), dtype=torch.int64, device=A.device)
         hip().tropical_spmv(A, x.contiguous(), y, col_lo)
         return y
-    # CPU: per-row lexicographic max over neighbor fields
+    # CPU: vectorized per-row lexicographic max via lexsort + group tails
     ip = A.indptr.cpu().numpy()
-    ix = (A.indices.cpu().numpy() - col_lo)
+    ix = (A.indices.cpu().numpy().astype(np.int64) - col_lo)
     xn = x.cpu().numpy()
+    nnz = len(ix)
     y = np.zeros((A.nrows, x.shape[1]), dtype=np.int64)
-    for i in range(A.nrows):
-        best = None
-        for p in range(ip[i], ip[i + 1]):
-            cand = xn[ix[p]]
-            if best is None or tuple(cand) > tuple(best):
-                best = cand
-        if best is not None:
-            y[i] = best
+    if nnz:
+        rows = np.repeat(np.arange(A.nrows), np.diff(ip))
+        cand = xn[ix]  # (nnz, f)
+        keys = tuple(cand[:, f] for f in range(cand.shape[1] - 1, -1, -1))
+        order = np.lexsort(keys + (rows,))
+        srows = rows[order]
+        # last entry of each row group is its lexicographic max
+        tail = np.r_[srows[1:] != srows[:-1], True]
+        y[srows[tail]] = cand[order][tail]
     return torch.as_tensor(y, device=A.device)
 
 
