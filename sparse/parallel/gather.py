@@ -54,30 +54,78 @@ class WindowGatherPlan:
             a, b = max(self.lo, xpart.start(p)), min(self.hi, xpart.stop(p))
             self.recv_counts.append(max(0, b - a))
         assert sum(self.recv_counts) == self.hi - self.lo
+        self._ctx_cache = {}
+
+    def _halo_ctx(self, xlocal: torch.Tensor):
+        """Persistent halo-exchange context for a (repeatedly updated
+        in-place) operand tensor: send views into x, one contiguous recv
+        buffer per side with per-peer slices — zero allocations and no
+        concatenation per exchange."""
+        key = (xlocal.data_ptr(), tuple(xlocal.shape), xlocal.dtype)
+        ctx = self._ctx_cache.get(key)
+        if ctx is not None:
+            return ctx
+        ws = comm.world_size(self.group)
+        me = comm.rank(self.group)
+        xs, xe = self.xpart.start(me), self.xpart.stop(me)
+        own_a, own_b = max(self.lo, xs), min(self.hi, xe)
+        send_views = []
+        for p in range(ws):
+            if p == me:
+                send_views.append(None)
+                continue
+            a, b = self.send_ranges[p]
+            send_views.append(xlocal[a:b] if b > a else None)
+        n_lo = sum(self.recv_counts[p] for p in range(me))
+        n_hi = sum(self.recv_counts[p] for p in range(me + 1, ws))
+        hlo = torch.empty(n_lo, dtype=xlocal.dtype, device=xlocal.device)
+        hhi = torch.empty(n_hi, dtype=xlocal.dtype, device=xlocal.device)
+        recv_views = []
+        off = 0
+        for p in range(ws):
+            c = self.recv_counts[p]
+            if p == me or c == 0:
+                recv_views.append(None)
+            elif p < me:
+                recv_views.append(hlo[off: off + c])
+            else:
+                recv_views.append(hhi[off - n_lo: off - n_lo + c])
+            if p != me:
+                off += c
+            else:
+                off = n_lo  # own piece not received; hhi offsets start at 0
+        own = xlocal[own_a - xs: own_b - xs]
+        ctx = (send_views, recv_views, hlo, own, hhi)
+        if len(self._ctx_cache) >= 8:
+            # contexts hold views (keep operand storage alive): bound them
+            self._ctx_cache.clear()
+        self._ctx_cache[key] = ctx
+        return ctx
 
     def gather_halos(self, xlocal: torch.Tensor):
         """Exchange ONLY the halo pieces; my own slab portion is used in
         place (no self-copy through the collective).  Returns
         (halo_lo, own_view, halo_hi)."""
         ws = comm.world_size(self.group)
-        me = comm.rank(self.group)
-        xs, xe = self.xpart.start(me), self.xpart.stop(me)
-        own_a, own_b = max(self.lo, xs), min(self.hi, xe)
         if ws == 1:
             return xlocal[:0], xlocal[self.lo: self.hi], xlocal[:0]
-        send = []
+        import torch.distributed as dist
+
+        send_views, recv_views, hlo, own, hhi = self._halo_ctx(xlocal)
+        p2p = []
+        me = comm.rank(self.group)
         for p in range(ws):
-            if p == me:
-                send.append(xlocal[:0])
-                continue
-            a, b = self.send_ranges[p]
-            send.append(xlocal[a:b].reshape(-1))
-        rc = [0 if p == me else self.recv_counts[p] for p in range(ws)]
-        recv = comm.all_to_all_v(send, group=self.group, recv_counts=rc)
-        halo_lo = torch.cat([recv[p] for p in range(me)]) if me > 0 else xlocal[:0]
-        halo_hi = torch.cat([recv[p] for p in range(me + 1, ws)]) if me < ws - 1 else xlocal[:0]
-        own = xlocal[own_a - xs: own_b - xs]
-        return halo_lo, own, halo_hi
+            if send_views[p] is not None and send_views[p].numel():
+                p2p.append(dist.P2POp(dist.isend, send_views[p].contiguous()
+                           if not send_views[p].is_contiguous() else send_views[p],
+                           p, group=self.group))
+            if recv_views[p] is not None:
+                p2p.append(dist.P2POp(dist.irecv, recv_views[p], p,
+                                      group=self.group))
+        if p2p:
+            for req in dist.batch_isend_irecv(p2p):
+                req.wait()
+        return hlo, own, hhi
 
     def gather(self, xlocal: torch.Tensor) -> torch.Tensor:
         """Return the window x[lo:hi) (dim 0 slices; works for 1-D and 2-D)."""

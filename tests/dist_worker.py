@@ -140,6 +140,12 @@ def extra_halo_check():
         hlo, own, hhi = plan.gather_halos(x)
         re = torch.cat([hlo, own, hhi])
         assert torch.equal(re, full), (rank, lo, hi)
+        # persistent-context path: repeat with mutated values in place
+        x.mul_(2.0)
+        full2 = plan.gather(x)
+        hlo, own, hhi = plan.gather_halos(x)
+        assert torch.equal(torch.cat([hlo, own, hhi]), full2), (rank, lo, hi, "ctx")
+        x.mul_(0.5)
 
 
 def extra_samplesort_check():
