@@ -30,7 +30,7 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
     int64_t* __restrict__ nnz_out, int64_t a_col_lo) {
   constexpr int RPB = WAVES_PER_BLOCK * (WAVE / LPR);
   __shared__ int64_t keys[RPB][HASH];
-  __shared__ int counts[RPB];
+  __shared__ int counts[2 * RPB];  // [0,RPB): distinct-col counts; [RPB,2RPB): scan totals
   const int slot_id = threadIdx.x / LPR;  // row slot within block
   const int sl = threadIdx.x % LPR;       // lane within row
   const int64_t li = (int64_t)blockIdx.x * RPB + slot_id;
@@ -40,11 +40,38 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
   if (sl == 0) counts[slot_id] = 0;
   __builtin_amdgcn_wave_barrier();
   __threadfence_block();
+  // product-per-lane: tile the A row by LPR entries; an LDS inclusive scan
+  // of the touched B-row sizes lets every lane take one (a,b) product —
+  // no idle lanes when B rows are short/uneven.
+  __shared__ int64_t bstart_s[RPB][LPR];
+  __shared__ int bcnt_s[RPB][LPR + 1];
   const int64_t as = aip[r], ae = aip[r + 1];
-  for (int64_t p = as + sl; p < ae; p += LPR) {
-    const int64_t brow = (int64_t)aix[p] - a_col_lo;
-    const int64_t bs = bip[brow], be = bip[brow + 1];
-    for (int64_t q = bs; q < be; ++q) {
+  for (int64_t t0 = as; t0 < ae; t0 += LPR) {
+    const int na = (int)min((int64_t)LPR, ae - t0);
+    if (sl < na) {
+      const int64_t brow = (int64_t)aix[t0 + sl] - a_col_lo;
+      bstart_s[slot_id][sl] = bip[brow];
+      bcnt_s[slot_id][sl] = (int)(bip[brow + 1] - bip[brow]);
+    }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+    if (sl == 0) {
+      int acc = 0;
+      for (int j2 = 0; j2 < na; ++j2) {
+        int c0 = bcnt_s[slot_id][j2];
+        bcnt_s[slot_id][j2] = acc;  // exclusive prefix
+        acc += c0;
+      }
+      counts[slot_id + RPB] = acc;  // total, stashed past counts[]
+    }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+    const int total = counts[slot_id + RPB];
+    for (int t = sl; t < total; t += LPR) {
+      // find j: largest with prefix[j] <= t (linear ok: na <= LPR)
+      int j = 0;
+      for (int j2 = 1; j2 < na; ++j2) j += (bcnt_s[slot_id][j2] <= t);
+      const int64_t q = bstart_s[slot_id][j] + (t - bcnt_s[slot_id][j]);
       int64_t c = (int64_t)bix[q];
       uint32_t h = hash_mul(c) & (HASH - 1);
       while (true) {
@@ -56,9 +83,9 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
         h = (h + 1) & (HASH - 1);
       }
     }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
   }
-  __builtin_amdgcn_wave_barrier();
-  __threadfence_block();
   if (sl == 0) nnz_out[r] = (int64_t)counts[slot_id];
 }
 
@@ -91,12 +118,38 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
   if (sl == 0) slots[slot_id] = 0;
   __builtin_amdgcn_wave_barrier();
   __threadfence_block();
+  __shared__ int64_t bstart_s[RPB][LPR];
+  __shared__ int bcnt_s[RPB][LPR + 1];
+  __shared__ int tot_s[RPB];
+  __shared__ __align__(16) char aval_raw[RPB * LPR * sizeof(T)];
+  auto aval_s = reinterpret_cast<T(*)[LPR]>(aval_raw);
   const int64_t as = aip[r], ae = aip[r + 1];
-  for (int64_t p = as + sl; p < ae; p += LPR) {
-    const int64_t brow = (int64_t)aix[p] - a_col_lo;
-    const T aval = av[p];
-    const int64_t bs = bip[brow], be = bip[brow + 1];
-    for (int64_t q = bs; q < be; ++q) {
+  for (int64_t t0 = as; t0 < ae; t0 += LPR) {
+    const int na = (int)min((int64_t)LPR, ae - t0);
+    if (sl < na) {
+      const int64_t brow = (int64_t)aix[t0 + sl] - a_col_lo;
+      bstart_s[slot_id][sl] = bip[brow];
+      bcnt_s[slot_id][sl] = (int)(bip[brow + 1] - bip[brow]);
+      aval_s[slot_id][sl] = av[t0 + sl];
+    }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+    if (sl == 0) {
+      int acc = 0;
+      for (int j2 = 0; j2 < na; ++j2) {
+        int c0 = bcnt_s[slot_id][j2];
+        bcnt_s[slot_id][j2] = acc;
+        acc += c0;
+      }
+      tot_s[slot_id] = acc;
+    }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
+    const int total = tot_s[slot_id];
+    for (int t = sl; t < total; t += LPR) {
+      int j = 0;
+      for (int j2 = 1; j2 < na; ++j2) j += (bcnt_s[slot_id][j2] <= t);
+      const int64_t q = bstart_s[slot_id][j] + (t - bcnt_s[slot_id][j]);
       int64_t c = (int64_t)bix[q];
       uint32_t h = hash_mul(c) & (HASH - 1);
       while (true) {
@@ -104,15 +157,15 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
                                 (unsigned long long)(-1ll),
                                 (unsigned long long)c);
         if (old == -1ll || old == c) {
-          atomic_add_any(&accs[slot_id][h], aval * bv[q]);
+          atomic_add_any(&accs[slot_id][h], aval_s[slot_id][j] * bv[q]);
           break;
         }
         h = (h + 1) & (HASH - 1);
       }
     }
+    __builtin_amdgcn_wave_barrier();
+    __threadfence_block();
   }
-  __builtin_amdgcn_wave_barrier();
-  __threadfence_block();
   // compact into [0, count) then LDS bitonic sort by column — emitting the
   // row SORTED skips a global argsort pass over C.
   const int64_t base = cip[r];
