@@ -56,9 +56,13 @@ def spmv_dot(A, x, y, p, dot_out, col_lo: int):
     ext().spmv_dot(A.indptr, A.indices, A.values, x, y, p, dot_out, int(col_lo))
 
 
-def axpby_norm2(y, x, a, b, isalpha, negate, dot_out):
-    """Fused axpby + dot_out += sum(y_new^2) (real dtypes)."""
-    ext().axpby_norm2(y, x, a, b, bool(isalpha), bool(negate), dot_out)
+def axpby_norm2(y, x, a, b, isalpha, negate):
+    """Fused axpby; returns sum(y_new^2) (real dtypes) as a 0-dim tensor."""
+    n = y.numel()
+    blocks = (n // 2 + 255) // 256 + 1
+    partial = torch.empty(blocks, dtype=y.dtype, device=y.device)
+    ext().axpby_norm2(y, x, a, b, bool(isalpha), bool(negate), partial)
+    return partial.sum()
 
 
 def spmm(A, B, C, col_lo: int):

@@ -63,9 +63,8 @@ def _axpby_norm2(y: DistArray, x: DistArray, a, b, negate: bool) -> torch.Tensor
     """Fused y += ±(a/b)x and all-reduced sum(y_new^2) (GPU real dtypes)."""
     from . import kernels
 
-    dot = torch.zeros((), dtype=y.local.dtype, device=y.local.device)
-    kernels.axpby_norm2(y.local, x.local, a.to(y.local.dtype), b.to(y.local.dtype),
-                        True, negate, dot)
+    dot = kernels.axpby_norm2(y.local, x.local, a.to(y.local.dtype),
+                              b.to(y.local.dtype), True, negate)
     comm.all_reduce_(dot)
     return dot
 
