@@ -116,10 +116,31 @@ def main():
 
     extra_samplesort_check()
     extra_halo_check()
+    extra_solver_checks()
 
     if rank == 0:
         print("DIST_ALL_OK")
     dist.destroy_process_group()
+
+
+def extra_solver_checks():
+    """eigsh + solve_ivp under world_size > 1."""
+    import scipy.sparse as sps2
+
+    from sparse import csr_array as _csr, integrate, linalg as _lin
+
+    n = 40
+    m = sps2.random(n, n, 0.3, random_state=9)
+    a = (m + m.T + n * sps2.eye(n)).tocsr()
+    w, V = _lin.eigsh(_csr(a), k=2)
+    ws_ = np.linalg.eigvalsh(a.toarray())
+    assert np.allclose(np.sort(np.abs(w)), np.sort(np.abs(ws_))[-2:], atol=1e-3), "dist eigsh"
+
+    y0 = np.linspace(1, 2, 37)
+    res = integrate.solve_ivp(lambda t, y: y * (-0.5), (0, 2.0), y0,
+                              rtol=1e-8, atol=1e-10)
+    assert res.success
+    assert np.allclose(res.y[:, -1], y0 * np.exp(-1.0), rtol=1e-6), "dist ivp"
 
 
 def extra_halo_check():
