@@ -77,7 +77,11 @@ class GMG:
             # materialize R row-partitioned: restriction becomes a row-split
             # ELL SpMV (no atomic scatter) at the cost of one transposed copy
             R = P.T.tocsr()
-            Ac = (R @ cur) @ P  # Galerkin triple product (distributed SpGEMM)
+            RA = R @ cur
+            Ac = RA @ P  # Galerkin triple product (distributed SpGEMM)
+            del RA
+            if torch.cuda.is_available():
+                torch.cuda.empty_cache()  # large-level intermediates
             self.levels[-1].Pdown = P
             self.levels[-1].Rdown = R
             cur = Ac

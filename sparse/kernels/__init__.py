@@ -281,6 +281,11 @@ def build_ell(A):
     mp = (m + 1) // 2 * 2
     if W * mp > 1.6 * A.nnz + 4096:
         return None
+    need = W * mp * (A.values.element_size() + A.indices.element_size())
+    free, _total = torch.cuda.mem_get_info(A.values.device)
+    if need > 0.5 * free:
+        # mirror would not fit comfortably — stay on the CSR kernels
+        return None
     eidx = torch.empty(W * mp, dtype=A.indices.dtype, device=A.device)
     evals = torch.empty(W * mp, dtype=A.values.dtype, device=A.device)
     pad_idx = int(A.indices[0].item())
