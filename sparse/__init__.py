@@ -5,7 +5,6 @@ sparse/__init__.py) backed by PyTorch-ROCm tensors, hand-written HIP/CDNA4
 (gfx950) kernels and RCCL collectives over xGMI — one process per GPU.
 """
 from .module import *  # noqa: F401,F403
-from .module import __all__ as _module_all
 from .coverage import clone_module, track_provenance  # noqa: F401
 
 from .csr import csr_array, csr_matrix  # noqa: F401

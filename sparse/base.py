@@ -15,7 +15,6 @@ import torch
 from .parallel import comm
 from .parallel.partition import RowPartition
 from .runtime import runtime
-from .types import to_torch_dtype
 
 
 class CompressedBase:

@@ -18,7 +18,7 @@ import scipy.sparse as _sps
 
 from .coverage import clone_scipy_arr_kind
 from .base import CompressedBase
-from .darray import DistArray, asdistarray
+from .darray import DistArray
 from .ops import local as ops
 from .parallel import comm
 from .parallel.partition import RowPartition

@@ -9,7 +9,7 @@ from __future__ import annotations
 
 from functools import wraps
 from types import FunctionType, MethodType, ModuleType
-from typing import Any, Callable
+from typing import Callable
 
 _PROFILING = False
 _STACK = []

@@ -20,7 +20,6 @@ import scipy.sparse as _sps
 
 from .coverage import clone_scipy_arr_kind
 from .base import CompressedBase
-from .parallel import comm
 from .runtime import runtime
 from .types import promote_value_dtype, to_numpy_dtype, to_torch_dtype
 

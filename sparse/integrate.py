@@ -12,8 +12,6 @@ State vectors are DistArrays (row-partitioned); stage matrices K are local
 from __future__ import annotations
 
 import math
-from typing import Callable, List, Optional
-
 import numpy as np
 import torch
 

@@ -10,7 +10,6 @@ missing.  There is no eager/PyTorch fallback on the GPU path.
 from __future__ import annotations
 
 import os
-from typing import Optional
 
 import torch
 

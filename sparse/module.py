@@ -12,9 +12,6 @@ the banded-SpMV benchmark uses at n=10M+ rows/GPU.
 """
 from __future__ import annotations
 
-import numbers
-from typing import Optional
-
 import numpy as np
 import torch
 

@@ -11,8 +11,6 @@ in the reference; the window plan is what its default build uses.)
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from . import comm

@@ -12,8 +12,6 @@ cases (COO->CSR with an equal row tiling).
 """
 from __future__ import annotations
 
-from typing import Tuple
-
 import torch
 
 from . import comm

@@ -18,7 +18,7 @@ flip(repeat(levels, poly))).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List
 
 import numpy as np
 
