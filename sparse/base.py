@@ -154,7 +154,28 @@ class DenseSparseBase:
     @classmethod
     def make_with_same_nnz_structure(cls, mat, arg, shape=None, dtype=None):
         """New array sharing mat's structure tensors with new values
-        (reference base.py:284-296)."""
-        if shape is None:
-            shape = mat.shape
-        return cls(arg, shape=shape, dtype=dtype, _partition=mat.partition)
+        (reference base.py:284-296).  arg is the new values: a global numpy
+        array (sliced to this rank's nnz), a local torch tensor, or the
+        reference-style (vals, crd, pos) tuple (only vals is taken — the
+        structure comes from mat)."""
+        import numpy as _np
+
+        vals = arg[0] if isinstance(arg, tuple) else arg
+        old = mat._values_tensor()
+        if isinstance(vals, torch.Tensor) and vals.numel() == old.numel():
+            local = vals.to(old.device)
+        else:
+            a = _np.asarray(vals)
+            counts = mat._nnz_counts()
+            off = sum(counts[: comm.rank()])
+            local = torch.as_tensor(
+                _np.ascontiguousarray(a[off: off + counts[comm.rank()]]),
+                device=old.device)
+        if dtype is not None:
+            from .types import to_torch_dtype
+
+            local = local.to(to_torch_dtype(dtype))
+        out = mat._with_values(lambda _v: local)
+        if shape is not None and tuple(shape) != tuple(mat.shape):
+            out.shape = tuple(int(x) for x in shape)
+        return out

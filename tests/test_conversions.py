@@ -154,3 +154,14 @@ def test_jacobi_smooth_cpu():
     r = b - A.dot(x)
     expect = np.asarray(x) + 0.7 * np.asarray(dinv) * np.asarray(r)
     assert np.allclose(np.asarray(out), expect, rtol=1e-12)
+
+
+def test_make_with_same_nnz_structure():
+    s = sample_csr(9, 11, 0.4, seed=70)
+    A = csr_array(s)
+    newvals = np.arange(1.0, s.nnz + 1.0)
+    B = csr_array.make_with_same_nnz_structure(A, newvals)
+    ref = s.copy()
+    ref.data = newvals
+    assert np.allclose(np.asarray(B.todense()), ref.toarray())
+    assert B._indices is A._indices  # structure shared
