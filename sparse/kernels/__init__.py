@@ -156,7 +156,8 @@ def spgemm_csr(A, B, a_col_lo, vdt):
     ub.index_add_(0, rows_of_nnz, bcounts[acols])
     counts = torch.zeros(m, dtype=torch.int64, device=dev)
     bins = [((ub <= 32), 64), ((ub > 32) & (ub <= 128), 256),
-            ((ub > 128) & (ub <= 512), 1024)]
+            ((ub > 128) & (ub <= 512), 1024),
+            ((ub > 512) & (ub <= 1024), 2048)]
     rowlists = [mask.nonzero(as_tuple=False).flatten() for mask, _ in bins]
     Av = A.values.to(vdt)
     Bv = B.values.to(vdt)
@@ -164,7 +165,7 @@ def spgemm_csr(A, B, a_col_lo, vdt):
         if rl.numel():
             ext().spgemm_nnz(A.indptr, A.indices, B.indptr, B.indices, rl,
                              counts, int(a_col_lo), H)
-    esc_rows = (ub > 512).nonzero(as_tuple=False).flatten()
+    esc_rows = (ub > 1024).nonzero(as_tuple=False).flatten()
     esc_sub = None
     if esc_rows.numel():
         sc = acounts[esc_rows]

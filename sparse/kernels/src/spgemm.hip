@@ -21,14 +21,15 @@ __device__ __forceinline__ uint32_t hash_mul(int64_t c) {
   return (uint32_t)((uint64_t)c * 2654435761u);
 }
 
-// LPR = lanes per row (16 or 64); rows per block = WAVES_PER_BLOCK*(WAVE/LPR)
-template <typename index_t, int HASH, int LPR>
+// LPR = lanes per row (16 or 64); WPB = waves per block;
+// rows per block = WPB*(WAVE/LPR)
+template <typename index_t, int HASH, int LPR, int WPB = WAVES_PER_BLOCK>
 __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
     const int64_t* __restrict__ aip, const index_t* __restrict__ aix,
     const int64_t* __restrict__ bip, const index_t* __restrict__ bix,
     const int64_t* __restrict__ rowlist, int64_t nrows,
     int64_t* __restrict__ nnz_out, int64_t a_col_lo) {
-  constexpr int RPB = WAVES_PER_BLOCK * (WAVE / LPR);
+  constexpr int RPB = WPB * (WAVE / LPR);
   __shared__ int64_t keys[RPB][HASH];
   __shared__ int counts[2 * RPB];  // [0,RPB): distinct-col counts; [RPB,2RPB): scan totals
   const int slot_id = threadIdx.x / LPR;  // row slot within block
@@ -89,7 +90,8 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
   if (sl == 0) nnz_out[r] = (int64_t)counts[slot_id];
 }
 
-template <typename T, typename index_t, int HASH, int LPR>
+template <typename T, typename index_t, int HASH, int LPR,
+          int WPB = WAVES_PER_BLOCK>
 __global__ __launch_bounds__(256) void spgemm_compute_kernel(
     const int64_t* __restrict__ aip, const index_t* __restrict__ aix,
     const T* __restrict__ av, const int64_t* __restrict__ bip,
@@ -97,7 +99,7 @@ __global__ __launch_bounds__(256) void spgemm_compute_kernel(
     const int64_t* __restrict__ rowlist, int64_t nrows,
     const int64_t* __restrict__ cip, index_t* __restrict__ cix,
     T* __restrict__ cv, int64_t a_col_lo) {
-  constexpr int RPB = WAVES_PER_BLOCK * (WAVE / LPR);
+  constexpr int RPB = WPB * (WAVE / LPR);
   __shared__ int64_t keys[RPB][HASH];
   __shared__ __align__(16) char accs_raw[RPB * HASH * sizeof(T)];
   auto accs = reinterpret_cast<T(*)[HASH]>(accs_raw);
@@ -220,16 +222,17 @@ void spgemm_nnz_hip(at::Tensor aip, at::Tensor aix, at::Tensor bip,
   int64_t nrows = rowlist.numel();
   if (nrows == 0) return;
   DISPATCH_INDEX(aix.scalar_type(), "spgemm_nnz", [&] {
-    auto launch = [&](auto kern, int rpb) {
-      hipLaunchKernelGGL(kern, dim3((nrows + rpb - 1) / rpb), dim3(256), 0,
+    auto launch = [&](auto kern, int rpb, int threads) {
+      hipLaunchKernelGGL(kern, dim3((nrows + rpb - 1) / rpb), dim3(threads), 0,
                          cur_stream(), aip.data_ptr<int64_t>(),
                          aix.data_ptr<index_t>(), bip.data_ptr<int64_t>(),
                          bix.data_ptr<index_t>(), rowlist.data_ptr<int64_t>(),
                          nrows, nnz_out.data_ptr<int64_t>(), a_col_lo);
     };
-    if (hash_size <= 64) launch(spgemm_nnz_kernel<index_t, 64, 16>, 16);
-    else if (hash_size <= 256) launch(spgemm_nnz_kernel<index_t, 256, 64>, 4);
-    else launch(spgemm_nnz_kernel<index_t, 1024, 64>, 4);
+    if (hash_size <= 64) launch(spgemm_nnz_kernel<index_t, 64, 16>, 16, 256);
+    else if (hash_size <= 256) launch(spgemm_nnz_kernel<index_t, 256, 64>, 4, 256);
+    else if (hash_size <= 1024) launch(spgemm_nnz_kernel<index_t, 1024, 64>, 4, 256);
+    else launch(spgemm_nnz_kernel<index_t, 2048, 64, 2>, 2, 128);
   });
 }
 
@@ -242,8 +245,8 @@ void spgemm_compute_hip(at::Tensor aip, at::Tensor aix, at::Tensor av,
   DISPATCH_VALUES(cv.scalar_type(), "spgemm_compute", [&] {
     using T = scalar_t;
     DISPATCH_INDEX(aix.scalar_type(), "spgemm_compute_idx", [&] {
-      auto launch = [&](auto kern, int rpb) {
-        hipLaunchKernelGGL(kern, dim3((nrows + rpb - 1) / rpb), dim3(256), 0,
+      auto launch = [&](auto kern, int rpb, int threads) {
+        hipLaunchKernelGGL(kern, dim3((nrows + rpb - 1) / rpb), dim3(threads), 0,
                            cur_stream(), aip.data_ptr<int64_t>(),
                            aix.data_ptr<index_t>(), av.data_ptr<T>(),
                            bip.data_ptr<int64_t>(), bix.data_ptr<index_t>(),
@@ -252,11 +255,13 @@ void spgemm_compute_hip(at::Tensor aip, at::Tensor aix, at::Tensor av,
                            cix.data_ptr<index_t>(), cv.data_ptr<T>(), a_col_lo);
       };
       if (hash_size <= 64) {
-        launch(spgemm_compute_kernel<T, index_t, 64, 16>, 16);
+        launch(spgemm_compute_kernel<T, index_t, 64, 16>, 16, 256);
       } else if (hash_size <= 256) {
-        launch(spgemm_compute_kernel<T, index_t, 256, 64>, 4);
+        launch(spgemm_compute_kernel<T, index_t, 256, 64>, 4, 256);
+      } else if (hash_size <= 1024) {
+        launch(spgemm_compute_kernel<T, index_t, 1024, 64>, 4, 256);
       } else {
-        launch(spgemm_compute_kernel<T, index_t, 1024, 64>, 4);
+        launch(spgemm_compute_kernel<T, index_t, 2048, 64, 2>, 2, 128);
       }
     });
   });
