@@ -128,13 +128,17 @@ class _SparseMatrixLinearOperator(LinearOperator):
 
     def __init__(self, A):
         self.A = A
+        self._AH = None  # adjoint, built lazily (conjugate for complex)
         super().__init__(A.shape, dtype=A.dtype)
 
     def matvec(self, x, out=None):
         return self.A.dot(asdistarray(x), out=out)
 
     def rmatvec(self, x, out=None):
-        return self.A.T.dot(asdistarray(x), out=out)
+        if self._AH is None:
+            self._AH = (self.A.conj().T if np.issubdtype(self.A.dtype,
+                        np.complexfloating) else self.A.T)
+        return self._AH.dot(asdistarray(x), out=out)
 
 
 def aslinearoperator(A) -> LinearOperator:

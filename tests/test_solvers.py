@@ -133,3 +133,29 @@ def test_eigsh(which):
     for i in range(k):
         v = V[:, i]
         assert np.linalg.norm(s @ v - w[i] * v) < 1e-3 * max(1, abs(w[i]))
+
+
+def test_bicg_complex():
+    """Complex nonsymmetric system: rmatvec must be the CONJUGATE transpose
+    (scipy semantics)."""
+    rng = np.random.default_rng(30)
+    n = 30
+    s = sps.random(n, n, 0.3, random_state=31).astype(np.complex128)
+    s.data = s.data + 1j * rng.random(len(s.data))
+    s = (s + n * sps.eye(n)).tocsr()
+    b = rng.random(n) + 1j * rng.random(n)
+    x, info = linalg.bicg(csr_array(s), b, tol=1e-10, maxiter=400,
+                          conv_test_iters=2)
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-5)
+
+
+def test_lsqr_complex_rmatvec():
+    rng = np.random.default_rng(32)
+    m, n = 25, 10
+    s = sps.random(m, n, 0.5, random_state=33).astype(np.complex128)
+    s.data = s.data + 1j * rng.random(len(s.data))
+    s = s.tocsr()
+    A = csr_array(s)
+    x = rng.random(m) + 1j * rng.random(m)
+    op = linalg.aslinearoperator(A)
+    assert np.allclose(np.asarray(op.rmatvec(x)), s.conj().T @ x)
