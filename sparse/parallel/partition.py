@@ -28,11 +28,21 @@ class RowPartition:
     @staticmethod
     def equal(n: int, world_size: int) -> "RowPartition":
         # Equal tiles, remainder spread over the first ranks (matches the
-        # reference runtime's default tiling semantics).
-        base, rem = divmod(n, world_size)
+        # reference runtime's default tiling semantics).  SPARSE_NUM_PROCS
+        # caps how many ranks own rows (reference LEGATE_SPARSE_NUM_PROCS);
+        # later ranks get empty slabs but still join collectives.
+        from ..settings import settings
+
+        owners = world_size
+        if settings.num_procs:
+            owners = max(1, min(world_size, settings.num_procs))
+        base, rem = divmod(n, owners)
         starts = [0]
         for r in range(world_size):
-            starts.append(starts[-1] + base + (1 if r < rem else 0))
+            if r < owners:
+                starts.append(starts[-1] + base + (1 if r < rem else 0))
+            else:
+                starts.append(starts[-1])
         return RowPartition(n, tuple(starts))
 
     @staticmethod

@@ -117,6 +117,7 @@ def main():
     extra_samplesort_check()
     extra_halo_check()
     extra_solver_checks()
+    extra_num_procs_check()
 
     if rank == 0:
         print("DIST_ALL_OK")
@@ -197,3 +198,26 @@ def extra_samplesort_check():
 
 if __name__ == "__main__":
     main()
+
+
+def extra_num_procs_check():
+    """SPARSE_NUM_PROCS=1: rank 0 owns everything, others hold empty slabs;
+    ops still agree with the oracle."""
+    import importlib
+
+    import sparse.settings as st
+
+    old = st.settings.num_procs
+    st.settings.num_procs = 1
+    try:
+        import scipy.sparse as sps3
+
+        from sparse import csr_array as _csr
+
+        s = sps3.random(19, 17, 0.3, random_state=42, format="csr")
+        A = _csr(s)
+        x = np.random.default_rng(1).random(17)
+        assert np.allclose(np.asarray(A @ x), s @ x), "num_procs spmv"
+        assert np.allclose(np.asarray((A + A).todense()), (s + s).toarray()), "num_procs add"
+    finally:
+        st.settings.num_procs = old
