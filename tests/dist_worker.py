@@ -196,10 +196,6 @@ def extra_samplesort_check():
     assert np.isclose(gv.sum(), allv.sum()), "payload preserved"
 
 
-if __name__ == "__main__":
-    main()
-
-
 def extra_num_procs_check():
     """SPARSE_NUM_PROCS=1: rank 0 owns everything, others hold empty slabs;
     ops still agree with the oracle."""
@@ -221,3 +217,7 @@ def extra_num_procs_check():
         assert np.allclose(np.asarray((A + A).todense()), (s + s).toarray()), "num_procs add"
     finally:
         st.settings.num_procs = old
+
+
+if __name__ == "__main__":
+    main()
