@@ -189,6 +189,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         self._window_cache = None
         self._plan_cache = {}
         self._ell_cache = None
+        self._dia_cache = None
         self._maxrow_cache = None
 
     @classmethod
@@ -220,6 +221,19 @@ class csr_array(CompressedBase, DenseSparseBase):
                 self._window_cache = (int(self._indices.min().item()),
                                       int(self._indices.max().item()) + 1)
         return self._window_cache
+
+    def _dia(self):
+        """Cached diagonal mirror (GPU banded fast SpMV; kernels.build_dia —
+        values only, no index stream)."""
+        if not self._values.is_cuda or self._dia_cache == "no":
+            return None
+        if self._dia_cache is None:
+            from . import kernels
+
+            kernels.require()
+            dm = kernels.build_dia(self.local, self.partition.start(comm.rank()))
+            self._dia_cache = dm or "no"
+        return None if self._dia_cache == "no" else self._dia_cache
 
     def _ell(self):
         """Cached padded-ELL mirror (GPU fast SpMV; kernels.build_ell)."""
@@ -263,6 +277,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         mine = np.asarray(v)[off: off + counts[comm.rank()]]
         self._values = torch.as_tensor(mine, device=self._values.device).to(self._values.dtype)
         self._ell_cache = None
+        self._dia_cache = None
 
     @property
     def indices(self) -> np.ndarray:
@@ -389,8 +404,17 @@ class csr_array(CompressedBase, DenseSparseBase):
             raise ValueError(f"dimension mismatch {self.shape} @ {x.shape}")
         plan = self._xplan(x.partition)
         vdt = self._out_dtype(x.local.dtype)
-        ell = self._ell() if self._values.dtype == vdt else None
-        if ell is not None:
+        dm = self._dia() if self._values.dtype == vdt else None
+        ell = None if dm is not None else (
+            self._ell() if self._values.dtype == vdt else None)
+        if dm is not None:
+            from . import kernels
+
+            pieces = plan.gather_halos(x.local.to(vdt))
+            ylocal = torch.empty(self.partition.count(comm.rank()), dtype=vdt,
+                                 device=self._values.device)
+            kernels.dia_spmv(dm, pieces, ylocal, plan.lo, plan.hi - plan.lo)
+        elif ell is not None:
             from . import kernels
 
             pieces = plan.gather_halos(x.local.to(vdt))
@@ -431,10 +455,19 @@ class csr_array(CompressedBase, DenseSparseBase):
         """One fused weighted-Jacobi sweep x' = x + omega*dinv*(b - A x)
         (the GMG/AMG smoother; reference WeightedJacobi gmg.py:247-285).
         Single kernel on the ELL fast path; generic fallback otherwise."""
-        ell = self._ell()
+        dm = self._dia()
+        ell = None if dm is not None else self._ell()
         if out is None:
             out = DistArray.from_local(torch.empty_like(x.local), x.partition,
                                        x.gshape)
+        if dm is not None:
+            from . import kernels
+
+            plan = self._xplan(x.partition)
+            pieces = plan.gather_halos(x.local)
+            kernels.dia_jacobi(dm, pieces, x.local, b.local, dinv.local,
+                               omega, out.local, plan.lo, plan.hi - plan.lo)
+            return out
         if ell is not None:
             from . import kernels
 
@@ -455,8 +488,13 @@ class csr_array(CompressedBase, DenseSparseBase):
         from . import kernels
 
         plan = self._xplan(p.partition)
-        ell = self._ell()
-        if ell is not None:
+        dm = self._dia()
+        ell = None if dm is not None else self._ell()
+        if dm is not None:
+            pieces = plan.gather_halos(p.local)
+            dot = kernels.dia_spmv_dot(dm, pieces, q.local, p.local, plan.lo,
+                                       plan.hi - plan.lo)
+        elif ell is not None:
             pieces = plan.gather_halos(p.local)
             dot = kernels.ell_spmv_dot(ell, pieces, q.local, p.local, plan.lo)
         else:

@@ -25,6 +25,16 @@ void ell_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
 void ell_jacobi_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, int64_t, int64_t, int64_t, double);
+void dia_spmv_plain_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                        at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
+                        int64_t, int64_t);
+void dia_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      int64_t, int64_t, int64_t, int64_t, int64_t);
+void dia_jacobi_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                    at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                    at::Tensor, int64_t, int64_t, int64_t, int64_t, int64_t,
+                    double);
 void add_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void add_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                      at::Tensor, at::Tensor, at::Tensor, at::Tensor, double,
@@ -87,6 +97,15 @@ TORCH_LIBRARY(sparse_hip, m) {
   m.def("ell_jacobi(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor xloc, Tensor b, Tensor dinv, Tensor(a!) xout, "
         "int W, int m, int col_lo, float omega) -> ()");
+  m.def("dia_spmv(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
+        "Tensor hhi, Tensor(a!) y, int W, int m, int col_lo, int row0, "
+        "int wsize) -> ()");
+  m.def("dia_spmv_dot(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
+        "Tensor hhi, Tensor(a!) y, Tensor pvec, Tensor(b!) dot_partial, "
+        "int W, int m, int col_lo, int row0, int wsize) -> ()");
+  m.def("dia_jacobi(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
+        "Tensor hhi, Tensor xloc, Tensor b, Tensor dinv, Tensor(a!) xout, "
+        "int W, int m, int col_lo, int row0, int wsize, float omega) -> ()");
   m.def("spmm(Tensor indptr, Tensor indices, Tensor vals, Tensor B, "
         "Tensor(a!) C, int col_lo) -> ()");
   m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "
@@ -127,6 +146,9 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("ell_spmv", ell_spmv_plain_hip);
   m.impl("ell_spmv_dot", ell_spmv_dot_hip);
   m.impl("ell_jacobi", ell_jacobi_hip);
+  m.impl("dia_spmv", dia_spmv_plain_hip);
+  m.impl("dia_spmv_dot", dia_spmv_dot_hip);
+  m.impl("dia_jacobi", dia_jacobi_hip);
   m.impl("spmm", spmm_hip);
   m.impl("rspmm", rspmm_hip);
   m.impl("sddmm", sddmm_hip);
