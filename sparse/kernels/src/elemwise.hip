@@ -299,7 +299,13 @@ void axpby_norm2_hip(at::Tensor y, at::Tensor x, at::Tensor a, at::Tensor b,
                      bool isalpha, bool negate, at::Tensor dot_out) {
   int64_t n = y.numel();
   if (n == 0) return;
-  int64_t blocks = std::min<int64_t>((n + 255) / 256, 1024);
+  // flat pair-per-thread, same grid as axpby_hip: ceil((n/2)/256) + 1 (the
+  // +1 block only contributes a zero partial).  NO cap: a capped grid left
+  // elements beyond blocks*512 untouched (silent corruption at n > 512K).
+  int64_t blocks = (n / 2 + 255) / 256 + 1;
+  TORCH_CHECK(dot_out.numel() >= blocks,
+              "axpby_norm2: dot_out too small (", dot_out.numel(), " < ",
+              blocks, ")");
   AT_DISPATCH_FLOATING_TYPES(y.scalar_type(), "axpby_norm2", [&] {
     using T = scalar_t;
     auto launch = [&](auto kern) {

@@ -264,11 +264,14 @@ def test_spmv_dot_fused_gpu():
     assert np.isclose(float(pq.item()), float(np.asarray(p) @ qref), rtol=1e-10)
 
 
-def test_axpby_norm2_gpu():
+@pytest.mark.parametrize("n", [50000, 3_000_001])
+def test_axpby_norm2_gpu(n):
+    """n=3M+1 regression: a capped launch grid once left elements beyond
+    512K untouched (silent corruption in the fused CG loop); odd n also
+    exercises the scalar tail."""
     from sparse import darray
     from sparse.linalg import _axpby_norm2
 
-    n = 50000
     y = darray.random((n,), seed=35)
     x = darray.random((n,), seed=36)
     y0 = np.asarray(y).copy()
