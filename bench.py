@@ -94,33 +94,87 @@ def main() -> None:
 
     graph = None
     if on_gpu:
-        rz_buf = rz.clone()
+        from sparse import kernels
 
-        def cg_step_gpu() -> None:  # fully-fused MI355X path (see linalg.cg)
-            # rz_buf is a persistent 0-dim buffer so the step is
-            # hipGraph-capturable (fixed addresses across replays)
-            pq = A.spmv_dot(pvec, q)
-            linalg.cg_axpby(xv, pvec, rz_buf, pq, isalpha=True, negate=False)
-            rz_new = linalg._axpby_norm2(r, q, rz_buf, pq, negate=True)
-            linalg.cg_axpby(pvec, r, rz_new, rz_buf, isalpha=False, negate=False)
-            rz_buf.copy_(rz_new)
+        rz_buf = rz.clone()
+        use2 = A._dia() is not None  # two-kernel CG iteration (DIA fast path)
+        if use2:
+            # K1 folds p = r + beta*p into the SpMV (q = Ap, p.q fused);
+            # K2 fuses x += alpha p, r -= alpha q and |r|^2 — two HBM
+            # passes per iteration fewer than the 4-kernel loop.
+            rz_old_buf = rz.clone()
+            p_b = darray.zeros((n,), dtype=dtype)
+
+            def prologue() -> None:  # iteration 0: p is r, beta undefined
+                pq = A.spmv_dot(pvec, q)
+                rz_new = kernels.cg_xr_norm2(xv.local, pvec.local, r.local,
+                                             q.local, rz_buf, pq)
+                comm.all_reduce_(rz_new)
+                rz_old_buf.copy_(rz_buf)
+                rz_buf.copy_(rz_new)
+
+            def step2(pc, pn) -> None:
+                pq = A.spmv_bpdot(r, pc, pn, q, rz_buf, rz_old_buf)
+                rz_new = kernels.cg_xr_norm2(xv.local, pn.local, r.local,
+                                             q.local, rz_buf, pq)
+                comm.all_reduce_(rz_new)
+                rz_old_buf.copy_(rz_buf)
+                rz_buf.copy_(rz_new)
+
+            prologue()
+            _parity = [0]
+
+            def cg_step_gpu() -> None:
+                if _parity[0] == 0:
+                    step2(pvec, p_b)
+                else:
+                    step2(p_b, pvec)
+                _parity[0] ^= 1
+
+        else:
+
+            def cg_step_gpu() -> None:  # fused 4-kernel path (see linalg.cg)
+                # rz_buf is a persistent 0-dim buffer so the step is
+                # hipGraph-capturable (fixed addresses across replays)
+                pq = A.spmv_dot(pvec, q)
+                linalg.cg_axpby(xv, pvec, rz_buf, pq, isalpha=True, negate=False)
+                rz_new = linalg._axpby_norm2(r, q, rz_buf, pq, negate=True)
+                linalg.cg_axpby(pvec, r, rz_new, rz_buf, isalpha=False, negate=False)
+                rz_buf.copy_(rz_new)
 
         cg_step = cg_step_gpu
         if args.gpus == 1 and not os.environ.get("SPARSE_NO_HIPGRAPH"):
-            # capture one CG iteration as a single hipGraph (launch-overhead
-            # free replay); multi-GPU stays eager (RCCL outside graphs)
+            # capture CG iterations as hipGraphs (launch-overhead free
+            # replay); multi-GPU stays eager (RCCL outside graphs).  The
+            # two-kernel path double-buffers p, so capture BOTH parities
+            # and alternate replays.
             try:
                 side = torch.cuda.Stream()
                 side.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(side):
-                    for _ in range(3):
+                    for _ in range(4):
                         cg_step_gpu()
                 torch.cuda.current_stream().wait_stream(side)
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
-                    cg_step_gpu()
-                graph = g
-                cg_step = graph.replay
+                if use2:
+                    _parity[0] = 0
+                    g_ab = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g_ab):
+                        step2(pvec, p_b)
+                    g_ba = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g_ba):
+                        step2(p_b, pvec)
+                    graph = (g_ab, g_ba)
+
+                    def cg_step() -> None:
+                        graph[_parity[0]].replay()
+                        _parity[0] ^= 1
+
+                else:
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        cg_step_gpu()
+                    graph = g
+                    cg_step = graph.replay
             except Exception as e:
                 print(f"# hipGraph capture unavailable ({e}); eager steps",
                       file=sys.stderr)

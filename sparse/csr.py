@@ -505,6 +505,30 @@ class csr_array(CompressedBase, DenseSparseBase):
         comm.all_reduce_(dot)
         return dot
 
+    def spmv_bpdot(self, r: DistArray, p_old: DistArray, p_new: DistArray,
+                   q: DistArray, beta_num: torch.Tensor,
+                   beta_den: torch.Tensor):
+        """Fused CG K1 (DIA fast path only): p_new = r + β p_old with
+        β = beta_num/beta_den read on device, q = A @ p_new, returns the
+        all-reduced p_new·q.  Returns None when this matrix has no DIA
+        mirror (caller falls back to the 3-kernel loop).  p_new must not
+        alias p_old (double-buffered by the caller).  This is the MI355X
+        two-kernel CG iteration: the separate p-update HBM pass disappears
+        (reference CG task chain, linalg.py:499-565)."""
+        dm = self._dia()
+        if dm is None:
+            return None
+        from . import kernels
+
+        plan = self._xplan(r.partition)
+        r_pieces = plan.gather_halos(r.local)
+        p_pieces = plan.gather_halos(p_old.local)
+        dot = kernels.dia_spmv_bpdot(dm, r_pieces, p_pieces, p_new.local,
+                                     q.local, beta_num, beta_den, plan.lo,
+                                     plan.hi - plan.lo)
+        comm.all_reduce_(dot)
+        return dot
+
     def _rspmm(self, A: DistArray) -> DistArray:
         # C = A(dense k x m) @ self(m x n): replicate A, local partial with my
         # row slab of B, ADD all-reduce (reference csr.py:1209-1240 semantics).

@@ -313,6 +313,34 @@ def dia_spmv_dot(dm: DiaMirror, pieces, y, p, col_lo: int, wsize: int):
     return partial.sum()
 
 
+def dia_spmv_bpdot(dm: DiaMirror, r_pieces, p_pieces, pnew, q,
+                   beta_num, beta_den, col_lo: int, wsize: int):
+    """Fused CG K1: pnew = r + (beta_num/beta_den)*p_old; q = A@pnew;
+    returns p.q (device 0-dim, local partial-sum).  r_pieces/p_pieces are
+    the (hlo, own, hhi) window pieces of r and p_old; pnew must be a
+    distinct buffer from p_old (double-buffered caller)."""
+    rlo, rown, rhi = r_pieces
+    plo, pown, phi = p_pieces
+    mp = dm.dvals.numel() // dm.W
+    nblocks = (mp // 2 + 255) // 256
+    partial = torch.empty(nblocks, dtype=dm.dvals.dtype, device=dm.dvals.device)
+    ext().dia_spmv_bpdot(dm.dvals, dm.offs, rlo.contiguous(), rown.contiguous(),
+                         rhi.contiguous(), plo.contiguous(), pown.contiguous(),
+                         phi.contiguous(), pnew, q, beta_num, beta_den,
+                         partial, dm.W, dm.m, int(col_lo), dm.row0, int(wsize))
+    return partial.sum()
+
+
+def cg_xr_norm2(x, p, r, q, a, b):
+    """Fused CG K2: x += (a/b)p; r -= (a/b)q; returns local sum(r_new^2)
+    as a device 0-dim tensor."""
+    n = x.numel()
+    blocks = (n // 2 + 255) // 256 + 1
+    partial = torch.empty(blocks, dtype=x.dtype, device=x.device)
+    ext().cg_xr_norm2(x, p, r, q, a.to(x.dtype), b.to(x.dtype), partial)
+    return partial.sum()
+
+
 def dia_jacobi(dm: DiaMirror, pieces, xloc, b, dinv, omega, xout,
                col_lo: int, wsize: int):
     hlo, own, hhi = pieces

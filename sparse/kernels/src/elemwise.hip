@@ -293,7 +293,71 @@ __global__ __launch_bounds__(256) void axpby_norm2_kernel(
   if (threadIdx.x == 0) dot_partial[blockIdx.x] = red[0];
 }
 
+// CG K2: x += (a/b) p ; r -= (a/b) q ; per-block partials of sum(r_new^2).
+// One pass over 4 streams (reads x,p,r,q; writes x,r) instead of the
+// separate x-axpby + r-axpby_norm2 pair — saves 2 full HBM passes per CG
+// iteration at n=268M.
+template <typename T>
+__global__ __launch_bounds__(256) void cg_xr_norm2_kernel(
+    T* __restrict__ x, const T* __restrict__ p, T* __restrict__ r,
+    const T* __restrict__ q, const T* __restrict__ a, const T* __restrict__ b,
+    T* __restrict__ dot_partial, int64_t n) {
+  __shared__ T red[256];
+  const T s = (*a) / (*b);
+  T acc = T(0);
+  const int64_t half = n / 2;
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < half) {
+    auto* x2 = reinterpret_cast<VPair<T>*>(x);
+    auto* r2 = reinterpret_cast<VPair<T>*>(r);
+    auto* p2 = reinterpret_cast<const VPair<T>*>(p);
+    auto* q2 = reinterpret_cast<const VPair<T>*>(q);
+    VPair<T> xv = x2[i];
+    VPair<T> rv = r2[i];
+    const VPair<T> pv = p2[i];
+    const VPair<T> qv = q2[i];
+    xv.a += s * pv.a;
+    xv.b += s * pv.b;
+    rv.a -= s * qv.a;
+    rv.b -= s * qv.b;
+    x2[i] = xv;
+    r2[i] = rv;
+    acc = rv.a * rv.a + rv.b * rv.b;
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
+    const int64_t t = n - 1;
+    x[t] += s * p[t];
+    const T rv = r[t] - s * q[t];
+    r[t] = rv;
+    acc += rv * rv;
+  }
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if ((int)threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) dot_partial[blockIdx.x] = red[0];
+}
+
 }  // namespace
+
+void cg_xr_norm2_hip(at::Tensor x, at::Tensor p, at::Tensor r, at::Tensor q,
+                     at::Tensor a, at::Tensor b, at::Tensor dot_out) {
+  int64_t n = x.numel();
+  if (n == 0) return;
+  int64_t blocks = (n / 2 + 255) / 256 + 1;
+  TORCH_CHECK(dot_out.numel() >= blocks,
+              "cg_xr_norm2: dot_out too small (", dot_out.numel(), " < ",
+              blocks, ")");
+  AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "cg_xr_norm2", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(cg_xr_norm2_kernel<T>, dim3(blocks), dim3(256), 0,
+                       cur_stream(), x.data_ptr<T>(), p.data_ptr<T>(),
+                       r.data_ptr<T>(), q.data_ptr<T>(), a.data_ptr<T>(),
+                       b.data_ptr<T>(), dot_out.data_ptr<T>(), n);
+  });
+}
 
 void axpby_norm2_hip(at::Tensor y, at::Tensor x, at::Tensor a, at::Tensor b,
                      bool isalpha, bool negate, at::Tensor dot_out) {

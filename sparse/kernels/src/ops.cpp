@@ -25,6 +25,13 @@ void ell_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
 void ell_jacobi_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, int64_t, int64_t, int64_t, double);
+void cg_xr_norm2_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                     at::Tensor, at::Tensor, at::Tensor);
+void dia_spmv_bpdot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                        at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                        at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                        at::Tensor, int64_t, int64_t, int64_t, int64_t,
+                        int64_t);
 void dia_spmv_plain_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                         at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
                         int64_t, int64_t);
@@ -97,6 +104,13 @@ TORCH_LIBRARY(sparse_hip, m) {
   m.def("ell_jacobi(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor xloc, Tensor b, Tensor dinv, Tensor(a!) xout, "
         "int W, int m, int col_lo, float omega) -> ()");
+  m.def("cg_xr_norm2(Tensor(a!) x, Tensor p, Tensor(b!) r, Tensor q, "
+        "Tensor a, Tensor b, Tensor(c!) dot_out) -> ()");
+  m.def("dia_spmv_bpdot(Tensor dvals, Tensor offs, Tensor r_hlo, Tensor r_own, "
+        "Tensor r_hhi, Tensor p_hlo, Tensor p_own, Tensor p_hhi, "
+        "Tensor(a!) pnew, Tensor(b!) q, Tensor beta_num, Tensor beta_den, "
+        "Tensor(c!) dot_partial, int W, int m, int col_lo, int row0, "
+        "int wsize) -> ()");
   m.def("dia_spmv(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor(a!) y, int W, int m, int col_lo, int row0, "
         "int wsize) -> ()");
@@ -146,6 +160,8 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("ell_spmv", ell_spmv_plain_hip);
   m.impl("ell_spmv_dot", ell_spmv_dot_hip);
   m.impl("ell_jacobi", ell_jacobi_hip);
+  m.impl("cg_xr_norm2", cg_xr_norm2_hip);
+  m.impl("dia_spmv_bpdot", dia_spmv_bpdot_hip);
   m.impl("dia_spmv", dia_spmv_plain_hip);
   m.impl("dia_spmv_dot", dia_spmv_dot_hip);
   m.impl("dia_jacobi", dia_jacobi_hip);

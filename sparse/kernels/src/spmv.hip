@@ -675,7 +675,129 @@ __global__ __launch_bounds__(BLK) void dia_jacobi_kernel(
   }
 }
 
+// Fully-fused CG K1 on the DIA mirror (two-kernel CG iteration):
+//   p_new = r + beta * p_old   (beta = *beta_num / *beta_den, device scalars)
+//   q     = A p_new
+//   pq    = p_new . q          (per-block partials)
+// p_new at neighbor columns is recomputed on the fly from the r / p_old
+// windows (deterministic IEEE => identical to the stored value), so the
+// separate p-update pass over 3 vector streams disappears.  p_old and
+// p_new MUST be distinct buffers (double-buffered by the caller).
+// Reference context: this is the MI355X fusion of the reference CG loop's
+// AXPBY + SpMV + dot task chain (linalg.py:499-565).
+template <typename T, bool SINGLE>
+__global__ __launch_bounds__(BLK) void dia_spmv_bpdot_kernel(
+    const T* __restrict__ dvals, const int64_t* __restrict__ offs,
+    const T* __restrict__ r_hlo, const T* __restrict__ r_own,
+    const T* __restrict__ r_hhi, const T* __restrict__ p_hlo,
+    const T* __restrict__ p_own, const T* __restrict__ p_hhi,
+    T* __restrict__ pnew, T* __restrict__ q,
+    const T* __restrict__ beta_num, const T* __restrict__ beta_den,
+    T* __restrict__ dot_partial, int64_t m, int64_t mp, int W,
+    int64_t col_lo, int64_t row0, int64_t nlo, int64_t nown, int64_t wsize,
+    int64_t own_off) {
+  __shared__ __align__(16) char red_raw[BLK * sizeof(T)];
+  T* red = reinterpret_cast<T*>(red_raw);
+  const T beta = (*beta_num) / (*beta_den);
+  const int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
+  const int64_t r0 = 2 * t;
+  T a0 = ZeroOf<T>::value(), a1 = ZeroOf<T>::value();
+  T p0 = ZeroOf<T>::value(), p1 = ZeroOf<T>::value();
+  if (r0 < mp) {
+    for (int k = 0; k < W; ++k) {
+      const int64_t base = (int64_t)k * mp + r0;
+      T va = nt_load(&dvals[base]);
+      T vb = nt_load(&dvals[base + 1]);
+      const int64_t c0 = row0 + r0 + offs[k] - col_lo;
+      const int64_t i0 = min(max(c0, (int64_t)0), wsize - 1);
+      const int64_t i1 = min(max(c0 + 1, (int64_t)0), wsize - 1);
+      T x0, x1;
+      if (SINGLE) {
+        x0 = r_own[i0] + beta * p_own[i0];
+        x1 = r_own[i1] + beta * p_own[i1];
+      } else {
+        x0 = xpiece(i0, r_hlo, nlo, r_own, nown, r_hhi)
+             + beta * xpiece(i0, p_hlo, nlo, p_own, nown, p_hhi);
+        x1 = xpiece(i1, r_hlo, nlo, r_own, nown, r_hhi)
+             + beta * xpiece(i1, p_hlo, nlo, p_own, nown, p_hhi);
+      }
+      a0 += va * x0;
+      a1 += vb * x1;
+    }
+    // own-row p_new (same formula as the window recompute => identical fp)
+    const int64_t w0 = own_off + r0;
+    if (SINGLE) {
+      p0 = r_own[w0] + beta * p_own[w0];
+      if (r0 + 1 < m) p1 = r_own[w0 + 1] + beta * p_own[w0 + 1];
+    } else {
+      p0 = xpiece(w0, r_hlo, nlo, r_own, nown, r_hhi)
+           + beta * xpiece(w0, p_hlo, nlo, p_own, nown, p_hhi);
+      if (r0 + 1 < m)
+        p1 = xpiece(w0 + 1, r_hlo, nlo, r_own, nown, r_hhi)
+             + beta * xpiece(w0 + 1, p_hlo, nlo, p_own, nown, p_hhi);
+    }
+    if (r0 + 1 < m) {
+      struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
+      TP po{p0, p1};
+      *reinterpret_cast<TP*>(&pnew[r0]) = po;
+      TP qo{a0, a1};
+      *reinterpret_cast<TP*>(&q[r0]) = qo;
+    } else if (r0 < m) {
+      pnew[r0] = p0;
+      q[r0] = a0;
+    }
+  }
+  T d = ZeroOf<T>::value();
+  if (r0 < m) d += a0 * p0;
+  if (r0 + 1 < m) d += a1 * p1;
+  red[threadIdx.x] = d;
+  __syncthreads();
+  for (int w = BLK / 2; w > 0; w >>= 1) {
+    if ((int)threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) dot_partial[blockIdx.x] = red[0];
+}
+
 }  // namespace
+
+void dia_spmv_bpdot_hip(at::Tensor dvals, at::Tensor offs, at::Tensor r_hlo,
+                        at::Tensor r_own, at::Tensor r_hhi, at::Tensor p_hlo,
+                        at::Tensor p_own, at::Tensor p_hhi, at::Tensor pnew,
+                        at::Tensor q, at::Tensor beta_num, at::Tensor beta_den,
+                        at::Tensor dot_partial, int64_t W, int64_t m,
+                        int64_t col_lo, int64_t row0, int64_t wsize) {
+  const int64_t mp = dvals.numel() / W;
+  const int64_t nblocks = (mp / 2 + BLK - 1) / BLK;
+  const int64_t nlo = r_hlo.numel();
+  const int64_t nown = r_own.numel();
+  TORCH_CHECK(pnew.data_ptr() != p_own.data_ptr(),
+              "dia_spmv_bpdot: pnew must not alias p_own (double-buffer)");
+  TORCH_CHECK(dot_partial.numel() >= nblocks, "dia_spmv_bpdot: partial small");
+  // own_off: window index of this rank's first own row (row0 - col_lo)
+  const int64_t own_off_w = row0 - col_lo;
+  // in SINGLE mode indices are into the own slab directly
+  const bool single = (nlo == 0 && r_hhi.numel() == 0);
+  DISPATCH_VALUES(dvals.scalar_type(), "dia_spmv_bpdot", [&] {
+    using T = scalar_t;
+    const T* rhlo_p = nlo ? r_hlo.data_ptr<T>() : r_own.data_ptr<T>();
+    const T* rhhi_p = r_hhi.numel() ? r_hhi.data_ptr<T>() : r_own.data_ptr<T>();
+    const T* phlo_p = nlo ? p_hlo.data_ptr<T>() : p_own.data_ptr<T>();
+    const T* phhi_p = p_hhi.numel() ? p_hhi.data_ptr<T>() : p_own.data_ptr<T>();
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(nblocks), dim3(BLK), 0, cur_stream(),
+                         dvals.data_ptr<T>(), offs.data_ptr<int64_t>(),
+                         rhlo_p, r_own.data_ptr<T>(), rhhi_p, phlo_p,
+                         p_own.data_ptr<T>(), phhi_p, pnew.data_ptr<T>(),
+                         q.data_ptr<T>(), beta_num.data_ptr<T>(),
+                         beta_den.data_ptr<T>(), dot_partial.data_ptr<T>(),
+                         m, mp, (int)W, col_lo, row0, nlo, nown, wsize,
+                         own_off_w);
+    };
+    if (single) launch(dia_spmv_bpdot_kernel<T, true>);
+    else launch(dia_spmv_bpdot_kernel<T, false>);
+  });
+}
 
 void dia_spmv_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
                   at::Tensor own, at::Tensor hhi, at::Tensor y,

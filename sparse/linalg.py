@@ -220,6 +220,39 @@ def cg(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
     fused = (ident_M and isinstance(A, _SparseMatrixLinearOperator)
              and getattr(A.A, "_format", None) == "csr"
              and b.local.is_cuda and not b.local.is_complex())
+    if fused and A.A._dia() is not None:
+        # two-kernel CG iteration on the DIA fast path: K1 folds the
+        # p-update into the SpMV (p = r + beta p, q = Ap, p.q), K2 fuses
+        # x += alpha p, r -= alpha q and |r|^2 — two HBM passes fewer per
+        # iteration than the 4-kernel fused loop below.
+        from . import kernels
+
+        p_b = darray.zeros((n,), dtype=A.dtype)
+        bufs = (p, p_b)
+        rz_cur, rz_old = rz, rz
+        for i in range(maxiter):
+            if i == 0:
+                pq = A.A.spmv_dot(p, q)
+                cur = p
+            else:
+                # p_old alternates: iter 1 reads p writes p_b, iter 2 reads
+                # p_b writes p, ...
+                pold, pnew = bufs[(i + 1) % 2], bufs[i % 2]
+                pq = A.A.spmv_bpdot(r, pold, pnew, q, rz_cur, rz_old)
+                cur = pnew
+            rz_new = kernels.cg_xr_norm2(x.local, cur.local, r.local,
+                                         q.local, rz_cur, pq)
+            comm.all_reduce_(rz_new)
+            rz_old, rz_cur = rz_cur, rz_new
+            if conv_test_iters and (i % conv_test_iters == 0 or i == maxiter - 1):
+                if math.sqrt(max(float(rz_cur.item()), 0.0)) < threshold:
+                    info = 0
+                    break
+            if callback is not None:
+                callback(x)
+        if info != 0 and float(r.norm().item()) < threshold:
+            info = 0
+        return x, info
     for i in range(maxiter):
         if fused:
             pq = A.A.spmv_dot(p, q)
