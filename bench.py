@@ -97,7 +97,9 @@ def main() -> None:
         from sparse import kernels
 
         rz_buf = rz.clone()
-        use2 = A._dia() is not None  # two-kernel CG iteration (DIA fast path)
+        # two-kernel CG iteration (opt-in: measured slower than the
+        # 4-kernel loop on MI355X — see sparse/linalg.py cg())
+        use2 = A._dia() is not None and os.environ.get("SPARSE_CG2") == "1"
         if use2:
             # K1 folds p = r + beta*p into the SpMV (q = Ap, p.q fused);
             # K2 fuses x += alpha p, r -= alpha q and |r|^2 — two HBM

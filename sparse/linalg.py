@@ -16,6 +16,7 @@ convergence checks.
 from __future__ import annotations
 
 import math
+import os
 from typing import Callable, Optional
 
 import numpy as np
@@ -220,11 +221,16 @@ def cg(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
     fused = (ident_M and isinstance(A, _SparseMatrixLinearOperator)
              and getattr(A.A, "_format", None) == "csr"
              and b.local.is_cuda and not b.local.is_complex())
-    if fused and A.A._dia() is not None:
+    if (fused and A.A._dia() is not None
+            and os.environ.get("SPARSE_CG2") == "1"):
         # two-kernel CG iteration on the DIA fast path: K1 folds the
         # p-update into the SpMV (p = r + beta p, q = Ap, p.q), K2 fuses
-        # x += alpha p, r -= alpha q and |r|^2 — two HBM passes fewer per
-        # iteration than the 4-kernel fused loop below.
+        # x += alpha p, r -= alpha q and |r|^2.  MEASURED SLOWER than the
+        # 4-kernel loop on MI355X at nx=16384 (6.14 vs 5.89 ms/iter:
+        # K1 streams TWO vector windows through L2 and is issue-bound at
+        # 5.0 TB/s while the axpby passes it replaces run at 6.0), so this
+        # is opt-in via SPARSE_CG2=1; kept because the tradeoff may flip
+        # for wider stencils (more dvals reuse per vector byte).
         from . import kernels
 
         p_b = darray.zeros((n,), dtype=A.dtype)
