@@ -247,10 +247,18 @@ class csr_array(CompressedBase, DenseSparseBase):
         return None if self._ell_cache == "no" else self._ell_cache
 
     def _xplan(self, xpart: RowPartition) -> WindowGatherPlan:
-        key = ("x", xpart.starts)
+        from .settings import settings
+
+        precise = settings.precise_images and comm.world_size() > 1
+        key = ("x", xpart.starts, precise)
         if key not in self._plan_cache:
-            lo, hi = self._col_window()
-            self._plan_cache[key] = WindowGatherPlan(lo, hi, xpart)
+            if precise:
+                from .parallel.gather import PreciseGatherPlan
+
+                self._plan_cache[key] = PreciseGatherPlan(self._indices, xpart)
+            else:
+                lo, hi = self._col_window()
+                self._plan_cache[key] = WindowGatherPlan(lo, hi, xpart)
         return self._plan_cache[key]
 
     # -- basic properties ----------------------------------------------------
@@ -403,10 +411,13 @@ class csr_array(CompressedBase, DenseSparseBase):
         if x.shape[0] != self.shape[1]:
             raise ValueError(f"dimension mismatch {self.shape} @ {x.shape}")
         plan = self._xplan(x.partition)
+        precise = not isinstance(plan, WindowGatherPlan)
         vdt = self._out_dtype(x.local.dtype)
-        dm = self._dia() if self._values.dtype == vdt else None
+        dm = (self._dia() if self._values.dtype == vdt and not precise
+              else None)
         ell = None if dm is not None else (
-            self._ell() if self._values.dtype == vdt else None)
+            self._ell() if self._values.dtype == vdt and not precise
+            else None)
         if dm is not None:
             from . import kernels
 
@@ -425,6 +436,10 @@ class csr_array(CompressedBase, DenseSparseBase):
             xw = plan.gather(x.local.to(vdt))
             self._max_row_nnz()
             lc = self.local
+            if precise:
+                lc = ops.LocalCSR(lc.indptr, plan.remap(lc.indices),
+                                  lc.values, lc.nrows, plan.hi,
+                                  lc.max_row_nnz)
             if lc.values.dtype != vdt:
                 lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt),
                                   lc.nrows, lc.ncols, lc.max_row_nnz)
@@ -442,8 +457,12 @@ class csr_array(CompressedBase, DenseSparseBase):
         vdt = self._out_dtype(B.local.dtype)
         Bw = plan.gather(B.local.to(vdt))
         lc = self.local
+        if not isinstance(plan, WindowGatherPlan):
+            lc = ops.LocalCSR(lc.indptr, plan.remap(lc.indices), lc.values,
+                              lc.nrows, plan.hi, lc.max_row_nnz)
         if lc.values.dtype != vdt:
-            lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt), lc.nrows, lc.ncols)
+            lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt),
+                              lc.nrows, lc.ncols, lc.max_row_nnz)
         Clocal = ops.spmm(lc, Bw, col_lo=plan.lo)
         if out is not None:
             out.local.copy_(Clocal.to(out.local.dtype))
@@ -488,8 +507,10 @@ class csr_array(CompressedBase, DenseSparseBase):
         from . import kernels
 
         plan = self._xplan(p.partition)
-        dm = self._dia()
-        ell = None if dm is not None else self._ell()
+        precise = not isinstance(plan, WindowGatherPlan)
+        dm = self._dia() if not precise else None
+        ell = None if dm is not None else (
+            self._ell() if not precise else None)
         if dm is not None:
             pieces = plan.gather_halos(p.local)
             dot = kernels.dia_spmv_dot(dm, pieces, q.local, p.local, plan.lo,
@@ -501,6 +522,10 @@ class csr_array(CompressedBase, DenseSparseBase):
             xw = plan.gather(p.local)
             dot = torch.zeros((), dtype=self._values.dtype, device=self._values.device)
             lc = self.local
+            if precise:
+                lc = ops.LocalCSR(lc.indptr, plan.remap(lc.indices),
+                                  lc.values, lc.nrows, plan.hi,
+                                  lc.max_row_nnz)
             kernels.spmv_dot(lc, xw, q.local, p.local, dot, plan.lo)
         comm.all_reduce_(dot)
         return dot

@@ -147,6 +147,68 @@ class WindowGatherPlan:
         return out.reshape(self.hi - self.lo, *tail) if tail else out
 
 
+class PreciseGatherPlan:
+    """Exact-index gather plan (reference precise images,
+    settings.py:23-33 / MinMaxImagePartition's precise alternative): the
+    operand slice communicated per rank is exactly the set of distinct
+    column indices its slab touches, not the min/max window.  Pays off for
+    scattered matrices at world size > 1 where the window spans nearly all
+    of x but only a few columns are read.
+
+    Interface-compatible with WindowGatherPlan's gather() path: gather()
+    returns a compact array of the needed values (sorted by global column),
+    lo == 0 and hi == len(cols), and remap(indices) rewrites a local index
+    tensor into positions in that compact array."""
+
+    def __init__(self, indices: torch.Tensor, xpart: RowPartition, group=None):
+        self.xpart = xpart
+        self.group = group
+        ws = comm.world_size(group)
+        me = comm.rank(group)
+        # distinct needed global columns, ascending (=> grouped by owner)
+        self.cols = torch.unique(indices.to(torch.int64))
+        self.lo = 0
+        self.hi = int(self.cols.numel())
+        starts = torch.tensor(xpart.starts, dtype=torch.int64,
+                              device=self.cols.device)
+        cuts = torch.searchsorted(self.cols, starts)
+        self._req_counts = [int(cuts[p + 1] - cuts[p]) for p in range(ws)]
+        if ws == 1:
+            self._send_idx = [self.cols]
+            self._local_cols = self.cols
+            self._remap_cache = None
+            return
+        # exchange the request lists: owners learn which of their entries
+        # each peer needs
+        reqs = [self.cols[cuts[p]: cuts[p + 1]].contiguous() for p in range(ws)]
+        got = comm.all_to_all_v(reqs, group=group)
+        s0 = xpart.start(me)
+        self._send_idx = [g - s0 for g in got]  # my-slab-local gather indices
+        self._remap_cache = None
+
+    def remap(self, indices: torch.Tensor) -> torch.Tensor:
+        """Rewrite global column indices into compact positions (cached)."""
+        if self._remap_cache is None:
+            self._remap_cache = torch.searchsorted(
+                self.cols, indices.to(torch.int64)).to(indices.dtype)
+        return self._remap_cache
+
+    def gather(self, xlocal: torch.Tensor) -> torch.Tensor:
+        ws = comm.world_size(self.group)
+        if ws == 1:
+            return xlocal[self.cols]
+        tail = xlocal.shape[1:]
+        k = 1
+        for t in tail:
+            k *= t
+        send = [(xlocal[idx] if idx.numel() else xlocal[:0]).reshape(-1)
+                for idx in self._send_idx]
+        recv = comm.all_to_all_v(send, group=self.group,
+                                 recv_counts=[c * k for c in self._req_counts])
+        out = torch.cat(recv, dim=0)
+        return out.reshape(self.hi, *tail) if tail else out
+
+
 class ReduceScatterPlan:
     """Inverse of the window gather: each rank holds partial contributions to
     y[lo:hi); owners receive and sum them.

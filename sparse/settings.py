@@ -1,8 +1,11 @@
 """Runtime settings (reference sparse/settings.py:22-36).
 
-precise_images: when True, gather plans use the exact index set instead of
-the min/max window (reference LEGATE_SPARSE_PRECISE_IMAGES).  The window
-plan is the default, as in the reference build.
+precise_images: when True (and world size > 1), SpMV/SpMM gather plans
+communicate the exact set of distinct columns each rank's slab touches
+(PreciseGatherPlan) instead of the min/max window — the reference's
+LEGATE_SPARSE_PRECISE_IMAGES (settings.py:23-33).  The window plan is the
+default, as in the reference build; banded matrices keep the window plan
+regardless (their windows are already tight and feed the DIA/ELL kernels).
 """
 from __future__ import annotations
 
@@ -15,13 +18,6 @@ class Settings:
             "SPARSE_PRECISE_IMAGES",
             os.environ.get("LEGATE_SPARSE_PRECISE_IMAGES", "0"),
         ) not in ("0", "", "false", "False")
-        if self.precise_images:
-            import warnings
-
-            warnings.warn(
-                "SPARSE_PRECISE_IMAGES: exact-index gather plans are not "
-                "implemented; min/max window plans (the reference default "
-                "build's behavior) are used", UserWarning)
         # cap on the number of ranks that own data (reference
         # LEGATE_SPARSE_NUM_PROCS, runtime.py:61-63); the remaining ranks
         # participate in collectives with empty slabs

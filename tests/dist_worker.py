@@ -118,6 +118,7 @@ def main():
     extra_halo_check()
     extra_solver_checks()
     extra_num_procs_check()
+    extra_precise_images_check()
 
     if rank == 0:
         print("DIST_ALL_OK")
@@ -194,6 +195,42 @@ def extra_samplesort_check():
     allv = _c.all_gather_rows(v, [257] * dist.get_world_size()).numpy()
     assert sorted(allk.tolist()) == gk.tolist(), "samplesort permutation"
     assert np.isclose(gv.sum(), allv.sum()), "payload preserved"
+
+
+def extra_precise_images_check():
+    """SPARSE_PRECISE_IMAGES: exact-index gather plans (PreciseGatherPlan)
+    must match the window-plan results and the scipy oracle (reference
+    settings.py:23-33)."""
+    import scipy.sparse as sps4
+
+    import sparse.settings as st
+    from sparse import csr_array as _csr
+    from sparse.parallel.gather import PreciseGatherPlan
+    from sparse.parallel.partition import RowPartition as _RP
+
+    rng = np.random.default_rng(9)
+    s = sps4.random(41, 400, 0.02, random_state=11, format="csr")
+    s.sort_indices()
+    x = rng.random(400)
+    B = rng.random((400, 3))
+    old = st.settings.precise_images
+    st.settings.precise_images = True
+    try:
+        A = _csr(s)
+        plan = A._xplan(_RP.equal(400, dist.get_world_size()))
+        assert isinstance(plan, PreciseGatherPlan), type(plan)
+        assert np.allclose(np.asarray(A @ x), s @ x), "precise spmv"
+        assert np.allclose(np.asarray(A @ B), s @ B), "precise spmm"
+        # CG through the precise plan (spmv_dot CPU fallback path)
+        from sparse import linalg as _lin
+
+        spd = (s[:, :41] + s[:, :41].T + 41 * sps4.eye(41)).tocsr()
+        Ap = _csr(spd)
+        bb = rng.random(41)
+        xs, info = _lin.cg(Ap, bb, tol=1e-10, conv_test_iters=5)
+        assert np.allclose(spd @ np.asarray(xs), bb, atol=1e-6), "precise cg"
+    finally:
+        st.settings.precise_images = old
 
 
 def extra_num_procs_check():
