@@ -67,3 +67,57 @@ def test_quantum_mis():
     assert "approx ratio" in out
     ratio = float(out.split("approx ratio = ")[1].split()[0])
     assert 0.3 < ratio <= 1.0
+
+
+def test_reference_amg():
+    out = run("reference_amg.py", "-n", "4096")
+    assert "iters=" in out and "info=0" in out
+
+
+def test_plot_helpers(tmp_path):
+    import subprocess as sp
+
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "import plot\n"
+        "plot.convergence_history([1, .1, .01], fname=%r)\n"
+        "plot.grid_scalar(np.arange(64.0))\n"
+        "plot.trimesh([[0,0],[1,0],[0,1],[1,1]], [[0,1,2],[1,3,2]])\n"
+        "print('OK')\n" % (EX, str(tmp_path / "c.png")))
+    r = sp.run([sys.executable, "-c", code], capture_output=True, timeout=120)
+    assert r.returncode == 0 and b"OK" in r.stdout, r.stderr.decode()[-800:]
+    assert (tmp_path / "c.png").exists()
+
+
+def test_pyamg_bridge_conversion():
+    """from_pyamg on a hand-built hierarchy object (pyamg itself optional):
+    the converted V-cycle must precondition CG on the 2-D Poisson system."""
+    import types
+
+    import numpy as np
+    import scipy.sparse as sps
+
+    sys.path.insert(0, EX)
+    from pyamg_to_sparse import from_pyamg
+
+    from sparse import csr_array, linalg
+
+    nx = 24
+    n = nx * nx
+    A = (sps.kron(sps.eye(nx), sps.diags([-1, 2, -1], [-1, 0, 1], (nx, nx)))
+         + sps.kron(sps.diags([-1, 2, -1], [-1, 0, 1], (nx, nx)),
+                    sps.eye(nx))).tocsr()
+    # tentative-aggregation hierarchy built by hand (stands in for pyamg's)
+    agg = np.arange(n) // 4
+    nc = int(agg.max()) + 1
+    P = sps.csr_matrix((np.ones(n), (np.arange(n), agg)), shape=(n, nc))
+    lvl0 = types.SimpleNamespace(A=A, P=P, R=P.T.tocsr())
+    lvl1 = types.SimpleNamespace(A=(P.T @ A @ P).tocsr())
+    ml = types.SimpleNamespace(levels=[lvl0, lvl1])
+    M = from_pyamg(ml)
+    b = np.ones(n)
+    x, info = linalg.cg(csr_array(A), b, M=M, tol=1e-8, maxiter=400,
+                        conv_test_iters=5)
+    assert info == 0
+    assert np.allclose(A @ np.asarray(x), b, atol=1e-5)
