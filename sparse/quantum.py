@@ -32,6 +32,11 @@ __all__ = [
 ]
 
 
+# graphs wider than this use the multi-limb (n, L) uint64 path; tests
+# lower it to force limb coverage on small graphs
+_INT64_MAX_NODES = 63
+
+
 def _neighbor_masks(graph) -> List[int]:
     import networkx as nx  # noqa: F401
 
@@ -44,6 +49,29 @@ def _neighbor_masks(graph) -> List[int]:
         nbr[iu] |= 1 << iv
         nbr[iv] |= 1 << iu
     return nbr
+
+
+def _neighbor_limbs(graph, L: int) -> np.ndarray:
+    """(n, L) uint64 adjacency bitsets, limb 0 = bits 0..63."""
+    n = graph.number_of_nodes()
+    nodes = list(graph.nodes())
+    index = {v: i for i, v in enumerate(nodes)}
+    nbr = np.zeros((n, L), dtype=np.uint64)
+    for u, v in graph.edges():
+        iu, iv = index[u], index[v]
+        nbr[iu, iv // 64] |= np.uint64(1) << np.uint64(iv % 64)
+        nbr[iv, iu // 64] |= np.uint64(1) << np.uint64(iu % 64)
+    return nbr
+
+
+def _limb_sortable(masks: np.ndarray) -> np.ndarray:
+    """View (m, L) uint64 limb masks as a structured array whose record
+    comparison is lexicographic most-significant-limb first — sortable and
+    searchsorted-compatible."""
+    L = masks.shape[1]
+    rev = np.ascontiguousarray(masks[:, ::-1])
+    dt = np.dtype([("f%d" % i, "<u8") for i in range(L)])
+    return rev.view(dt).reshape(masks.shape[0])
 
 
 def _popcount64(a: np.ndarray) -> np.ndarray:
@@ -64,8 +92,8 @@ def enumerate_independent_sets(graph, k: int, prevk_sets=None, prevk_queues=None
     queues[i] = candidate nodes with index greater than every member of
     sets[i] and not adjacent to it (the canonical-extension frontier)."""
     n = graph.number_of_nodes()
-    if n > 63:
-        return _enumerate_py(graph, k, prevk_sets, prevk_queues)
+    if n > _INT64_MAX_NODES:
+        return _enumerate_limbs(graph, k, prevk_sets, prevk_queues)
     nbr = np.array(_neighbor_masks(graph), dtype=np.int64)
     if k == 1:
         sets = np.array([1 << v for v in range(n)], dtype=np.int64)
@@ -91,35 +119,55 @@ def enumerate_independent_sets(graph, k: int, prevk_sets=None, prevk_queues=None
     return np.concatenate(out_s), np.concatenate(out_q)
 
 
-def _enumerate_py(graph, k, prevk_sets, prevk_queues):
-    """Arbitrary-width fallback (> 63 nodes)."""
+def _enumerate_limbs(graph, k, prevk_sets, prevk_queues):
+    """Vectorized arbitrary-width path (> 63 nodes): masks are (m, L)
+    uint64 limb arrays, L = ceil(n/64) — the MI355X counterpart of the
+    reference's templated IntSet<N,T> bitsets (quantum.h:27-...)."""
     n = graph.number_of_nodes()
-    nbr = _neighbor_masks(graph)
+    L = (n + 63) // 64
+    nbr = _neighbor_limbs(graph, L)
+    one = np.uint64(1)
     if k == 1:
-        sets = [1 << v for v in range(n)]
-        queues = []
+        sets = np.zeros((n, L), dtype=np.uint64)
+        above = np.zeros((n, L), dtype=np.uint64)
         for v in range(n):
-            q = 0
-            for u in range(v + 1, n):
-                if not (nbr[v] >> u) & 1:
-                    q |= 1 << u
-            queues.append(q)
+            li, sh = divmod(v, 64)
+            sets[v, li] = one << np.uint64(sh)
+            above[v, li] = (~np.uint64(0)) << np.uint64(sh)
+            above[v, li] &= ~(one << np.uint64(sh))
+            above[v, li + 1:] = ~np.uint64(0)
+            if n % 64:
+                above[v, L - 1] &= (one << np.uint64(n % 64)) - one
+        queues = above & ~nbr
         return sets, queues
     assert prevk_sets is not None and prevk_queues is not None
-    sets, queues = [], []
-    for S, Q in zip(prevk_sets, prevk_queues):
-        q = Q
-        while q:
-            v = (q & -q).bit_length() - 1
-            q &= q - 1
-            sets.append(S | (1 << v))
-            queues.append(Q & ~((1 << (v + 1)) - 1) & ~nbr[v])
-    return sets, queues
+    S = np.asarray(prevk_sets, dtype=np.uint64)
+    Q = np.asarray(prevk_queues, dtype=np.uint64)
+    out_s, out_q = [], []
+    for v in range(n):
+        li, sh = divmod(v, 64)
+        rows = (Q[:, li] >> np.uint64(sh)) & one != 0
+        if not rows.any():
+            continue
+        ns = S[rows].copy()
+        ns[:, li] |= one << np.uint64(sh)
+        lowmask = np.zeros(L, dtype=np.uint64)
+        lowmask[:li] = ~np.uint64(0)
+        lowmask[li] = ((one << np.uint64(sh)) - one) | (one << np.uint64(sh))
+        nq = Q[rows] & ~lowmask & ~nbr[v]
+        out_s.append(ns)
+        out_q.append(nq)
+    if not out_s:
+        z = np.zeros((0, L), dtype=np.uint64)
+        return z, z.copy()
+    return np.concatenate(out_s), np.concatenate(out_q)
 
 
 def sets_to_sizes(queues, graph) -> np.ndarray:
     """Popcount of each candidate queue (reference SETS_TO_SIZES)."""
     if isinstance(queues, np.ndarray):
+        if queues.ndim == 2:  # multi-limb
+            return _popcount64(queues).sum(axis=1)
         return _popcount64(queues)
     return np.array([bin(int(q)).count("1") for q in queues], dtype=np.int64)
 
@@ -178,11 +226,25 @@ class HamiltonianDriver:
         # ascending ids: group k starts at offsets[k]; within each group ids
         # follow VALUE-sorted mask order so subset lookup is a searchsorted
         offsets = np.concatenate([[0], np.cumsum(self.ip)])
-        groups = []
+        limbs = any(isinstance(g, np.ndarray) and g.ndim == 2
+                    for g in all_sets[1:])
+        if limbs:
+            L = next(g.shape[1] for g in all_sets[1:]
+                     if isinstance(g, np.ndarray) and g.ndim == 2)
+            all_sets[0] = np.zeros((1, L), dtype=np.uint64)
+        groups, sortables = [], []
         for k, group in enumerate(all_sets):
-            g = np.asarray(group, dtype=np.int64) if not isinstance(
-                group, np.ndarray) else group
-            groups.append(np.sort(g))
+            if limbs:
+                g = np.asarray(group, dtype=np.uint64)
+                key = _limb_sortable(g)
+                order = np.argsort(key)
+                groups.append(g[order])
+                sortables.append(key[order])
+            else:
+                g = np.asarray(group, dtype=np.int64) if not isinstance(
+                    group, np.ndarray) else group
+                groups.append(np.sort(g))
+                sortables.append(groups[-1])
         rows_l, cols_l = [], []
         for k in range(1, len(groups)):
             Sk = groups[k]
@@ -190,12 +252,24 @@ class HamiltonianDriver:
             # peel the k set bits of every mask, vectorized per position
             rem = Sk.copy()
             for _ in range(k):
-                low = rem & -rem
-                Tm = Sk & ~low  # subset with that member removed
-                tid = offsets[k - 1] + np.searchsorted(groups[k - 1], Tm)
+                if limbs:
+                    m = len(Sk)
+                    ar = np.arange(m)
+                    li = np.argmax(rem != 0, axis=1)
+                    limbv = rem[ar, li]
+                    low = limbv & (np.uint64(0) - limbv)
+                    Tm = Sk.copy()
+                    Tm[ar, li] = Sk[ar, li] & ~low
+                    tid = offsets[k - 1] + np.searchsorted(
+                        sortables[k - 1], _limb_sortable(Tm))
+                    rem[ar, li] = limbv & ~low
+                else:
+                    low = rem & -rem
+                    Tm = Sk & ~low  # subset with that member removed
+                    tid = offsets[k - 1] + np.searchsorted(groups[k - 1], Tm)
+                    rem = rem & ~low
                 rows_l.append(sid)
                 cols_l.append(tid)
-                rem = rem & ~low
         rows = np.concatenate(rows_l)
         cols = np.concatenate(cols_l)
         # reference state ordering: largest sets first, empty set last

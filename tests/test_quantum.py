@@ -101,3 +101,42 @@ def test_annealing_evolution_small():
     assert res.success
     # norm conserved
     assert np.isclose(np.linalg.norm(res.y[:, -1]), 1.0, atol=1e-6)
+
+
+def test_limbs_path_matches_int64():
+    """Force the multi-limb (>63-node) enumeration on a small graph and
+    compare sets/queues/polynomial/driver against the int64 path."""
+    import networkx as nx
+
+    from sparse import quantum
+
+    g = nx.random_geometric_graph(18, 0.45, seed=5)
+    ip_fast = quantum.independence_polynomial(g)
+    H_fast = quantum.HamiltonianDriver(graph=g).hamiltonian.to_scipy_sparse_csr()
+    old = quantum._INT64_MAX_NODES
+    quantum._INT64_MAX_NODES = 0
+    try:
+        ip_limbs = quantum.independence_polynomial(g)
+        H_limbs = quantum.HamiltonianDriver(graph=g).hamiltonian.to_scipy_sparse_csr()
+    finally:
+        quantum._INT64_MAX_NODES = old
+    assert ip_fast == ip_limbs
+    assert (H_fast != H_limbs).nnz == 0
+
+
+def test_wide_graph_polynomial():
+    """>63-node graph (multi-limb path): disjoint union of 5 cliques of 14
+    nodes (70 total) — an independent set picks at most one node per
+    clique, so ip[k] = C(5,k) * 14^k."""
+    import math
+
+    import networkx as nx
+
+    from sparse import quantum
+
+    g = nx.disjoint_union_all([nx.complete_graph(14) for _ in range(5)])
+    assert g.number_of_nodes() == 70
+    ip = quantum.independence_polynomial(g)
+    assert len(ip) == 6
+    for k in range(6):
+        assert ip[k] == math.comb(5, k) * 14 ** k, k
