@@ -31,8 +31,11 @@ parser.add_argument("-theta", type=float, default=0.0)
 parser.add_argument("-maxiter", type=int, default=400)
 parser.add_argument("-tol", type=float, default=1e-8)
 parser.add_argument("-max_coarse", type=int, default=512)
-args, _ = parser.parse_known_args()
-_, timer, npx, sparse, linalg, use_sparse = parse_common_args()
+# parse argv only when run as a script: importing this module (e.g. the
+# pyamg bridge reusing vcycle) must not react to the host program's argv
+args, _ = parser.parse_known_args(None if __name__ == "__main__" else [])
+_, timer, npx, sparse, linalg, use_sparse = parse_common_args(
+    None if __name__ == "__main__" else [])
 
 from sparse import csr_array, darray, gallery
 from sparse.parallel import comm

@@ -77,11 +77,11 @@ def get_phase_procs(use_sparse: bool):
     return DummyScope(), DummyScope()
 
 
-def parse_common_args():
+def parse_common_args(argv=None):
     parser = argparse.ArgumentParser(add_help=False)
     parser.add_argument("--package", default="sparse",
                         choices=["sparse", "legate", "scipy"])
-    args, _ = parser.parse_known_args()
+    args, _ = parser.parse_known_args(argv)
     if args.package in ("sparse", "legate"):
         import sparse
         import sparse.linalg as linalg
