@@ -375,6 +375,8 @@ class csr_array(CompressedBase, DenseSparseBase):
         from .csc import csc_array
 
         if isinstance(other, (csr_array, csc_array, coo_array)):
+            if isinstance(other, csc_array) and comm.world_size() > 1:
+                return self._spgemm_2d(other)
             B = other.tocsr() if not isinstance(other, csr_array) else other
             return self._spgemm(B)
         x = other
@@ -585,6 +587,71 @@ class csr_array(CompressedBase, DenseSparseBase):
         C_l = ops.spgemm(A_l, B_l, a_col_lo=lo)
         return csr_array.from_local(C_l.indptr, C_l.indices, C_l.values,
                                     self.partition, (self.shape[0], B.shape[1]))
+
+    def _spgemm_2d(self, B) -> "csr_array":
+        """C = A(csr) @ B(csc) on a 2-D (gx, gy) process grid — the MI355X
+        realization of the reference's SPGEMM_CSR_CSR_CSC 2-D replicated
+        algorithm (csr.py:1495-1728, spgemm_csr_csr_csc.cu): rank (pi, pj)
+        gathers A's row block pi and B's column block pj (B stays CSC — no
+        global transpose), multiplies locally, then shuffles tile rows to
+        the 1-D row owners.  Per-rank comm is O(nnz(A)/gx + nnz(B)/gy +
+        nnz(C)/W) vs the 1-D algorithm's O(nnz(B)) B-row broadcast."""
+        from .parallel.shuffle import gather_csr_rows, shuffle_to_owner
+        from .utils import factor_int
+
+        if self.shape[1] != B.shape[0]:
+            raise ValueError(f"dimension mismatch {self.shape} @ {B.shape}")
+        W = comm.world_size()
+        gy, gx = factor_int(W)  # gx >= gy: more row blocks than col blocks
+        pi, pj = divmod(comm.rank(), gy)
+        m, K, n = self.shape[0], self.shape[1], B.shape[1]
+        rb = RowPartition.equal(m, gx)
+        cb = RowPartition.equal(n, gy)
+        r0, r1 = rb.start(pi), rb.stop(pi)
+        c0, c1 = cb.start(pj), cb.stop(pj)
+        # A row block pi (global rows [r0, r1)); every rank participates in
+        # both collective gathers with its own target range
+        aip, aix, avs = gather_csr_rows(self._indptr, self._indices,
+                                        self._values, self.partition, r0, r1)
+        # B column block pj, still column-compressed
+        bip, bix, bvs = gather_csr_rows(B._colptr, B._indices, B._values,
+                                        B.partition, c0, c1)
+        # local CSC (K x (c1-c0)) -> CSR by stable row sort (device torch ops)
+        ccols = torch.repeat_interleave(
+            torch.arange(c1 - c0, dtype=torch.int64, device=bix.device),
+            bip[1:] - bip[:-1])
+        order = torch.argsort(bix.to(torch.int64), stable=True)
+        brows = bix.to(torch.int64)[order]
+        bcsr_ip = torch.zeros(K + 1, dtype=torch.int64, device=bix.device)
+        if brows.numel():
+            bcsr_ip[1:] = torch.cumsum(
+                torch.bincount(brows, minlength=K), dim=0)
+        vdt = common_value_dtype(self._values.dtype, bvs.dtype)
+        A_l = ops.LocalCSR(aip, aix, avs.to(vdt), r1 - r0, K)
+        B_l = ops.LocalCSR(bcsr_ip, ccols[order].to(self._indices.dtype),
+                           bvs[order].to(vdt), K, c1 - c0)
+        C_l = ops.spgemm(A_l, B_l, a_col_lo=0)
+        # tile (r1-r0 x c1-c0) -> global COO -> shuffle rows to 1-D owners
+        trows = r0 + torch.repeat_interleave(
+            torch.arange(r1 - r0, dtype=torch.int64, device=aix.device),
+            C_l.indptr[1:] - C_l.indptr[:-1])
+        tcols = C_l.indices.to(torch.int64) + c0
+        part = RowPartition.equal(m, W)
+        i, j, v = shuffle_to_owner(trows, part, tcols, C_l.values)
+        me = comm.rank()
+        mloc = part.count(me)
+        key = (i - part.start(me)) * max(1, n) + j
+        key, order = torch.sort(key)
+        v = v[order]
+        rows_l = torch.div(key, max(1, n), rounding_mode="floor")
+        cols_l = key - rows_l * max(1, n)
+        indptr = torch.zeros(mloc + 1, dtype=torch.int64, device=v.device)
+        if rows_l.numel():
+            indptr[1:] = torch.cumsum(
+                torch.bincount(rows_l, minlength=mloc), dim=0)
+        idt = torch.int32 if n < 2**31 - 1 else torch.int64
+        return csr_array.from_local(indptr, cols_l.to(idt), v.to(vdt),
+                                    part, (m, n))
 
     def sddmm(self, C, D) -> "csr_array":
         """vals'[i,j] = vals[i,j] * (C[i,:] @ D[:,j]) (reference csr.py:693-705)."""

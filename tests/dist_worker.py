@@ -119,6 +119,7 @@ def main():
     extra_solver_checks()
     extra_num_procs_check()
     extra_precise_images_check()
+    extra_spgemm_2d_check()
 
     if rank == 0:
         print("DIST_ALL_OK")
@@ -195,6 +196,31 @@ def extra_samplesort_check():
     allv = _c.all_gather_rows(v, [257] * dist.get_world_size()).numpy()
     assert sorted(allk.tolist()) == gk.tolist(), "samplesort permutation"
     assert np.isclose(gv.sum(), allv.sum()), "payload preserved"
+
+
+def extra_spgemm_2d_check():
+    """csr @ csc takes the 2-D grid algorithm at ws>1 (reference
+    SPGEMM_CSR_CSR_CSC parity): compare against the scipy oracle,
+    including a non-square case and an empty-product case."""
+    import scipy.sparse as sps5
+
+    from sparse import csc_array as _csc
+    from sparse import csr_array as _csr
+
+    a = sps5.random(33, 47, 0.15, random_state=21, format="csr")
+    b = sps5.random(47, 29, 0.15, random_state=22, format="csc")
+    C = _csr(a) @ _csc(b)
+    ref = (a @ b).tocsr()
+    ref.sum_duplicates()
+    ref.sort_indices()
+    got = C.to_scipy_sparse_csr()
+    assert got.shape == ref.shape
+    assert np.allclose(got.toarray(), ref.toarray()), "spgemm 2d"
+    # empty product
+    a0 = sps5.csr_matrix((10, 8))
+    b0 = sps5.csc_matrix((8, 6))
+    C0 = _csr(a0) @ _csc(b0)
+    assert C0.nnz == 0 and C0.shape == (10, 6)
 
 
 def extra_precise_images_check():
