@@ -190,6 +190,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         self._plan_cache = {}
         self._ell_cache = None
         self._dia_cache = None
+        self._csc_cache = None
         self._maxrow_cache = None
 
     @classmethod
@@ -286,6 +287,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         self._values = torch.as_tensor(mine, device=self._values.device).to(self._values.dtype)
         self._ell_cache = None
         self._dia_cache = None
+        self._csc_cache = None
 
     @property
     def indices(self) -> np.ndarray:
@@ -386,6 +388,16 @@ class csr_array(CompressedBase, DenseSparseBase):
             x = asdistarray(x)
         if isinstance(x, DistArray):
             if x.ndim == 1:
+                if spmv_domain_part and comm.world_size() > 1:
+                    # domain (column) partition SpMV: each rank multiplies
+                    # its COLUMN block against its own x slab (no x gather)
+                    # and partial-y windows are sum-reduced to the row
+                    # owners — the reference's CSR_SPMV_COL_SPLIT strategy
+                    # (csr.py:869-927), executed via a cached CSC mirror +
+                    # ReduceScatterPlan.
+                    if self._csc_cache is None:
+                        self._csc_cache = self.tocsc()
+                    return self._csc_cache._spmv(x, out=out)
                 return self._spmv(x, out=out)
             if x.ndim == 2:
                 return self._spmm(x, out=out)

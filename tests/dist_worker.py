@@ -120,6 +120,7 @@ def main():
     extra_num_procs_check()
     extra_precise_images_check()
     extra_spgemm_2d_check()
+    extra_domain_part_spmv_check()
 
     if rank == 0:
         print("DIST_ALL_OK")
@@ -196,6 +197,21 @@ def extra_samplesort_check():
     allv = _c.all_gather_rows(v, [257] * dist.get_world_size()).numpy()
     assert sorted(allk.tolist()) == gk.tolist(), "samplesort permutation"
     assert np.isclose(gv.sum(), allv.sum()), "payload preserved"
+
+
+def extra_domain_part_spmv_check():
+    """dot(..., spmv_domain_part=True) runs the column-split + reduce path
+    at ws>1 (reference CSR_SPMV_COL_SPLIT) and matches the oracle."""
+    import scipy.sparse as sps6
+
+    from sparse import csr_array as _csr
+
+    s = sps6.random(31, 44, 0.2, random_state=31, format="csr")
+    x = np.random.default_rng(32).random(44)
+    A = _csr(s)
+    y = A.dot(x, spmv_domain_part=True)
+    assert np.allclose(np.asarray(y), s @ x), "domain-part spmv"
+    assert A._csc_cache is not None
 
 
 def extra_spgemm_2d_check():
