@@ -39,9 +39,13 @@ def _nccl(group=None) -> bool:
 
 
 def _to_wire_device(t: torch.Tensor, group=None) -> torch.Tensor:
-    """NCCL moves only CUDA tensors; host scalars ride via a device copy."""
+    """NCCL moves only CUDA tensors; host scalars ride via a device copy.
+    gloo moves only host tensors; CUDA data is staged through the host
+    (the several-ranks-per-GPU test configuration)."""
     if _nccl(group) and not t.is_cuda:
         return t.cuda()
+    if initialized() and not _nccl(group) and t.is_cuda:
+        return t.cpu()
     return t
 
 
@@ -51,12 +55,12 @@ def all_reduce_(t: torch.Tensor, op: str = "sum", group=None, async_op: bool = F
     if not initialized() or world_size(group) == 1:
         return None
     ops = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
-    if _nccl(group) and not t.is_cuda:
-        # host tensor under RCCL: reduce a device copy, write back (sync)
+    if (_nccl(group) and not t.is_cuda) or (not _nccl(group) and t.is_cuda):
+        # backend/device mismatch: reduce a wire-device copy, write back
         tt = _as_real(t) if t.is_complex() else t
-        dev = tt.cuda()
+        dev = _to_wire_device(tt, group)
         dist.all_reduce(dev, op=ops[op], group=group)
-        tt.copy_(dev.cpu())
+        tt.copy_(dev.to(tt.device))
         return None
     if t.is_complex():
         assert op == "sum", "complex all-reduce supports sum only"
@@ -76,7 +80,7 @@ def all_gather_rows(local: torch.Tensor, counts: Sequence[int], group=None) -> t
     local_c = local.contiguous()
     cplx = local_c.is_complex()
     wire = torch.view_as_real(local_c) if cplx else local_c
-    host = not wire.is_cuda
+    orig_device = wire.device
     wire = _to_wire_device(wire, group)
     maxc = max(counts)
     if maxc == 0:
@@ -89,8 +93,8 @@ def all_gather_rows(local: torch.Tensor, counts: Sequence[int], group=None) -> t
     dist.all_gather(outs, pad, group=group)
     pieces = [outs[r][: counts[r]] for r in range(ws)]
     out = torch.cat(pieces, dim=0)
-    if host and out.is_cuda:
-        out = out.cpu()
+    if out.device != orig_device:
+        out = out.to(orig_device)
     if cplx:
         out = torch.view_as_complex(out)
     return out
@@ -99,7 +103,11 @@ def all_gather_rows(local: torch.Tensor, counts: Sequence[int], group=None) -> t
 def bcast_(t: torch.Tensor, src: int = 0, group=None) -> None:
     if not initialized() or world_size(group) == 1:
         return
-    dist.broadcast(torch.view_as_real(t) if t.is_complex() else t, src=src, group=group)
+    tt = torch.view_as_real(t) if t.is_complex() else t
+    wire = _to_wire_device(tt, group)
+    dist.broadcast(wire, src=src, group=group)
+    if wire.data_ptr() != tt.data_ptr():
+        tt.copy_(wire.to(tt.device))
 
 
 def all_to_all_v(send: List[torch.Tensor], group=None,
@@ -120,9 +128,11 @@ def all_to_all_v(send: List[torch.Tensor], group=None,
     wire = [(_as_real(s.contiguous())) for s in send]
     wdtype = wire[0].dtype
     backend = dist.get_backend(group)
-    host = not wire[0].is_cuda
-    if backend == "nccl" and host:
+    orig_device = wire[0].device
+    if backend == "nccl" and not wire[0].is_cuda:
         wire = [w.cuda() for w in wire]
+    elif backend != "nccl" and wire[0].is_cuda:
+        wire = [w.cpu() for w in wire]
     if recv_counts is None:
         cdev = wire[0].device if backend == "nccl" else torch.device("cpu")
         counts = torch.tensor([int(s.numel()) for s in wire], dtype=torch.int64, device=cdev)
@@ -146,8 +156,8 @@ def all_to_all_v(send: List[torch.Tensor], group=None,
         for req in dist.batch_isend_irecv(p2p):
             req.wait()
     recv[me].copy_(wire[me])
-    if host and backend == "nccl":
-        recv = [r.cpu() for r in recv]
+    if recv and recv[0].device != orig_device:
+        recv = [r.to(orig_device) for r in recv]
     if cplx:
         recv = [torch.view_as_complex(r.view(-1, 2)) for r in recv]
     return recv

@@ -110,19 +110,33 @@ class WindowGatherPlan:
         import torch.distributed as dist
 
         send_views, recv_views, hlo, own, hhi = self._halo_ctx(xlocal)
+        # gloo cannot move CUDA tensors: stage the halos through the host
+        # (the several-ranks-per-GPU battery configuration; RCCL runs the
+        # zero-copy device path below)
+        stage_host = (dist.get_backend(self.group) != "nccl"
+                      and xlocal.is_cuda)
         p2p = []
         me = comm.rank(self.group)
+        host_recv = {}
         for p in range(ws):
             if send_views[p] is not None and send_views[p].numel():
-                p2p.append(dist.P2POp(dist.isend, send_views[p].contiguous()
-                           if not send_views[p].is_contiguous() else send_views[p],
-                           p, group=self.group))
+                sv = send_views[p]
+                if not sv.is_contiguous():
+                    sv = sv.contiguous()
+                if stage_host:
+                    sv = sv.cpu()
+                p2p.append(dist.P2POp(dist.isend, sv, p, group=self.group))
             if recv_views[p] is not None:
-                p2p.append(dist.P2POp(dist.irecv, recv_views[p], p,
-                                      group=self.group))
+                rv = recv_views[p]
+                if stage_host:
+                    rv = torch.empty(rv.shape, dtype=rv.dtype)
+                    host_recv[p] = rv
+                p2p.append(dist.P2POp(dist.irecv, rv, p, group=self.group))
         if p2p:
             for req in dist.batch_isend_irecv(p2p):
                 req.wait()
+        for p, rv in host_recv.items():
+            recv_views[p].copy_(rv)
         return hlo, own, hhi
 
     def gather(self, xlocal: torch.Tensor) -> torch.Tensor:

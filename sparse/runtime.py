@@ -30,7 +30,8 @@ class Runtime:
 
         self.use_gpu = torch.cuda.is_available()
         if self.use_gpu:
-            local_rank = int(os.environ.get("LOCAL_RANK", self.rank % max(1, torch.cuda.device_count())))
+            local_rank = int(os.environ.get(
+                "LOCAL_RANK", self.rank)) % max(1, torch.cuda.device_count())
             torch.cuda.set_device(local_rank)
             self.device = torch.device("cuda", local_rank)
             # Side stream for halo/collective overlap with interior compute.
@@ -42,7 +43,11 @@ class Runtime:
 
     # -- distributed bring-up -------------------------------------------------
     def _init_dist_from_env(self) -> None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # SPARSE_DIST_BACKEND overrides (e.g. gloo on a GPU box: several
+        # ranks sharing one device — the distributed-battery-on-hardware
+        # test configuration)
+        backend = os.environ.get("SPARSE_DIST_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo")
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29571")
         dist.init_process_group(backend=backend)
