@@ -26,6 +26,29 @@ __global__ __launch_bounds__(256) void spmm_kernel(
   if (j < k) C[r * k + j] = acc;
 }
 
+// Small-k variant: lanes tile (rows x columns) so k < WAVE does not idle
+// 64-k lanes (k=8 would idle 87% of the wave in spmm_kernel).  kp =
+// next_pow2(k) lanes per row, WAVE/kp rows per wave.
+template <typename T, typename index_t>
+__global__ __launch_bounds__(256) void spmm_smallk_kernel(
+    const int64_t* __restrict__ indptr, const index_t* __restrict__ indices,
+    const T* __restrict__ vals, const T* __restrict__ B, T* __restrict__ C,
+    int64_t m, int64_t k, int64_t col_lo, int kp) {
+  const int lane = threadIdx.x;
+  const int rpw = WAVE / kp;
+  const int64_t r = ((int64_t)blockIdx.x * blockDim.y + threadIdx.y) * rpw
+                    + lane / kp;
+  const int64_t j = lane % kp;
+  if (r >= m) return;
+  T acc = ZeroOf<T>::value();
+  const int64_t e = indptr[r + 1];
+  for (int64_t p = indptr[r]; p < e; ++p) {
+    const int64_t c = (int64_t)indices[p] - col_lo;
+    if (j < k) acc += vals[p] * B[c * k + j];
+  }
+  if (j < k) C[r * k + j] = acc;
+}
+
 // C[i, c] += A[i, r] * v for each nz (r,c,v): one wave per nz, lanes over i
 template <typename T, typename index_t>
 __global__ void rspmm_kernel(const int64_t* __restrict__ indptr,
@@ -73,6 +96,18 @@ void spmm_hip(at::Tensor indptr, at::Tensor indices, at::Tensor vals,
   DISPATCH_VALUES(vals.scalar_type(), "spmm", [&] {
     using T = scalar_t;
     DISPATCH_INDEX(indices.scalar_type(), "spmm_idx", [&] {
+      if (k <= WAVE / 2) {
+        int kp = 1;
+        while (kp < k) kp <<= 1;
+        const int rpw = WAVE / kp;
+        dim3 block(WAVE, 4);
+        dim3 grid((m + (int64_t)4 * rpw - 1) / ((int64_t)4 * rpw));
+        hipLaunchKernelGGL((spmm_smallk_kernel<T, index_t>), grid, block, 0,
+                           cur_stream(), indptr.data_ptr<int64_t>(),
+                           indices.data_ptr<index_t>(), vals.data_ptr<T>(),
+                           B.data_ptr<T>(), C.data_ptr<T>(), m, k, col_lo, kp);
+        return;
+      }
       dim3 block(WAVE, 4);
       dim3 grid((k + WAVE - 1) / WAVE, (m + 3) / 4);
       hipLaunchKernelGGL((spmm_kernel<T, index_t>), grid, block, 0, cur_stream(),

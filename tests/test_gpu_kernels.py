@@ -63,11 +63,13 @@ def test_spmv_long_row_carry_gpu():
 
 
 @pytest.mark.parametrize("dt", types)
-def test_spmm_gpu(dt):
+@pytest.mark.parametrize("k", [3, 8, 33])
+def test_spmm_gpu(dt, k):
+    """k<=32 takes the small-k lane-tiled kernel; k=33 the wide kernel."""
     from sparse import csr_array
 
     s = sample_csr(300, 200, 0.05, seed=4, dtype=dt)
-    B = sample_dense((200, 33), seed=5, dtype=dt)
+    B = sample_dense((200, k), seed=5, dtype=dt)
     assert np.allclose(np.asarray(csr_array(s) @ B), s @ B, **tol(dt))
 
 
