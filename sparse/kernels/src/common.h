@@ -7,6 +7,8 @@
 //  - Complex arithmetic uses c10::complex<T> (device-ready in ROCm torch).
 #pragma once
 
+#include <type_traits>
+
 #include <hip/hip_runtime.h>
 
 #include <ATen/ATen.h>
@@ -88,4 +90,16 @@ __device__ __forceinline__ int64_t xcd_swizzle(int64_t bid, int64_t nwg) {
 
 inline hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+// Non-temporal load for single-use streams (value/index planes): keeps them
+// from evicting reusable lines (x windows, B rows) out of L2.  Complex
+// types fall back to plain loads (builtin needs scalar/vector types).
+template <typename T>
+__device__ __forceinline__ T nt_load(const T* __restrict__ p) {
+  if constexpr (std::is_arithmetic_v<T>) {
+    return __builtin_nontemporal_load(p);
+  } else {
+    return *p;
+  }
 }

@@ -20,8 +20,8 @@ __global__ __launch_bounds__(256) void spmm_kernel(
   T acc = ZeroOf<T>::value();
   int64_t e = indptr[r + 1];
   for (int64_t p = indptr[r]; p < e; ++p) {
-    int64_t c = (int64_t)indices[p] - col_lo;
-    if (j < k) acc += vals[p] * B[c * k + j];
+    int64_t c = (int64_t)nt_load(&indices[p]) - col_lo;
+    if (j < k) acc += nt_load(&vals[p]) * B[c * k + j];
   }
   if (j < k) C[r * k + j] = acc;
 }
@@ -43,8 +43,8 @@ __global__ __launch_bounds__(256) void spmm_smallk_kernel(
   T acc = ZeroOf<T>::value();
   const int64_t e = indptr[r + 1];
   for (int64_t p = indptr[r]; p < e; ++p) {
-    const int64_t c = (int64_t)indices[p] - col_lo;
-    if (j < k) acc += vals[p] * B[c * k + j];
+    const int64_t c = (int64_t)nt_load(&indices[p]) - col_lo;
+    if (j < k) acc += nt_load(&vals[p]) * B[c * k + j];
   }
   if (j < k) C[r * k + j] = acc;
 }

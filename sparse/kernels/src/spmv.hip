@@ -297,19 +297,9 @@ __global__ void build_ell_kernel(const int64_t* __restrict__ indptr,
   }
 }
 
-// Non-temporal load for single-use streams (values/index planes): keeps
-// them from evicting the reusable x window lines out of L2.  Measured on
-// the 16384^2 5-pt DIA SpMV: 2.580 -> 2.474 ms (tools/dia_nt_bench.hip);
-// nt STORES measured slower, so y is written with plain stores.  Complex
-// types fall back to plain loads (builtin needs scalar/vector types).
-template <typename T>
-__device__ __forceinline__ T nt_load(const T* __restrict__ p) {
-  if constexpr (std::is_arithmetic_v<T>) {
-    return __builtin_nontemporal_load(p);
-  } else {
-    return *p;
-  }
-}
+// nt_load (common.h): measured on the 16384^2 5-pt DIA SpMV:
+// 2.580 -> 2.474 ms (tools/dia_nt_bench.hip); nt STORES measured slower,
+// so outputs use plain stores.
 
 // Window pair load [c0, c0+1] from a single piece (SINGLE mode), clamped
 // at the edges.  Rows r0, r0+1 of one thread access consecutive columns on
