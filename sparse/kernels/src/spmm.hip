@@ -20,14 +20,17 @@ __global__ __launch_bounds__(256) void spmm_kernel(
   T acc = ZeroOf<T>::value();
   int64_t e = indptr[r + 1];
   for (int64_t p = indptr[r]; p < e; ++p) {
-    int64_t c = (int64_t)nt_load(&indices[p]) - col_lo;
-    if (j < k) acc += nt_load(&vals[p]) * B[c * k + j];
+    int64_t c = (int64_t)indices[p] - col_lo;
+    if (j < k) acc += vals[p] * B[c * k + j];
   }
   if (j < k) C[r * k + j] = acc;
 }
 
 // Small-k variant: lanes tile (rows x columns) so k < WAVE does not idle
-// 64-k lanes (k=8 would idle 87% of the wave in spmm_kernel).  kp =
+// 64-k lanes (k=8 would idle 87% of the wave in spmm_kernel).  NOTE:
+// nt loads on vals/indices measured 20-50% SLOWER here (unlike SpMV) —
+// the same val/index address is read by all kp lanes of the wave, and the
+// nontemporal policy defeats the cache broadcast that makes that free.  kp =
 // next_pow2(k) lanes per row, WAVE/kp rows per wave.
 template <typename T, typename index_t>
 __global__ __launch_bounds__(256) void spmm_smallk_kernel(
@@ -43,8 +46,8 @@ __global__ __launch_bounds__(256) void spmm_smallk_kernel(
   T acc = ZeroOf<T>::value();
   const int64_t e = indptr[r + 1];
   for (int64_t p = indptr[r]; p < e; ++p) {
-    const int64_t c = (int64_t)nt_load(&indices[p]) - col_lo;
-    if (j < k) acc += nt_load(&vals[p]) * B[c * k + j];
+    const int64_t c = (int64_t)indices[p] - col_lo;
+    if (j < k) acc += vals[p] * B[c * k + j];
   }
   if (j < k) C[r * k + j] = acc;
 }
