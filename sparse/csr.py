@@ -12,6 +12,8 @@ from __future__ import annotations
 import numbers
 from typing import Optional
 
+import os
+
 import numpy as np
 import torch
 
@@ -222,6 +224,20 @@ class csr_array(CompressedBase, DenseSparseBase):
                 self._window_cache = (int(self._indices.min().item()),
                                       int(self._indices.max().item()) + 1)
         return self._window_cache
+
+    def _dia_interior(self, dm, plan):
+        """Even row bounds [a, b) of the interior — rows whose whole
+        window falls inside the own piece, safe to compute while the halo
+        exchange is in flight (overlap of comm with interior compute)."""
+        me = comm.rank()
+        xs, xe = plan.xpart.start(me), plan.xpart.stop(me)
+        own_a, own_b = max(plan.lo, xs), min(plan.hi, xe)
+        row0 = self.partition.start(me)
+        a = min(max(0, own_a - row0 - dm.off_min), dm.m)
+        b = min(max(a, own_b - row0 - dm.off_max), dm.m)
+        a = (a + 1) // 2 * 2
+        b = max(a, b // 2 * 2)
+        return a, b
 
     def _dia(self):
         """Cached diagonal mirror (GPU banded fast SpMV; kernels.build_dia —
@@ -435,10 +451,20 @@ class csr_array(CompressedBase, DenseSparseBase):
         if dm is not None:
             from . import kernels
 
-            pieces = plan.gather_halos(x.local.to(vdt))
             ylocal = torch.empty(self.partition.count(comm.rank()), dtype=vdt,
                                  device=self._values.device)
-            kernels.dia_spmv(dm, pieces, ylocal, plan.lo, plan.hi - plan.lo)
+            ws_ = plan.hi - plan.lo
+            if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
+                h = plan.gather_halos_begin(x.local.to(vdt))
+                a, bnd = self._dia_interior(dm, plan)
+                kernels.dia_spmv(dm, plan.handle_pieces(h), ylocal, plan.lo,
+                                 ws_, a, bnd)
+                pieces = plan.gather_halos_end(h)
+                kernels.dia_spmv(dm, pieces, ylocal, plan.lo, ws_, 0, a)
+                kernels.dia_spmv(dm, pieces, ylocal, plan.lo, ws_, bnd, -1)
+            else:
+                pieces = plan.gather_halos(x.local.to(vdt))
+                kernels.dia_spmv(dm, pieces, ylocal, plan.lo, ws_)
         elif ell is not None:
             from . import kernels
 
@@ -497,9 +523,22 @@ class csr_array(CompressedBase, DenseSparseBase):
             from . import kernels
 
             plan = self._xplan(x.partition)
-            pieces = plan.gather_halos(x.local)
-            kernels.dia_jacobi(dm, pieces, x.local, b.local, dinv.local,
-                               omega, out.local, plan.lo, plan.hi - plan.lo)
+            ws_ = plan.hi - plan.lo
+            if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
+                h = plan.gather_halos_begin(x.local)
+                a, bnd = self._dia_interior(dm, plan)
+                kernels.dia_jacobi(dm, plan.handle_pieces(h), x.local,
+                                   b.local, dinv.local, omega, out.local,
+                                   plan.lo, ws_, a, bnd)
+                pieces = plan.gather_halos_end(h)
+                kernels.dia_jacobi(dm, pieces, x.local, b.local, dinv.local,
+                                   omega, out.local, plan.lo, ws_, 0, a)
+                kernels.dia_jacobi(dm, pieces, x.local, b.local, dinv.local,
+                                   omega, out.local, plan.lo, ws_, bnd, -1)
+            else:
+                pieces = plan.gather_halos(x.local)
+                kernels.dia_jacobi(dm, pieces, x.local, b.local, dinv.local,
+                                   omega, out.local, plan.lo, ws_)
             return out
         if ell is not None:
             from . import kernels
@@ -517,7 +556,14 @@ class csr_array(CompressedBase, DenseSparseBase):
 
     def spmv_dot(self, p: DistArray, q: DistArray) -> torch.Tensor:
         """Fused q = A@p and all-reduced sum(p*q) — the CG p·Ap in one kernel
-        (GPU, real dtypes; MI355X fusion: saves re-reading p and q)."""
+        (GPU, real dtypes; MI355X fusion: saves re-reading p and q).
+        CPU: eager SpMV + dot (same semantics)."""
+        if not self._values.is_cuda:
+            self._spmv(p, out=q)
+            dot = torch.dot(q.local, p.local) if q.local.numel() else \
+                torch.zeros((), dtype=q.local.dtype)
+            comm.all_reduce_(dot)
+            return dot
         from . import kernels
 
         plan = self._xplan(p.partition)
@@ -526,9 +572,21 @@ class csr_array(CompressedBase, DenseSparseBase):
         ell = None if dm is not None else (
             self._ell() if not precise else None)
         if dm is not None:
-            pieces = plan.gather_halos(p.local)
-            dot = kernels.dia_spmv_dot(dm, pieces, q.local, p.local, plan.lo,
-                                       plan.hi - plan.lo)
+            ws_ = plan.hi - plan.lo
+            if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
+                h = plan.gather_halos_begin(p.local)
+                a, bnd = self._dia_interior(dm, plan)
+                dot = kernels.dia_spmv_dot(dm, plan.handle_pieces(h), q.local,
+                                           p.local, plan.lo, ws_, a, bnd)
+                pieces = plan.gather_halos_end(h)
+                dot = dot + kernels.dia_spmv_dot(dm, pieces, q.local, p.local,
+                                                 plan.lo, ws_, 0, a)
+                dot = dot + kernels.dia_spmv_dot(dm, pieces, q.local, p.local,
+                                                 plan.lo, ws_, bnd, -1)
+            else:
+                pieces = plan.gather_halos(p.local)
+                dot = kernels.dia_spmv_dot(dm, pieces, q.local, p.local,
+                                           plan.lo, ws_)
         elif ell is not None:
             pieces = plan.gather_halos(p.local)
             dot = kernels.ell_spmv_dot(ell, pieces, q.local, p.local, plan.lo)

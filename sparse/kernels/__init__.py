@@ -260,9 +260,11 @@ class DiaMirror:
     """Padded diagonal planes of a banded CSR slab: values only, no index
     stream (12 -> 8 B/nnz for fp64+int32 vs ELL)."""
 
-    __slots__ = ("dvals", "offs", "W", "m", "row0")
+    __slots__ = ("dvals", "offs", "W", "m", "row0", "off_min", "off_max")
 
-    def __init__(self, dvals, offs, W, m, row0):
+    def __init__(self, dvals, offs, W, m, row0, off_min=0, off_max=0):
+        self.off_min = off_min
+        self.off_max = off_max
         self.dvals = dvals
         self.offs = offs
         self.W = W
@@ -292,24 +294,33 @@ def build_dia(A, row0: int):
     dvals = torch.zeros(W * mp, dtype=A.values.dtype, device=A.device)
     k_idx = torch.searchsorted(offs, diag)
     dvals[k_idx * mp + (rows - row0)] = A.values
-    return DiaMirror(dvals, offs, W, m, row0)
+    return DiaMirror(dvals, offs, W, m, row0,
+                     off_min=int(offs[0].item()), off_max=int(offs[-1].item()))
 
 
-def dia_spmv(dm: DiaMirror, pieces, y, col_lo: int, wsize: int):
+def dia_spmv(dm: DiaMirror, pieces, y, col_lo: int, wsize: int,
+             rbase: int = 0, rhi: int = -1):
+    """rbase/rhi select a row sub-range (rbase even; -1 = all rows): the
+    interior/boundary split that overlaps halo exchange with interior
+    compute at ws>1."""
     hlo, own, hhi = pieces
     ext().dia_spmv(dm.dvals, dm.offs, hlo.contiguous(), own.contiguous(),
                    hhi.contiguous(), y, dm.W, dm.m, int(col_lo), dm.row0,
-                   int(wsize))
+                   int(wsize), int(rbase), int(rhi))
 
 
-def dia_spmv_dot(dm: DiaMirror, pieces, y, p, col_lo: int, wsize: int):
+def dia_spmv_dot(dm: DiaMirror, pieces, y, p, col_lo: int, wsize: int,
+                 rbase: int = 0, rhi: int = -1):
     hlo, own, hhi = pieces
     mp = dm.dvals.numel() // dm.W
-    nblocks = (mp // 2 + 255) // 256
+    hi = mp if rhi < 0 else rhi
+    nblocks = max(1, ((hi - rbase) // 2 + 256) // 256)
     partial = torch.empty(nblocks, dtype=dm.dvals.dtype, device=dm.dvals.device)
+    if hi <= rbase:
+        return partial[:1].zero_().sum()
     ext().dia_spmv_dot(dm.dvals, dm.offs, hlo.contiguous(), own.contiguous(),
                        hhi.contiguous(), y, p, partial, dm.W, dm.m,
-                       int(col_lo), dm.row0, int(wsize))
+                       int(col_lo), dm.row0, int(wsize), int(rbase), int(hi))
     return partial.sum()
 
 
@@ -342,11 +353,12 @@ def cg_xr_norm2(x, p, r, q, a, b):
 
 
 def dia_jacobi(dm: DiaMirror, pieces, xloc, b, dinv, omega, xout,
-               col_lo: int, wsize: int):
+               col_lo: int, wsize: int, rbase: int = 0, rhi: int = -1):
     hlo, own, hhi = pieces
     ext().dia_jacobi(dm.dvals, dm.offs, hlo.contiguous(), own.contiguous(),
                      hhi.contiguous(), xloc, b, dinv, xout, dm.W, dm.m,
-                     int(col_lo), dm.row0, int(wsize), float(omega))
+                     int(col_lo), dm.row0, int(wsize), float(omega),
+                     int(rbase), int(rhi))
 
 
 # -- ELL fast path ------------------------------------------------------------

@@ -34,14 +34,15 @@ void dia_spmv_bpdot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                         int64_t);
 void dia_spmv_plain_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                         at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
-                        int64_t, int64_t);
+                        int64_t, int64_t, int64_t, int64_t);
 void dia_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                      int64_t, int64_t, int64_t, int64_t, int64_t);
+                      int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
+                      int64_t);
 void dia_jacobi_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, int64_t, int64_t, int64_t, int64_t, int64_t,
-                    double);
+                    double, int64_t, int64_t);
 void add_nnz_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void add_compute_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                      at::Tensor, at::Tensor, at::Tensor, at::Tensor, double,
@@ -113,13 +114,15 @@ TORCH_LIBRARY(sparse_hip, m) {
         "int wsize) -> ()");
   m.def("dia_spmv(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor(a!) y, int W, int m, int col_lo, int row0, "
-        "int wsize) -> ()");
+        "int wsize, int rbase, int rhi) -> ()");
   m.def("dia_spmv_dot(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor(a!) y, Tensor pvec, Tensor(b!) dot_partial, "
-        "int W, int m, int col_lo, int row0, int wsize) -> ()");
+        "int W, int m, int col_lo, int row0, int wsize, int rbase, "
+        "int rhi) -> ()");
   m.def("dia_jacobi(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor xloc, Tensor b, Tensor dinv, Tensor(a!) xout, "
-        "int W, int m, int col_lo, int row0, int wsize, float omega) -> ()");
+        "int W, int m, int col_lo, int row0, int wsize, float omega, "
+        "int rbase, int rhi) -> ()");
   m.def("spmm(Tensor indptr, Tensor indices, Tensor vals, Tensor B, "
         "Tensor(a!) C, int col_lo) -> ()");
   m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "

@@ -601,13 +601,15 @@ __global__ __launch_bounds__(BLK) void dia_spmv_kernel(
     const T* __restrict__ hhi, T* __restrict__ y,
     const T* __restrict__ pvec, T* __restrict__ dot_partial,
     int64_t m, int64_t mp, int W, int64_t col_lo, int64_t row0,
-    int64_t nlo, int64_t nown, int64_t wsize) {
+    int64_t nlo, int64_t nown, int64_t wsize, int64_t rbase, int64_t rhi) {
+  // [rbase, rhi): row sub-range (rbase even) — the interior/boundary split
+  // that overlaps halo exchange with interior compute at ws>1
   __shared__ __align__(16) char red_raw[BLK * sizeof(T)];
   T* red = reinterpret_cast<T*>(red_raw);
   const int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
-  const int64_t r0 = 2 * t;
+  const int64_t r0 = rbase + 2 * t;
   T a0 = ZeroOf<T>::value(), a1 = ZeroOf<T>::value();
-  if (r0 < mp) {
+  if (r0 < rhi) {
     for (int k = 0; k < W; ++k) {
       const int64_t base = (int64_t)k * mp + r0;
       struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
@@ -628,7 +630,7 @@ __global__ __launch_bounds__(BLK) void dia_spmv_kernel(
         a1 += vv.b * xpiece(i1, hlo, nlo, own, nown, hhi);
       }
     }
-    if (r0 + 1 < m) {
+    if (r0 + 1 < min(m, rhi)) {
       struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
       TP out{a0, a1};
       *reinterpret_cast<TP*>(&y[r0]) = out;
@@ -638,8 +640,8 @@ __global__ __launch_bounds__(BLK) void dia_spmv_kernel(
   }
   if (FUSE_DOT) {
     T d = ZeroOf<T>::value();
-    if (r0 < m) d += a0 * pvec[r0];
-    if (r0 + 1 < m) d += a1 * pvec[r0 + 1];
+    if (r0 < min(m, rhi)) d += a0 * pvec[r0];
+    if (r0 + 1 < min(m, rhi)) d += a1 * pvec[r0 + 1];
     red[threadIdx.x] = d;
     __syncthreads();
     for (int w = BLK / 2; w > 0; w >>= 1) {
@@ -657,10 +659,11 @@ __global__ __launch_bounds__(BLK) void dia_jacobi_kernel(
     const T* __restrict__ hhi, const T* __restrict__ xloc,
     const T* __restrict__ b, const T* __restrict__ dinv,
     T* __restrict__ xout, int64_t m, int64_t mp, int W, int64_t col_lo,
-    int64_t row0, int64_t nlo, int64_t nown, int64_t wsize, T omega) {
+    int64_t row0, int64_t nlo, int64_t nown, int64_t wsize, T omega,
+    int64_t rbase, int64_t rhi) {
   const int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
-  const int64_t r0 = 2 * t;
-  if (r0 >= mp) return;
+  const int64_t r0 = rbase + 2 * t;
+  if (r0 >= rhi) return;
   T a0 = ZeroOf<T>::value(), a1 = ZeroOf<T>::value();
   for (int k = 0; k < W; ++k) {
     const int64_t base = (int64_t)k * mp + r0;
@@ -681,7 +684,7 @@ __global__ __launch_bounds__(BLK) void dia_jacobi_kernel(
       a1 += vv.b * xpiece(i1, hlo, nlo, own, nown, hhi);
     }
   }
-  if (r0 + 1 < m) {
+  if (r0 + 1 < min(m, rhi)) {
     struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
     const TP xv = *reinterpret_cast<const TP*>(&xloc[r0]);
     const TP bv = *reinterpret_cast<const TP*>(&b[r0]);
@@ -829,10 +832,13 @@ void dia_spmv_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
                   int64_t W, int64_t m, int64_t col_lo, int64_t row0,
                   int64_t wsize,
                   const c10::optional<at::Tensor>& pvec,
-                  const c10::optional<at::Tensor>& dot_partial) {
+                  const c10::optional<at::Tensor>& dot_partial,
+                  int64_t rbase, int64_t rhi) {
   const int64_t mp = dvals.numel() / W;
+  if (rhi < 0) rhi = mp;
+  if (rhi <= rbase) return;
   const bool fuse = pvec.has_value();
-  const int64_t nblocks = (mp / 2 + BLK - 1) / BLK;
+  const int64_t nblocks = ((rhi - rbase) / 2 + BLK) / BLK;
   const int64_t nlo = hlo.numel();
   const int64_t nown = own.numel();
   DISPATCH_VALUES(dvals.scalar_type(), "dia_spmv", [&] {
@@ -844,7 +850,8 @@ void dia_spmv_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
       hipLaunchKernelGGL(kern, dim3(nblocks), dim3(BLK), 0, cur_stream(),
                          dvals.data_ptr<T>(), offs.data_ptr<int64_t>(), hlo_p,
                          own.data_ptr<T>(), hhi_p, y.data_ptr<T>(), pv, dp,
-                         m, mp, (int)W, col_lo, row0, nlo, nown, wsize);
+                         m, mp, (int)W, col_lo, row0, nlo, nown, wsize,
+                         rbase, rhi);
     };
     if (fuse && single)
       launch(dia_spmv_kernel<T, true, true>, pvec->data_ptr<T>(),
@@ -862,26 +869,29 @@ void dia_spmv_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
 void dia_spmv_plain_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
                         at::Tensor own, at::Tensor hhi, at::Tensor y,
                         int64_t W, int64_t m, int64_t col_lo, int64_t row0,
-                        int64_t wsize) {
+                        int64_t wsize, int64_t rbase, int64_t rhi) {
   dia_spmv_hip(dvals, offs, hlo, own, hhi, y, W, m, col_lo, row0, wsize,
-               c10::nullopt, c10::nullopt);
+               c10::nullopt, c10::nullopt, rbase, rhi);
 }
 
 void dia_spmv_dot_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
                       at::Tensor own, at::Tensor hhi, at::Tensor y,
                       at::Tensor pvec, at::Tensor dot_partial, int64_t W,
-                      int64_t m, int64_t col_lo, int64_t row0, int64_t wsize) {
+                      int64_t m, int64_t col_lo, int64_t row0, int64_t wsize,
+                      int64_t rbase, int64_t rhi) {
   dia_spmv_hip(dvals, offs, hlo, own, hhi, y, W, m, col_lo, row0, wsize,
-               pvec, dot_partial);
+               pvec, dot_partial, rbase, rhi);
 }
 
 void dia_jacobi_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
                     at::Tensor own, at::Tensor hhi, at::Tensor xloc,
                     at::Tensor b, at::Tensor dinv, at::Tensor xout,
                     int64_t W, int64_t m, int64_t col_lo, int64_t row0,
-                    int64_t wsize, double omega) {
+                    int64_t wsize, double omega, int64_t rbase, int64_t rhi) {
   const int64_t mp = dvals.numel() / W;
-  const int64_t nblocks = (mp / 2 + BLK - 1) / BLK;
+  if (rhi < 0) rhi = mp;
+  if (rhi <= rbase) return;
+  const int64_t nblocks = ((rhi - rbase) / 2 + BLK) / BLK;
   const int64_t nlo = hlo.numel();
   const int64_t nown = own.numel();
   DISPATCH_VALUES(dvals.scalar_type(), "dia_jacobi", [&] {
@@ -895,7 +905,7 @@ void dia_jacobi_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
                          own.data_ptr<T>(), hhi_p, xloc.data_ptr<T>(),
                          b.data_ptr<T>(), dinv.data_ptr<T>(),
                          xout.data_ptr<T>(), m, mp, (int)W, col_lo, row0,
-                         nlo, nown, wsize, static_cast<T>(omega));
+                         nlo, nown, wsize, static_cast<T>(omega), rbase, rhi);
     };
     if (single) launch(dia_jacobi_kernel<T, true>);
     else launch(dia_jacobi_kernel<T, false>);

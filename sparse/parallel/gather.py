@@ -100,13 +100,17 @@ class WindowGatherPlan:
         self._ctx_cache[key] = ctx
         return ctx
 
-    def gather_halos(self, xlocal: torch.Tensor):
-        """Exchange ONLY the halo pieces; my own slab portion is used in
-        place (no self-copy through the collective).  Returns
-        (halo_lo, own_view, halo_hi)."""
+    def gather_halos_begin(self, xlocal: torch.Tensor):
+        """Start the halo exchange without waiting: returns an opaque
+        handle for gather_halos_end.  Interior rows (which never read the
+        halo pieces) can compute while the exchange is in flight — the
+        MI355X realization of the reference's Legion-overlapped gathers.
+        The pieces in the handle are valid for interior use immediately
+        (halo buffers unfilled until _end)."""
         ws = comm.world_size(self.group)
         if ws == 1:
-            return xlocal[:0], xlocal[self.lo: self.hi], xlocal[:0]
+            return ("done", (xlocal[:0], xlocal[self.lo: self.hi],
+                             xlocal[:0]))
         import torch.distributed as dist
 
         send_views, recv_views, hlo, own, hhi = self._halo_ctx(xlocal)
@@ -116,7 +120,6 @@ class WindowGatherPlan:
         stage_host = (dist.get_backend(self.group) != "nccl"
                       and xlocal.is_cuda)
         p2p = []
-        me = comm.rank(self.group)
         host_recv = {}
         for p in range(ws):
             if send_views[p] is not None and send_views[p].numel():
@@ -132,12 +135,32 @@ class WindowGatherPlan:
                     rv = torch.empty(rv.shape, dtype=rv.dtype)
                     host_recv[p] = rv
                 p2p.append(dist.P2POp(dist.irecv, rv, p, group=self.group))
-        if p2p:
-            for req in dist.batch_isend_irecv(p2p):
-                req.wait()
+        reqs = dist.batch_isend_irecv(p2p) if p2p else []
+        return ("pending", reqs, host_recv, recv_views, (hlo, own, hhi))
+
+    @staticmethod
+    def handle_pieces(handle):
+        """The (hlo, own, hhi) views of an in-flight handle — valid for
+        INTERIOR rows only until gather_halos_end returns."""
+        return handle[1] if handle[0] == "done" else handle[4]
+
+    def gather_halos_end(self, handle):
+        """Complete an exchange started by gather_halos_begin; returns the
+        (halo_lo, own_view, halo_hi) pieces with halos filled."""
+        if handle[0] == "done":
+            return handle[1]
+        _, reqs, host_recv, recv_views, pieces = handle
+        for req in reqs:
+            req.wait()
         for p, rv in host_recv.items():
             recv_views[p].copy_(rv)
-        return hlo, own, hhi
+        return pieces
+
+    def gather_halos(self, xlocal: torch.Tensor):
+        """Exchange ONLY the halo pieces; my own slab portion is used in
+        place (no self-copy through the collective).  Returns
+        (halo_lo, own_view, halo_hi)."""
+        return self.gather_halos_end(self.gather_halos_begin(xlocal))
 
     def gather(self, xlocal: torch.Tensor) -> torch.Tensor:
         """Return the window x[lo:hi) (dim 0 slices; works for 1-D and 2-D)."""
