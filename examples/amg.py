@@ -247,8 +247,7 @@ def vcycle(levels, li, b):
     x = b * dinv
     x.local.mul_(omega)
     x = A.jacobi_smooth(x, b, dinv, omega)
-    r = A.dot(x)
-    r.local.sub_(b.local).neg_()
+    r = A.residual(x, b)
     rc = lvl["R"].dot(r)
     xc = vcycle(levels, li + 1, rc)
     x += lvl["P"].dot(xc)

@@ -115,8 +115,7 @@ class GMG:
         x = b * lvl.dinv
         x.local.mul_(lvl.omega)  # pre-smooth from zero
         x = self._smooth(lvl, x, b, self.smooth_iters - 1)
-        r = lvl.A.dot(x)
-        r.local.sub_(b.local).neg_()
+        r = lvl.A.residual(x, b)
         P = lvl.Pdown
         rc = lvl.Rdown.dot(r)  # restriction (CSC col-split SpMV w/ reduction)
         xc = self._vcycle(li + 1, rc)

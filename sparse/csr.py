@@ -554,6 +554,45 @@ class csr_array(CompressedBase, DenseSparseBase):
         out.local.addcmul_(r.local, dinv.local, value=omega)
         return out
 
+    def residual(self, x: DistArray, b: DistArray,
+                 out: Optional[DistArray] = None) -> DistArray:
+        """r = b - A @ x in ONE kernel on the DIA fast path (the V-cycle
+        residual; saves the separate 3-pass pointwise subtract of the
+        naive r = A.dot(x); r -= b; r.neg_()).  Falls back to SpMV +
+        pointwise elsewhere.  Reference context: the implicit copy+ADD
+        chain of the reference's GMG residual (examples/gmg.py:118-119
+        there)."""
+        dm = self._dia()
+        if out is None:
+            out = DistArray.from_local(
+                torch.empty(self.partition.count(comm.rank()),
+                            dtype=self._values.dtype,
+                            device=self._values.device),
+                self.partition, (self.shape[0],))
+        if dm is not None:
+            from . import kernels
+
+            plan = self._xplan(x.partition)
+            ws_ = plan.hi - plan.lo
+            if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
+                h = plan.gather_halos_begin(x.local)
+                a, bnd = self._dia_interior(dm, plan)
+                kernels.dia_residual(dm, plan.handle_pieces(h), b.local,
+                                     out.local, plan.lo, ws_, a, bnd)
+                pieces = plan.gather_halos_end(h)
+                kernels.dia_residual(dm, pieces, b.local, out.local,
+                                     plan.lo, ws_, 0, a)
+                kernels.dia_residual(dm, pieces, b.local, out.local,
+                                     plan.lo, ws_, bnd, -1)
+            else:
+                pieces = plan.gather_halos(x.local)
+                kernels.dia_residual(dm, pieces, b.local, out.local,
+                                     plan.lo, ws_)
+            return out
+        self._spmv(x, out=out)
+        out.local.sub_(b.local).neg_()
+        return out
+
     def spmv_dot(self, p: DistArray, q: DistArray) -> torch.Tensor:
         """Fused q = A@p and all-reduced sum(p*q) — the CG p·Ap in one kernel
         (GPU, real dtypes; MI355X fusion: saves re-reading p and q).

@@ -352,6 +352,15 @@ def cg_xr_norm2(x, p, r, q, a, b):
     return partial.sum()
 
 
+def dia_residual(dm: DiaMirror, pieces, b, y, col_lo: int, wsize: int,
+                 rbase: int = 0, rhi: int = -1):
+    """Fused V-cycle residual y = b - A@x (x given as window pieces)."""
+    hlo, own, hhi = pieces
+    ext().dia_residual(dm.dvals, dm.offs, hlo.contiguous(), own.contiguous(),
+                       hhi.contiguous(), b, y, dm.W, dm.m, int(col_lo),
+                       dm.row0, int(wsize), int(rbase), int(rhi))
+
+
 def dia_jacobi(dm: DiaMirror, pieces, xloc, b, dinv, omega, xout,
                col_lo: int, wsize: int, rbase: int = 0, rhi: int = -1):
     hlo, own, hhi = pieces

@@ -39,6 +39,9 @@ void dia_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
                       int64_t);
+void dia_residual_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                      at::Tensor, at::Tensor, at::Tensor, int64_t, int64_t,
+                      int64_t, int64_t, int64_t, int64_t, int64_t);
 void dia_jacobi_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, int64_t, int64_t, int64_t, int64_t, int64_t,
@@ -119,6 +122,9 @@ TORCH_LIBRARY(sparse_hip, m) {
         "Tensor hhi, Tensor(a!) y, Tensor pvec, Tensor(b!) dot_partial, "
         "int W, int m, int col_lo, int row0, int wsize, int rbase, "
         "int rhi) -> ()");
+  m.def("dia_residual(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
+        "Tensor hhi, Tensor b, Tensor(a!) y, int W, int m, int col_lo, "
+        "int row0, int wsize, int rbase, int rhi) -> ()");
   m.def("dia_jacobi(Tensor dvals, Tensor offs, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor xloc, Tensor b, Tensor dinv, Tensor(a!) xout, "
         "int W, int m, int col_lo, int row0, int wsize, float omega, "
@@ -167,6 +173,7 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("dia_spmv_bpdot", dia_spmv_bpdot_hip);
   m.impl("dia_spmv", dia_spmv_plain_hip);
   m.impl("dia_spmv_dot", dia_spmv_dot_hip);
+  m.impl("dia_residual", dia_residual_hip);
   m.impl("dia_jacobi", dia_jacobi_hip);
   m.impl("spmm", spmm_hip);
   m.impl("rspmm", rspmm_hip);

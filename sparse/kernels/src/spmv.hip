@@ -593,7 +593,7 @@ void csr_row_spmv_hip(at::Tensor indptr, at::Tensor indices, at::Tensor values,
 // banded operators (Poisson, dot_microbenchmark).
 namespace {
 
-template <typename T, bool FUSE_DOT, bool SINGLE>
+template <typename T, bool FUSE_DOT, bool SINGLE, bool RESID = false>
 __global__ __launch_bounds__(BLK) void dia_spmv_kernel(
     const T* __restrict__ dvals,  // (W, mp) column-major planes
     const int64_t* __restrict__ offs,  // W diagonal offsets
@@ -633,9 +633,16 @@ __global__ __launch_bounds__(BLK) void dia_spmv_kernel(
     if (r0 + 1 < min(m, rhi)) {
       struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
       TP out{a0, a1};
+      if (RESID) {
+        // y = b - A x (pvec carries b): the V-cycle residual fused into
+        // the SpMV — saves the separate 3-pass pointwise kernel
+        const TP bv = *reinterpret_cast<const TP*>(&pvec[r0]);
+        out.a = bv.a - a0;
+        out.b = bv.b - a1;
+      }
       *reinterpret_cast<TP*>(&y[r0]) = out;
     } else if (r0 < m) {
-      y[r0] = a0;
+      y[r0] = RESID ? (pvec[r0] - a0) : a0;
     }
   }
   if (FUSE_DOT) {
@@ -863,6 +870,35 @@ void dia_spmv_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
       launch(dia_spmv_kernel<T, false, true>, nullptr, nullptr);
     else
       launch(dia_spmv_kernel<T, false, false>, nullptr, nullptr);
+  });
+}
+
+void dia_residual_hip(at::Tensor dvals, at::Tensor offs, at::Tensor hlo,
+                      at::Tensor own, at::Tensor hhi, at::Tensor b,
+                      at::Tensor y, int64_t W, int64_t m, int64_t col_lo,
+                      int64_t row0, int64_t wsize, int64_t rbase,
+                      int64_t rhi) {
+  // y = b - A x: the fused V-cycle residual (pvec slot carries b)
+  const int64_t mp = dvals.numel() / W;
+  if (rhi < 0) rhi = mp;
+  if (rhi <= rbase) return;
+  const int64_t nblocks = ((rhi - rbase) / 2 + BLK) / BLK;
+  const int64_t nlo = hlo.numel();
+  const int64_t nown = own.numel();
+  DISPATCH_VALUES(dvals.scalar_type(), "dia_residual", [&] {
+    using T = scalar_t;
+    const T* hlo_p = nlo ? hlo.data_ptr<T>() : own.data_ptr<T>();
+    const T* hhi_p = hhi.numel() ? hhi.data_ptr<T>() : own.data_ptr<T>();
+    const bool single = (nlo == 0 && hhi.numel() == 0);
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(nblocks), dim3(BLK), 0, cur_stream(),
+                         dvals.data_ptr<T>(), offs.data_ptr<int64_t>(), hlo_p,
+                         own.data_ptr<T>(), hhi_p, y.data_ptr<T>(),
+                         b.data_ptr<T>(), nullptr, m, mp, (int)W, col_lo,
+                         row0, nlo, nown, wsize, rbase, rhi);
+    };
+    if (single) launch(dia_spmv_kernel<T, false, true, true>);
+    else launch(dia_spmv_kernel<T, false, false, true>);
   });
 }
 

@@ -561,3 +561,29 @@ def test_precise_plan_gpu():
         got[r] = (A._values[sl] * xw[remapped[sl].long()]).sum()
     ref = s @ x.cpu().numpy()
     assert np.allclose(got.cpu().numpy(), ref, rtol=1e-10)
+
+
+def test_dia_residual_gpu():
+    """Fused r = b - A@x on the DIA path vs the eager reference."""
+    from sparse import darray, gallery
+
+    n = 5000
+    A = gallery.banded(n, ndiags=9)
+    assert A._dia() is not None
+    x = darray.random((n,), seed=95)
+    b = darray.random((n,), seed=96)
+    r = A.residual(x, b)
+    sref = A.to_scipy_sparse_csr()
+    expect = np.asarray(b) - sref @ np.asarray(x)
+    assert np.allclose(np.asarray(r), expect, rtol=1e-12)
+    # scattered matrix takes the fallback path
+    s = sample_csr(800, 800, 0.01, seed=97)
+    from sparse import csr_array
+
+    A2 = csr_array(s)
+    assert A2._dia() is None
+    x2 = darray.random((800,), seed=98)
+    b2 = darray.random((800,), seed=99)
+    r2 = A2.residual(x2, b2)
+    assert np.allclose(np.asarray(r2), np.asarray(b2) - s @ np.asarray(x2),
+                       rtol=1e-10)
