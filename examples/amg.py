@@ -186,12 +186,22 @@ def build_hierarchy(A: csr_array, theta: float, max_coarse: int):
             break
         if nc >= cur.shape[0]:
             break
-        # tentative prolongator: T[i, agg[i]] = 1, column-normalized
-        counts = np.bincount(agg, minlength=nc).astype(np.float64)
-        T = sps.csr_matrix((1.0 / np.sqrt(counts[agg]),
-                            (np.arange(cur.shape[0]), agg)),
-                           shape=(cur.shape[0], nc))
-        Td = csr_array(T)
+        # tentative prolongator: T[i, agg[i]] = 1, column-normalized —
+        # exactly one nnz per row, built directly as device slabs (no
+        # scipy round-trip)
+        import torch as _t
+
+        dev = cur._values.device
+        agg_t = _t.as_tensor(agg, device=dev)
+        counts_t = _t.bincount(agg_t, minlength=nc).to(_t.float64)
+        vals_all = 1.0 / _t.sqrt(counts_t[agg_t])
+        me_r = comm.rank()
+        s0, s1 = cur.partition.start(me_r), cur.partition.stop(me_r)
+        idt = _t.int32 if nc < 2**31 - 1 else _t.int64
+        Td = csr_array.from_local(
+            _t.arange(s1 - s0 + 1, dtype=_t.int64, device=dev),
+            agg_t[s0:s1].to(idt), vals_all[s0:s1].to(cur._values.dtype),
+            cur.partition, (cur.shape[0], nc))
         # ---- smoothed prolongator: P = (I - omega D^-1 A) T (distributed) --
         dinv_vec = 1.0 / cur.diagonal().gather()
         # rho(D^-1 A) power iteration
