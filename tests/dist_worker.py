@@ -225,6 +225,9 @@ def extra_banded_overlap_check():
     assert np.isclose(float(dotv),
                       float(np.asarray(x) @ (sref @ np.asarray(x))),
                       rtol=1e-8), "banded dist spmv_dot"
+    rr = A.residual(x, _d.asdistarray(np.ones(n)))
+    assert np.allclose(np.asarray(rr), 1.0 - sref @ np.asarray(x),
+                       rtol=1e-10), "banded dist residual"
     bb = _d.random((n,), seed=56)
     dinv = _d.asdistarray(1.0 / sref.diagonal())
     outj = A.jacobi_smooth(x, bb, dinv, 0.7)
