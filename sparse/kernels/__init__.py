@@ -282,7 +282,15 @@ def build_dia(A, row0: int):
     rows = torch.repeat_interleave(
         torch.arange(row0, row0 + m, dtype=torch.int64, device=A.device), counts)
     diag = A.indices.long() - rows
-    offs = torch.unique(diag)
+    # chunked: torch.unique / index_put hit CUB 2^31-element limits at
+    # capacity scale (3B nnz on one 288 GB GPU)
+    CH = 1 << 30
+    if diag.numel() <= CH:
+        offs = torch.unique(diag)
+    else:
+        parts = [torch.unique(diag[i: i + CH])
+                 for i in range(0, diag.numel(), CH)]
+        offs = torch.unique(torch.cat(parts))
     W = int(offs.numel())
     mp = (m + 1) // 2 * 2
     if W == 0 or W > 48 or W * mp > 1.6 * A.nnz + 4096:
@@ -292,8 +300,10 @@ def build_dia(A, row0: int):
     if need > 0.5 * free:
         return None
     dvals = torch.zeros(W * mp, dtype=A.values.dtype, device=A.device)
-    k_idx = torch.searchsorted(offs, diag)
-    dvals[k_idx * mp + (rows - row0)] = A.values
+    for i in range(0, diag.numel(), CH):
+        dch = diag[i: i + CH]
+        k_idx = torch.searchsorted(offs, dch)
+        dvals[k_idx * mp + (rows[i: i + CH] - row0)] = A.values[i: i + CH]
     return DiaMirror(dvals, offs, W, m, row0,
                      off_min=int(offs[0].item()), off_max=int(offs[-1].item()))
 
