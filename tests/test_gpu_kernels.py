@@ -587,3 +587,49 @@ def test_dia_residual_gpu():
     r2 = A2.residual(x2, b2)
     assert np.allclose(np.asarray(r2), np.asarray(b2) - s @ np.asarray(x2),
                        rtol=1e-10)
+
+
+def test_solvers_gpu():
+    """Non-CG solvers end-to-end on the GPU (gmres/bicgstab/lsqr/eigsh run
+    through DistArray torch ops + HIP SpMV)."""
+    from sparse import csr_array, linalg
+
+    n = 60
+    rng = np.random.default_rng(101)
+    s = (sps.random(n, n, 0.25, random_state=102) + n * sps.eye(n)).tocsr()
+    b = rng.random(n)
+    x, _ = linalg.gmres(csr_array(s), b, tol=1e-10, restart=20, maxiter=300)
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-5)
+    x, _ = linalg.bicgstab(csr_array(s), b, tol=1e-10, maxiter=400,
+                           conv_test_iters=2)
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-5)
+    spd = spd_csr(n, seed=103)
+    w, V = linalg.eigsh(csr_array(spd), k=3, which="LA")
+    ws_ = np.linalg.eigvalsh(spd.toarray())
+    assert np.allclose(np.sort(w), ws_[-3:], atol=1e-4)
+    m = 50
+    srect = sps.random(m, 20, 0.4, random_state=104).tocsr()
+    brect = rng.random(m)
+    r = linalg.lsqr(csr_array(srect), brect, atol=1e-12, btol=1e-12,
+                    iter_lim=200)
+    xref = sps.linalg.lsqr(srect, brect, atol=1e-12, btol=1e-12,
+                           iter_lim=200)[0]
+    assert np.allclose(np.asarray(r[0]), xref, atol=1e-5)
+
+
+def test_integrate_banded_gpu():
+    """solve_ivp with a sparse banded RHS operator on GPU (SpMV inside the
+    RK stages)."""
+    from sparse import asdistarray, gallery, integrate
+
+    n = 2000
+    A = gallery.banded(n, ndiags=5)
+    A = A * (-0.01)
+    y0 = np.linspace(1.0, 2.0, n)
+
+    def f(t, y):
+        return A.dot(asdistarray(y))
+
+    res = integrate.solve_ivp(f, (0, 1.0), y0, method="RK45", rtol=1e-8,
+                              atol=1e-10)
+    assert res.success
