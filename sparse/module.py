@@ -15,6 +15,8 @@ from __future__ import annotations
 import numpy as np
 import torch
 
+import scipy.sparse as _sp
+
 from .coo import coo_array
 from .csc import csc_array
 from .csr import csr_array
@@ -29,6 +31,7 @@ __all__ = [
     "spdiags", "diags", "eye", "identity", "kron", "random", "rand",
     "is_sparse_matrix", "issparse", "isspmatrix", "isspmatrix_csr",
     "isspmatrix_csc", "isspmatrix_coo", "isspmatrix_dia",
+    "hstack", "vstack", "bmat", "block_diag", "tril", "triu",
 ]
 
 
@@ -252,3 +255,57 @@ def isspmatrix_coo(o) -> bool:
 
 def isspmatrix_dia(o) -> bool:
     return isinstance(o, dia_array)
+
+
+# -- block composition / triangle extraction ---------------------------------
+# (superset of the reference: its clone_module only wraps its own functions,
+# so scipy.sparse's hstack/vstack/bmat/tril/triu are absent there.  Here
+# they assemble via scipy on the host — fine for construction-time use —
+# and return distributed csr_arrays.)
+def _to_scipy(x):
+    if is_sparse_matrix(x):
+        return x.tocsr().to_scipy_sparse_csr()
+    return _sp.csr_matrix(np.asarray(x))
+
+
+def hstack(blocks, format="csr", dtype=None):
+    """Stack sparse matrices horizontally (scipy.sparse.hstack parity)."""
+    out = _sp.hstack([_to_scipy(b) for b in blocks], format="csr",
+                     dtype=dtype)
+    return csr_array(out).asformat(format)
+
+
+def vstack(blocks, format="csr", dtype=None):
+    """Stack sparse matrices vertically (scipy.sparse.vstack parity)."""
+    out = _sp.vstack([_to_scipy(b) for b in blocks], format="csr",
+                     dtype=dtype)
+    return csr_array(out).asformat(format)
+
+
+def bmat(blocks, format="csr", dtype=None):
+    """Assemble from a 2-D grid of blocks (None = zero block)."""
+    grid = [[None if b is None else _to_scipy(b) for b in row]
+            for row in blocks]
+    out = _sp.bmat(grid, format="csr", dtype=dtype)
+    return csr_array(out).asformat(format)
+
+
+def block_diag(mats, format="csr", dtype=None):
+    """Block-diagonal assembly (scipy.sparse.block_diag parity)."""
+    out = _sp.block_diag([_to_scipy(b) for b in mats], format="csr",
+                         dtype=dtype)
+    return csr_array(out).asformat(format)
+
+
+def tril(A, k=0, format=None):
+    """Lower triangle (scipy.sparse.tril parity)."""
+    out = _sp.tril(_to_scipy(A), k=k, format="csr")
+    r = csr_array(out)
+    return r.asformat(format) if format else r
+
+
+def triu(A, k=0, format=None):
+    """Upper triangle (scipy.sparse.triu parity)."""
+    out = _sp.triu(_to_scipy(A), k=k, format="csr")
+    r = csr_array(out)
+    return r.asformat(format) if format else r

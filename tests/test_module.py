@@ -74,3 +74,34 @@ def test_issparse_predicates():
     assert sparse.isspmatrix_csc(A.tocsc())
     assert sparse.isspmatrix_dia(A.tocoo().todia())
     assert not sparse.issparse(np.eye(3))
+
+
+def test_stack_and_triangles():
+    """hstack/vstack/bmat/block_diag/tril/triu (superset: the reference's
+    namespace clone does not provide these)."""
+    import scipy.sparse as sps
+
+    import sparse
+
+    a = sps.random(6, 8, 0.4, random_state=1, format="csr")
+    b = sps.random(6, 5, 0.4, random_state=2, format="csr")
+    c = sps.random(4, 8, 0.4, random_state=3, format="csr")
+    H = sparse.hstack([sparse.csr_array(a), sparse.csr_array(b)])
+    assert np.allclose(np.asarray(H.todense()),
+                       sps.hstack([a, b]).toarray())
+    V = sparse.vstack([sparse.csr_array(a), sparse.csr_array(c)])
+    assert np.allclose(np.asarray(V.todense()),
+                       sps.vstack([a, c]).toarray())
+    B = sparse.bmat([[sparse.csr_array(a), None],
+                     [None, sparse.csr_array(b)]])
+    assert np.allclose(np.asarray(B.todense()),
+                       sps.bmat([[a, None], [None, b]]).toarray())
+    D = sparse.block_diag([sparse.csr_array(a), sparse.csr_array(b)])
+    assert np.allclose(np.asarray(D.todense()),
+                       sps.block_diag([a, b]).toarray())
+    sq = sps.random(7, 7, 0.5, random_state=4, format="csr")
+    for k in (-2, 0, 1):
+        assert np.allclose(np.asarray(sparse.tril(sparse.csr_array(sq), k=k).todense()),
+                           sps.tril(sq, k=k).toarray())
+        assert np.allclose(np.asarray(sparse.triu(sparse.csr_array(sq), k=k).todense()),
+                           sps.triu(sq, k=k).toarray())
