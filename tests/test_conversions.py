@@ -165,3 +165,20 @@ def test_make_with_same_nnz_structure():
     ref.data = newvals
     assert np.allclose(np.asarray(B.todense()), ref.toarray())
     assert B._indices is A._indices  # structure shared
+
+
+def test_csc_sddmm():
+    """CSC SDDMM (reference test_csr_sddmm.py's CSC half, CSC_SDDMM task)."""
+    import numpy as np
+    import scipy.sparse as sps
+
+    import sparse
+
+    s = sps.random(40, 30, 0.2, random_state=5, format="csc")
+    A = sparse.csc_array(s)
+    rng = np.random.default_rng(6)
+    C = rng.random((40, 4))
+    D = rng.random((4, 30))
+    out = A.sddmm(C, D)
+    assert np.allclose(np.asarray(out.todense()),
+                       s.multiply(C @ D).toarray())
