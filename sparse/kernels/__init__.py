@@ -58,7 +58,8 @@ def spmv_dot(A, x, y, p, dot_out, col_lo: int):
 def axpby_norm2(y, x, a, b, isalpha, negate):
     """Fused axpby; returns sum(y_new^2) (real dtypes) as a 0-dim tensor."""
     n = y.numel()
-    blocks = (n // 2 + 255) // 256 + 1
+    ept = max(1, 16 // y.element_size())
+    blocks = (n // ept + 255) // 256 + 1
     partial = torch.empty(blocks, dtype=y.dtype, device=y.device)
     ext().axpby_norm2(y, x, a, b, bool(isalpha), bool(negate), partial)
     return partial.sum()
@@ -356,7 +357,8 @@ def cg_xr_norm2(x, p, r, q, a, b):
     """Fused CG K2: x += (a/b)p; r -= (a/b)q; returns local sum(r_new^2)
     as a device 0-dim tensor."""
     n = x.numel()
-    blocks = (n // 2 + 255) // 256 + 1
+    ept = max(1, 16 // x.element_size())
+    blocks = (n // ept + 255) // 256 + 1
     partial = torch.empty(blocks, dtype=x.dtype, device=x.device)
     ext().cg_xr_norm2(x, p, r, q, a.to(x.dtype), b.to(x.dtype), partial)
     return partial.sum()

@@ -113,40 +113,47 @@ __global__ void mult_dense_kernel(const int64_t* __restrict__ indptr,
 }
 
 // fused y = y ± (a/b) x  (isalpha) | y = x ± (a/b) y ; a,b 0-dim device.
-// 2-element vector loads/stores (16B for fp64): 8B accesses run at ~0.6x
-// the 16B rate on gfx950.
+// 16-byte vector loads/stores per thread (EPT = 16/sizeof(T) elements:
+// 4 for fp32, 2 for fp64; 8B accesses run at ~0.6x the 16B rate on
+// gfx950).  Flat (no grid-stride loop): measured 5865 vs 5070 GB/s for
+// the strided form at 268M fp64 (tools/axpby_bench.hip).
 template <typename T>
-struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) VPair {
-  T a, b;
+struct EptOf {
+  static constexpr int value =
+      (16 / sizeof(T)) >= 1 ? (int)(16 / sizeof(T)) : 1;
+};
+
+template <typename T>
+struct alignas(EptOf<T>::value * sizeof(T) <= 16
+                   ? EptOf<T>::value * sizeof(T)
+                   : 16) VVec {
+  T v[EptOf<T>::value];
 };
 
 template <typename T, bool ISALPHA, bool NEG>
 __global__ __launch_bounds__(256) void axpby_kernel(
     T* __restrict__ y, const T* __restrict__ x, const T* __restrict__ a,
     const T* __restrict__ b, int64_t n) {
-  // flat pair-per-thread (no grid-stride loop): measured 5865 vs 5070 GB/s
-  // for the strided form at 268M fp64 (tools/axpby_bench.hip)
+  constexpr int EPT = EptOf<T>::value;
   T s = (*a) / (*b);
   if (NEG) s = -s;
-  const int64_t half = n / 2;
+  const int64_t nv = n / EPT;
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < half) {
-    auto* y2 = reinterpret_cast<VPair<T>*>(y);
-    auto* x2 = reinterpret_cast<const VPair<T>*>(x);
-    VPair<T> yv = y2[i];
-    const VPair<T> xv = x2[i];
-    if (ISALPHA) {
-      yv.a = yv.a + s * xv.a;
-      yv.b = yv.b + s * xv.b;
-    } else {
-      yv.a = xv.a + s * yv.a;
-      yv.b = xv.b + s * yv.b;
+  if (i < nv) {
+    auto* y2 = reinterpret_cast<VVec<T>*>(y);
+    auto* x2 = reinterpret_cast<const VVec<T>*>(x);
+    VVec<T> yv = y2[i];
+    const VVec<T> xv = x2[i];
+#pragma unroll
+    for (int j = 0; j < EPT; ++j) {
+      yv.v[j] = ISALPHA ? (yv.v[j] + s * xv.v[j]) : (xv.v[j] + s * yv.v[j]);
     }
     y2[i] = yv;
   }
-  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
-    int64_t t = n - 1;
-    y[t] = ISALPHA ? (y[t] + s * x[t]) : (x[t] + s * y[t]);
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    for (int64_t t = nv * EPT; t < n; ++t) {
+      y[t] = ISALPHA ? (y[t] + s * x[t]) : (x[t] + s * y[t]);
+    }
   }
 }
 
@@ -234,7 +241,8 @@ void axpby_hip(at::Tensor y, at::Tensor x, at::Tensor a, at::Tensor b,
                bool isalpha, bool negate) {
   int64_t n = y.numel();
   if (n == 0) return;
-  int64_t blocks = (n / 2 + 255) / 256 + 1;
+  const int64_t ept = std::max<int64_t>(1, 16 / y.element_size());
+  int64_t blocks = (n / ept + 255) / 256 + 1;
   DISPATCH_VALUES(y.scalar_type(), "axpby", [&] {
     using T = scalar_t;
     auto launch = [&](auto kern) {
@@ -256,33 +264,32 @@ template <typename T, bool ISALPHA, bool NEG>
 __global__ __launch_bounds__(256) void axpby_norm2_kernel(
     T* __restrict__ y, const T* __restrict__ x, const T* __restrict__ a,
     const T* __restrict__ b, T* __restrict__ dot_partial, int64_t n) {
-  // flat pair-per-thread + per-block partial (wrapper sums; no atomics)
+  // flat 16B-per-thread + per-block partial (wrapper sums; no atomics)
+  constexpr int EPT = EptOf<T>::value;
   __shared__ T red[256];
   T s = (*a) / (*b);
   if (NEG) s = -s;
   T acc = T(0);
-  const int64_t half = n / 2;
+  const int64_t nv = n / EPT;
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < half) {
-    auto* y2 = reinterpret_cast<VPair<T>*>(y);
-    auto* x2 = reinterpret_cast<const VPair<T>*>(x);
-    VPair<T> yv = y2[i];
-    const VPair<T> xv = x2[i];
-    if (ISALPHA) {
-      yv.a = yv.a + s * xv.a;
-      yv.b = yv.b + s * xv.b;
-    } else {
-      yv.a = xv.a + s * yv.a;
-      yv.b = xv.b + s * yv.b;
+  if (i < nv) {
+    auto* y2 = reinterpret_cast<VVec<T>*>(y);
+    auto* x2 = reinterpret_cast<const VVec<T>*>(x);
+    VVec<T> yv = y2[i];
+    const VVec<T> xv = x2[i];
+#pragma unroll
+    for (int j = 0; j < EPT; ++j) {
+      yv.v[j] = ISALPHA ? (yv.v[j] + s * xv.v[j]) : (xv.v[j] + s * yv.v[j]);
+      acc += yv.v[j] * yv.v[j];
     }
     y2[i] = yv;
-    acc += yv.a * yv.a + yv.b * yv.b;
   }
-  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
-    int64_t t = n - 1;
-    T v = ISALPHA ? (y[t] + s * x[t]) : (x[t] + s * y[t]);
-    y[t] = v;
-    acc += v * v;
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    for (int64_t t = nv * EPT; t < n; ++t) {
+      T v = ISALPHA ? (y[t] + s * x[t]) : (x[t] + s * y[t]);
+      y[t] = v;
+      acc += v * v;
+    }
   }
   red[threadIdx.x] = acc;
   __syncthreads();
@@ -302,34 +309,37 @@ __global__ __launch_bounds__(256) void cg_xr_norm2_kernel(
     T* __restrict__ x, const T* __restrict__ p, T* __restrict__ r,
     const T* __restrict__ q, const T* __restrict__ a, const T* __restrict__ b,
     T* __restrict__ dot_partial, int64_t n) {
+  constexpr int EPT = EptOf<T>::value;
   __shared__ T red[256];
   const T s = (*a) / (*b);
   T acc = T(0);
-  const int64_t half = n / 2;
+  const int64_t nv = n / EPT;
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < half) {
-    auto* x2 = reinterpret_cast<VPair<T>*>(x);
-    auto* r2 = reinterpret_cast<VPair<T>*>(r);
-    auto* p2 = reinterpret_cast<const VPair<T>*>(p);
-    auto* q2 = reinterpret_cast<const VPair<T>*>(q);
-    VPair<T> xv = x2[i];
-    VPair<T> rv = r2[i];
-    const VPair<T> pv = p2[i];
-    const VPair<T> qv = q2[i];
-    xv.a += s * pv.a;
-    xv.b += s * pv.b;
-    rv.a -= s * qv.a;
-    rv.b -= s * qv.b;
+  if (i < nv) {
+    auto* x2 = reinterpret_cast<VVec<T>*>(x);
+    auto* r2 = reinterpret_cast<VVec<T>*>(r);
+    auto* p2 = reinterpret_cast<const VVec<T>*>(p);
+    auto* q2 = reinterpret_cast<const VVec<T>*>(q);
+    VVec<T> xv = x2[i];
+    VVec<T> rv = r2[i];
+    const VVec<T> pv = p2[i];
+    const VVec<T> qv = q2[i];
+#pragma unroll
+    for (int j = 0; j < EPT; ++j) {
+      xv.v[j] += s * pv.v[j];
+      rv.v[j] -= s * qv.v[j];
+      acc += rv.v[j] * rv.v[j];
+    }
     x2[i] = xv;
     r2[i] = rv;
-    acc = rv.a * rv.a + rv.b * rv.b;
   }
-  if (blockIdx.x == 0 && threadIdx.x == 0 && (n & 1)) {
-    const int64_t t = n - 1;
-    x[t] += s * p[t];
-    const T rv = r[t] - s * q[t];
-    r[t] = rv;
-    acc += rv * rv;
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    for (int64_t t = nv * EPT; t < n; ++t) {
+      x[t] += s * p[t];
+      const T rv = r[t] - s * q[t];
+      r[t] = rv;
+      acc += rv * rv;
+    }
   }
   red[threadIdx.x] = acc;
   __syncthreads();
@@ -346,7 +356,8 @@ void cg_xr_norm2_hip(at::Tensor x, at::Tensor p, at::Tensor r, at::Tensor q,
                      at::Tensor a, at::Tensor b, at::Tensor dot_out) {
   int64_t n = x.numel();
   if (n == 0) return;
-  int64_t blocks = (n / 2 + 255) / 256 + 1;
+  const int64_t ept = std::max<int64_t>(1, 16 / x.element_size());
+  int64_t blocks = (n / ept + 255) / 256 + 1;
   TORCH_CHECK(dot_out.numel() >= blocks,
               "cg_xr_norm2: dot_out too small (", dot_out.numel(), " < ",
               blocks, ")");
@@ -363,10 +374,12 @@ void axpby_norm2_hip(at::Tensor y, at::Tensor x, at::Tensor a, at::Tensor b,
                      bool isalpha, bool negate, at::Tensor dot_out) {
   int64_t n = y.numel();
   if (n == 0) return;
-  // flat pair-per-thread, same grid as axpby_hip: ceil((n/2)/256) + 1 (the
-  // +1 block only contributes a zero partial).  NO cap: a capped grid left
-  // elements beyond blocks*512 untouched (silent corruption at n > 512K).
-  int64_t blocks = (n / 2 + 255) / 256 + 1;
+  // flat 16B-per-thread, same grid as axpby_hip: ceil((n/EPT)/256) + 1
+  // (the +1 block only contributes a zero partial).  NO cap: a capped grid
+  // left elements beyond its reach untouched (silent corruption at large
+  // n — regression-tested).
+  const int64_t ept = std::max<int64_t>(1, 16 / y.element_size());
+  int64_t blocks = (n / ept + 255) / 256 + 1;
   TORCH_CHECK(dot_out.numel() >= blocks,
               "axpby_norm2: dot_out too small (", dot_out.numel(), " < ",
               blocks, ")");
