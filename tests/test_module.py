@@ -105,3 +105,20 @@ def test_stack_and_triangles():
                            sps.tril(sq, k=k).toarray())
         assert np.allclose(np.asarray(sparse.triu(sparse.csr_array(sq), k=k).todense()),
                            sps.triu(sq, k=k).toarray())
+
+
+def test_bounds_checks_env(monkeypatch):
+    """SPARSE_BOUNDS_CHECKS=1 validates structure at construction
+    (reference Legion_BOUNDS_CHECKS parity)."""
+    import scipy.sparse as sps
+
+    import sparse
+
+    monkeypatch.setenv("SPARSE_BOUNDS_CHECKS", "1")
+    s = sps.random(10, 10, 0.3, random_state=1, format="csr")
+    A = sparse.csr_array(s)  # valid structure passes
+    assert A.nnz == s.nnz
+    bad_indices = np.array([0, 99], dtype=np.int64)  # col 99 out of range
+    bad_indptr = np.array([0, 1, 2, 2, 2, 2, 2, 2, 2, 2, 2], dtype=np.int64)
+    with pytest.raises((ValueError, AssertionError, RuntimeError)):
+        sparse.csr_array((np.ones(2), bad_indices, bad_indptr), shape=(10, 10))
