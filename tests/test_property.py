@@ -1,0 +1,71 @@
+"""Property-based fuzzing vs the scipy oracle (hypothesis).
+
+Random shapes/densities/dtypes through the core op surface — catches
+edge cases (empty rows, duplicate-free randoms, single-column windows)
+that the fixed-seed tests miss.
+"""
+import numpy as np
+import scipy.sparse as sps
+from hypothesis import given, settings, strategies as st
+
+from sparse import csr_array, csc_array
+
+DT = st.sampled_from([np.float64, np.float32])
+
+
+def _rand(m, n, density, seed, dt):
+    s = sps.random(m, n, density, random_state=seed, format="csr",
+                   dtype=np.float64)
+    s.sort_indices()
+    return s.astype(dt)
+
+
+@settings(max_examples=25, deadline=None)
+@given(m=st.integers(1, 40), n=st.integers(1, 40),
+       density=st.floats(0.0, 0.5), seed=st.integers(0, 10**6), dt=DT)
+def test_spmv_matches_scipy(m, n, density, seed, dt):
+    s = _rand(m, n, density, seed, dt)
+    x = np.random.default_rng(seed).random(n).astype(dt)
+    got = np.asarray(csr_array(s) @ x)
+    rtol = 1e-10 if dt == np.float64 else 1e-4
+    assert np.allclose(got, s @ x, rtol=rtol, atol=1e-6)
+
+
+@settings(max_examples=20, deadline=None)
+@given(m=st.integers(1, 30), n=st.integers(1, 30), k=st.integers(1, 30),
+       d1=st.floats(0.05, 0.5), d2=st.floats(0.05, 0.5),
+       seed=st.integers(0, 10**6))
+def test_spgemm_matches_scipy(m, n, k, d1, d2, seed):
+    a = _rand(m, n, d1, seed, np.float64)
+    b = _rand(n, k, d2, seed + 1, np.float64)
+    C = csr_array(a) @ csr_array(b)
+    ref = (a @ b).tocsr()
+    assert C.shape == ref.shape
+    assert np.allclose(np.asarray(C.todense()), ref.toarray(),
+                       rtol=1e-10, atol=1e-12)
+
+
+@settings(max_examples=20, deadline=None)
+@given(m=st.integers(1, 30), n=st.integers(1, 30),
+       d1=st.floats(0.0, 0.5), d2=st.floats(0.0, 0.5),
+       seed=st.integers(0, 10**6))
+def test_add_mult_match_scipy(m, n, d1, d2, seed):
+    a = _rand(m, n, d1, seed, np.float64)
+    b = _rand(m, n, d2, seed + 7, np.float64)
+    S = csr_array(a) + csr_array(b)
+    assert np.allclose(np.asarray(S.todense()), (a + b).toarray())
+    M = csr_array(a).multiply(csr_array(b))
+    assert np.allclose(np.asarray(M.todense()), a.multiply(b).toarray())
+
+
+@settings(max_examples=15, deadline=None)
+@given(m=st.integers(1, 30), n=st.integers(1, 30),
+       density=st.floats(0.0, 0.5), seed=st.integers(0, 10**6))
+def test_conversion_roundtrips(m, n, density, seed):
+    s = _rand(m, n, density, seed, np.float64)
+    A = csr_array(s)
+    assert np.allclose(np.asarray(A.tocsc().tocsr().todense()), s.toarray())
+    assert np.allclose(np.asarray(A.tocoo().tocsr().todense()), s.toarray())
+    assert np.allclose(np.asarray(A.T.todense()), s.T.toarray())
+    d = np.asarray(A.diagonal())
+    assert np.allclose(d, s.diagonal())
