@@ -50,13 +50,20 @@ __global__ void gen_dia(double* dvals, double* x, int64_t nx, int64_t N,
   if (r < N) x[r] = 1.0 + (double)(r % 97) * 0.01;
 }
 
+__device__ __forceinline__ int64_t swz(int64_t bid, int64_t nwg) {
+  int64_t q = nwg / 8, rr = nwg % 8;
+  int64_t xcd = bid % 8, idx = bid / 8;
+  return (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+}
+
 template <int NT>
 __global__ __launch_bounds__(BLK) void dia_v(const double* __restrict__ dvals,
                                              const int64_t* __restrict__ offs,
                                              const double* __restrict__ x,
                                              double* __restrict__ y,
                                              int64_t m, int64_t mp, int W) {
-  const int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
+  const int64_t b = NT >= 3 ? swz(blockIdx.x, gridDim.x) : blockIdx.x;
+  const int64_t t = b * BLK + threadIdx.x;
   const int64_t r0 = 2 * t;
   if (r0 >= mp) return;
   double a0 = 0.0, a1 = 0.0;
@@ -76,7 +83,7 @@ __global__ __launch_bounds__(BLK) void dia_v(const double* __restrict__ dvals,
     a1 += vv.b * x[i1];
   }
   if (r0 + 1 < m) {
-    if (NT >= 2) {
+    if (NT == 2) {
       __builtin_nontemporal_store(a0, &y[r0]);
       __builtin_nontemporal_store(a1, &y[r0 + 1]);
     } else {
@@ -107,18 +114,20 @@ int main(int argc, char** argv) {
   const int64_t nblocks = (mp / 2 + BLK - 1) / BLK;
   const int iters = 30;
   double bytes = (double)W * mp * 8 + 2.0 * N * 8;
-  const char* names[3] = {"v0_plain", "v1_nt_vals", "v2_nt_vals_y"};
-  std::vector<std::vector<float>> ms(3);
+  const char* names[4] = {"v0_plain", "v1_nt_vals", "v2_nt_vals_y",
+                          "v3_nt_swz"};
+  std::vector<std::vector<float>> ms(4);
   hipEvent_t e0, e1;
   CHECK(hipEventCreate(&e0));
   CHECK(hipEventCreate(&e1));
   for (int rep = 0; rep < iters; ++rep) {
-    for (int v = 0; v < 3; ++v) {
+    for (int v = 0; v < 4; ++v) {
       CHECK(hipEventRecord(e0));
       switch (v) {
         case 0: dia_v<0><<<nblocks, BLK>>>(dvals, offs, x, y, N, mp, W); break;
         case 1: dia_v<1><<<nblocks, BLK>>>(dvals, offs, x, y, N, mp, W); break;
         case 2: dia_v<2><<<nblocks, BLK>>>(dvals, offs, x, y, N, mp, W); break;
+        case 3: dia_v<3><<<nblocks, BLK>>>(dvals, offs, x, y, N, mp, W); break;
       }
       CHECK(hipEventRecord(e1));
       CHECK(hipEventSynchronize(e1));
@@ -128,7 +137,7 @@ int main(int argc, char** argv) {
     }
   }
   // checksum parity
-  for (int v = 0; v < 3; ++v) {
+  for (int v = 0; v < 4; ++v) {
     std::sort(ms[v].begin(), ms[v].end());
     float med = ms[v][ms[v].size() / 2];
     printf("%-14s median %.3f ms  %.0f GB/s\n", names[v], med,
