@@ -125,6 +125,12 @@ _UNARY_TORCH = {
     "rad2deg": torch.rad2deg,
 }
 
+def _abs_dunder(self):
+    return self._with_values(torch.abs)
+
+
+CompressedBase.__abs__ = _abs_dunder
+
 for _name, _fn in _UNARY_TORCH.items():
 
     def _method(self, _fn=_fn):

@@ -122,3 +122,10 @@ def test_bounds_checks_env(monkeypatch):
     bad_indptr = np.array([0, 1, 2, 2, 2, 2, 2, 2, 2, 2, 2], dtype=np.int64)
     with pytest.raises((ValueError, AssertionError, RuntimeError)):
         sparse.csr_array((np.ones(2), bad_indices, bad_indptr), shape=(10, 10))
+
+
+def test_abs_dunder():
+    s = sps.random(15, 15, 0.3, random_state=8, format="csr") - 0.5 * sps.eye(15)
+    s = s.tocsr()
+    A = csr_array(s)
+    assert np.allclose(np.asarray(abs(A).todense()), abs(s).toarray())
