@@ -24,7 +24,10 @@ import torch
 from benchmark import parse_common_args
 
 parser = argparse.ArgumentParser()
-parser.add_argument("-N", type=int, default=255, help="grid edge (odd, 2^k-1)")
+parser.add_argument("-N", "-nx", "--num", type=int, default=255,
+                    dest="N", help="grid edge")
+parser.add_argument("-ny", type=int, default=None,
+                    help="accepted for CLI parity; grids are square")
 parser.add_argument("-dim", type=int, default=2, choices=[2, 3])
 parser.add_argument("-levels", type=int, default=None)
 parser.add_argument("-maxiter", type=int, default=200)
