@@ -28,7 +28,10 @@ def test_pde_throughput():
 
 def test_dot_microbenchmark():
     out = run("dot_microbenchmark.py", "-n", "100000", "-iters", "5", "-warmup", "1")
-    assert "SpMVs" in out
+    assert "spmvs" in out
+    out = run("dot_microbenchmark.py", "-n", "50000", "-i", "3", "-op",
+              "spmm", "-k", "8")
+    assert "spmms" in out
 
 
 def test_gmg():
