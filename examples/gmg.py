@@ -34,6 +34,19 @@ parser.add_argument("-maxiter", type=int, default=200)
 parser.add_argument("-tol", type=float, default=1e-8)
 parser.add_argument("-throughput", action="store_true")
 parser.add_argument("-smooth_iters", type=int, default=2)
+parser.add_argument("-data", "-d", choices=["poisson", "diffusion"],
+                    default="poisson", dest="data")
+parser.add_argument("-gridop", "-g", choices=["linear", "injection"],
+                    default="linear", dest="gridop",
+                    help="intergrid transfer (reference gmg.py parity)")
+parser.add_argument("-smoother", "-s", choices=["jacobi", "rbgs"],
+                    default="jacobi", dest="smoother",
+                    help="jacobi = fused weighted Jacobi; rbgs = red-black "
+                         "Gauss-Seidel (a WORKING substitute for the "
+                         "reference's symgs option, whose symgs_c symbol "
+                         "is undefined there)")
+parser.add_argument("-epsilon", type=float, default=0.1)
+parser.add_argument("-theta", type=float, default=0.785398)
 args, _ = parser.parse_known_args()
 _, timer, npx, sparse, linalg, use_sparse = parse_common_args()
 
@@ -66,14 +79,27 @@ class GMG:
     """V-cycle preconditioner (2-D bilinear or 3-D trilinear transfers)."""
 
     def __init__(self, A, nx, levels=None, smooth_iters=2, coarse_threshold=1024,
-                 dim=2):
+                 dim=2, gridop="linear", smoother="jacobi"):
         self.levels = []
+        self.smoother = smoother
         cur_nx = nx
         cur = A
         maxl = levels or 64
-        interp = gallery.interpolation2d if dim == 2 else gallery.interpolation3d
+        if gridop == "injection":
+            assert dim == 2, "injection gridop: 2-D grids"
+            interp = gallery.injection2d
+        else:
+            interp = gallery.interpolation2d if dim == 2 else gallery.interpolation3d
         while True:
             self.levels.append(Level(cur))
+            if smoother == "rbgs" and dim == 2:
+                # red-black masks from grid parity (per level)
+                lvl = self.levels[-1]
+                part = cur.partition
+                me = comm.rank()
+                rr = torch.arange(part.start(me), part.stop(me),
+                                  device=cur._values.device)
+                lvl.red = ((rr % cur_nx) + (rr // cur_nx)) % 2 == 0
             if len(self.levels) >= maxl or cur.shape[0] <= coarse_threshold or cur_nx < 7:
                 break
             P = interp(cur_nx)
@@ -100,7 +126,17 @@ class GMG:
         self._graph_tried = False
 
     def _smooth(self, lvl, x, b, iters):
-        # fused weighted-Jacobi sweeps (single ELL kernel each)
+        if self.smoother == "rbgs" and hasattr(lvl, "red"):
+            # red-black Gauss-Seidel: two masked GS half-sweeps (exact GS
+            # for 5-pt stencils; each half = one fused-Jacobi kernel +
+            # masked merge)
+            for _ in range(iters):
+                t = lvl.A.jacobi_smooth(x, b, lvl.dinv, 1.0)
+                x.local[lvl.red] = t.local[lvl.red]
+                t = lvl.A.jacobi_smooth(x, b, lvl.dinv, 1.0)
+                x.local[~lvl.red] = t.local[~lvl.red]
+            return x
+        # fused weighted-Jacobi sweeps (single DIA/ELL kernel each)
         for _ in range(iters):
             x = lvl.A.jacobi_smooth(x, b, lvl.dinv, lvl.omega)
         return x
@@ -182,7 +218,10 @@ def main():
     N = args.N
     assert N % 2 == 1, "N must be odd (2^k - 1)"
     h = 1.0 / (N + 1)
-    if args.dim == 2:
+    if args.data == "diffusion":
+        assert args.dim == 2, "-data diffusion is 2-D"
+        A = gallery.diffusion2d(N, epsilon=args.epsilon, theta=args.theta)
+    elif args.dim == 2:
         A = gallery.poisson2d(N, scale=1.0 / (h * h))
     else:
         A = gallery.poisson3d(N, scale=1.0 / (h * h))
@@ -196,7 +235,7 @@ def main():
     t0 = timer
     t0.start()
     mg = GMG(A, N, levels=args.levels, smooth_iters=args.smooth_iters,
-             dim=args.dim)
+             dim=args.dim, gridop=args.gridop, smoother=args.smoother)
     setup_ms = t0.stop()
 
     # warm the preconditioner (captures the hipGraph) outside the timer

@@ -77,6 +77,63 @@ def poisson2d(nx: int, ny: Optional[int] = None, dtype=np.float64,
     return _assemble(rows, cols, vals, valid, (N, N), part, dtype)
 
 
+def stencil2d(stencil, nx: int, ny: Optional[int] = None,
+              dtype=np.float64) -> csr_array:
+    """General 3x3-stencil operator on an nx*ny grid (Dirichlet): the
+    reference's stencil_grid (examples/gmg.py:21-80) for 2-D, built as
+    distributed device slabs.  stencil[di+1][dj+1] multiplies the
+    neighbor at (iy+di, ix+dj)."""
+    st = np.asarray(stencil, dtype=np.float64)
+    assert st.shape == (3, 3), "stencil2d wants a 3x3 stencil"
+    ny = nx if ny is None else ny
+    N = nx * ny
+    part = RowPartition.equal(N, comm.world_size())
+    r = comm.rank()
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64,
+                        device=_dev())
+    ix = rows % nx
+    iy = rows // nx
+    one = torch.ones_like(rows, dtype=torch.float64)
+    cols, vals, valid = [], [], []
+    for di in (-1, 0, 1):
+        for dj in (-1, 0, 1):
+            w = float(st[di + 1, dj + 1])
+            if w == 0.0 and not (di == 0 and dj == 0):
+                continue
+            cols.append(rows + di * nx + dj)
+            vals.append(one * w)
+            ok = torch.ones_like(ix, dtype=torch.bool)
+            if dj < 0:
+                ok = ok & (ix > 0)
+            if dj > 0:
+                ok = ok & (ix < nx - 1)
+            if di < 0:
+                ok = ok & (iy > 0)
+            if di > 0:
+                ok = ok & (iy < ny - 1)
+            valid.append(ok)
+    return _assemble(rows, cols, vals, valid, (N, N), part, dtype)
+
+
+def diffusion2d(N: int, epsilon: float = 1.0, theta: float = 0.0,
+                dtype=np.float64) -> csr_array:
+    """Rotated anisotropic diffusion 9-pt stencil (the reference gmg.py
+    'diffusion' problem instance, examples/gmg.py:114-131 there; the
+    standard pyamg test operator)."""
+    import math as _m
+
+    C, S = _m.cos(theta), _m.sin(theta)
+    CS, CC, SS = C * S, C * C, S * S
+    eps = float(epsilon)
+    a = (-1 * eps - 1) * CC + (-1 * eps - 1) * SS + (3 * eps - 3) * CS
+    b = (2 * eps - 4) * CC + (-4 * eps + 2) * SS
+    c = (-1 * eps - 1) * CC + (-1 * eps - 1) * SS + (-3 * eps + 3) * CS
+    d = (-4 * eps + 2) * CC + (2 * eps - 4) * SS
+    e = (8 * eps + 8) * CC + (8 * eps + 8) * SS
+    st = np.array([[a, b, c], [d, e, d], [c, b, a]]) / 6.0
+    return stencil2d(st, N, dtype=dtype)
+
+
 def poisson3d(nx: int, ny: Optional[int] = None, nz: Optional[int] = None,
               dtype=np.float64, scale: float = 1.0) -> csr_array:
     """7-point 3-D Laplacian (Dirichlet) — the GMG 3-D benchmark operator."""
@@ -115,6 +172,25 @@ def banded(n: int, ndiags: int = 11, dtype=np.float64) -> csr_array:
         vals.append(one * (1.0 if k else float(ndiags)))
         valid.append((c >= 0) & (c < n))
     return _assemble(rows, cols, vals, valid, (n, n), part, dtype)
+
+
+def injection2d(nx: int, dtype=np.float64):
+    """Injection prolongation P (N_f x N_c) for GMG on a 2-D grid with
+    n_f = 2*n_c + 1 per axis: row f carries a single 1 iff fine point f
+    coincides with a coarse point (odd, odd).  Reference parity:
+    examples/gmg.py injection_operator (gmg.py:287-301 there)."""
+    nxc = (nx - 1) // 2
+    Nf, Nc = nx * nx, nxc * nxc
+    part = RowPartition.equal(Nf, comm.world_size())
+    r = comm.rank()
+    rows = torch.arange(part.start(r), part.stop(r), dtype=torch.int64,
+                        device=_dev())
+    ixf = rows % nx
+    iyf = rows // nx
+    hit = (ixf % 2 == 1) & (iyf % 2 == 1)
+    cols = (iyf // 2) * nxc + (ixf // 2)
+    vals = torch.ones_like(rows, dtype=torch.float64)
+    return _assemble(rows, [cols], [vals], [hit], (Nf, Nc), part, dtype)
 
 
 def interpolation2d(nx: int, ny: Optional[int] = None, dtype=np.float64):

@@ -124,3 +124,14 @@ def test_pyamg_bridge_conversion():
                         conv_test_iters=5)
     assert info == 0
     assert np.allclose(A @ np.asarray(x), b, atol=1e-5)
+
+
+def test_gmg_variants():
+    """Reference gmg.py option parity: diffusion problem, injection
+    transfer, and a working Gauss-Seidel smoother option."""
+    out = run("gmg.py", "-N", "31", "-data", "diffusion")
+    assert "info=0" in out
+    out = run("gmg.py", "-N", "31", "-gridop", "injection")
+    assert "info=0" in out
+    out = run("gmg.py", "-N", "31", "-smoother", "rbgs")
+    assert "info=0" in out
