@@ -39,10 +39,11 @@ parser.add_argument("-data", "-d", choices=["poisson", "diffusion"],
 parser.add_argument("-gridop", "-g", choices=["linear", "injection"],
                     default="linear", dest="gridop",
                     help="intergrid transfer (reference gmg.py parity)")
-parser.add_argument("-smoother", "-s", choices=["jacobi", "rbgs"],
+parser.add_argument("-smoother", "-s", choices=["jacobi", "rbgs", "symgs"],
                     default="jacobi", dest="smoother",
                     help="jacobi = fused weighted Jacobi; rbgs = red-black "
-                         "Gauss-Seidel (a WORKING substitute for the "
+                         "Gauss-Seidel; symgs = symmetric (RB then BR) "
+                         "red-black GS (a WORKING version of the "
                          "reference's symgs option, whose symgs_c symbol "
                          "is undefined there)")
 parser.add_argument("-epsilon", type=float, default=0.1)
@@ -92,7 +93,7 @@ class GMG:
             interp = gallery.interpolation2d if dim == 2 else gallery.interpolation3d
         while True:
             self.levels.append(Level(cur))
-            if smoother == "rbgs" and dim == 2:
+            if smoother in ("rbgs", "symgs") and dim == 2:
                 # red-black masks from grid parity (per level)
                 lvl = self.levels[-1]
                 part = cur.partition
@@ -126,15 +127,19 @@ class GMG:
         self._graph_tried = False
 
     def _smooth(self, lvl, x, b, iters):
-        if self.smoother == "rbgs" and hasattr(lvl, "red"):
-            # red-black Gauss-Seidel: two masked GS half-sweeps (exact GS
-            # for 5-pt stencils; each half = one fused-Jacobi kernel +
-            # masked merge)
+        if self.smoother in ("rbgs", "symgs") and hasattr(lvl, "red"):
+            # red-black Gauss-Seidel: masked GS half-sweeps (exact GS for
+            # 5-pt stencils; each half = one fused-Jacobi kernel + masked
+            # merge).  symgs adds the reversed (black-red) sweep.
+            def half(mask):
+                t = lvl.A.jacobi_smooth(x, b, lvl.dinv, 1.0)
+                x.local[mask] = t.local[mask]
             for _ in range(iters):
-                t = lvl.A.jacobi_smooth(x, b, lvl.dinv, 1.0)
-                x.local[lvl.red] = t.local[lvl.red]
-                t = lvl.A.jacobi_smooth(x, b, lvl.dinv, 1.0)
-                x.local[~lvl.red] = t.local[~lvl.red]
+                half(lvl.red)
+                half(~lvl.red)
+                if self.smoother == "symgs":
+                    half(~lvl.red)
+                    half(lvl.red)
             return x
         # fused weighted-Jacobi sweeps (single DIA/ELL kernel each)
         for _ in range(iters):
