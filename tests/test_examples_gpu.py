@@ -33,4 +33,13 @@ def test_amg_gpu():
 
 def test_dot_micro_gpu():
     out = run("dot_microbenchmark.py", "-n", "1000000", "-iters", "20", "-warmup", "3")
-    assert "SpMVs" in out
+    assert "spmvs" in out
+
+
+def test_gmg_variants_gpu():
+    out = run("gmg.py", "-N", "127", "-data", "diffusion")
+    assert "info=0" in out
+    out = run("gmg.py", "-N", "127", "-smoother", "symgs")
+    assert "info=0" in out
+    out = run("gmg.py", "-N", "127", "-gridop", "injection")
+    assert "info=0" in out
