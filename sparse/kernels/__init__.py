@@ -203,12 +203,56 @@ def spgemm_csr(A, B, a_col_lo, vdt):
     return LocalCSR(indptr, indices, values, m, B.ncols)
 
 
+_ESC_LIMIT = 1 << 28  # max materialized products per ESC batch
+
+
 def _spgemm_esc(A, B, a_col_lo, vdt):
-    """Expansion-sort-compress SpGEMM fallback in torch ops (still on-GPU)."""
+    """Expansion-sort-compress SpGEMM fallback in torch ops (still on-GPU).
+    Row-batched: the expansion materializes every intermediate product, so
+    unbatched dense-product chains (e.g. repeated A@A squaring) would
+    allocate tens of GB and abort; batches are capped at ~2^28 products."""
     from ..ops.local import LocalCSR
 
     dev = A.device
-    acols = A.indices.long() - a_col_lo
+    LIMIT = _ESC_LIMIT
+    acols_all = A.indices.long() - a_col_lo
+    bc_all = B.indptr[acols_all + 1] - B.indptr[acols_all]
+    total_all = int(bc_all.sum().item())
+    if total_all > LIMIT and A.nrows > 1:
+        acounts = A.indptr[1:] - A.indptr[:-1]
+        arows_all = torch.repeat_interleave(
+            torch.arange(A.nrows, dtype=torch.int64, device=dev), acounts)
+        rowprod = torch.zeros(A.nrows, dtype=torch.int64, device=dev)
+        rowprod.index_add_(0, arows_all, bc_all)
+        cum = torch.cumsum(rowprod, 0).cpu().numpy()
+        pieces = []
+        r0 = 0
+        base = 0
+        import numpy as _np
+
+        while r0 < A.nrows:
+            r1 = int(_np.searchsorted(cum, base + LIMIT, side="right"))
+            r1 = max(r1, r0 + 1)
+            ip = (A.indptr[r0: r1 + 1] - A.indptr[r0]).contiguous()
+            lo, hi = int(A.indptr[r0].item()), int(A.indptr[r1].item())
+            sub = LocalCSR(ip, A.indices[lo:hi], A.values[lo:hi],
+                           r1 - r0, A.ncols)
+            pieces.append(_spgemm_esc(sub, B, a_col_lo, vdt))
+            base = cum[r1 - 1]
+            r0 = r1
+        indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=dev)
+        off = 0
+        chunks_i, chunks_v = [], []
+        r0 = 0
+        for pc in pieces:
+            indptr[r0 + 1: r0 + pc.nrows + 1] = pc.indptr[1:] + off
+            off += int(pc.indptr[-1].item())
+            r0 += pc.nrows
+            chunks_i.append(pc.indices)
+            chunks_v.append(pc.values)
+        return LocalCSR(indptr, torch.cat(chunks_i), torch.cat(chunks_v),
+                        A.nrows, B.ncols)
+    acols = acols_all
     acounts = A.indptr[1:] - A.indptr[:-1]
     arows = torch.repeat_interleave(
         torch.arange(A.nrows, dtype=torch.int64, device=dev), acounts)

@@ -100,3 +100,20 @@ def test_dot_vector_matmul_operator():
     x = sample_dense(6, seed=21)
     # x @ A (vector from the left)
     assert np.allclose(np.asarray(x @ csr_array(s)), x @ s)
+
+
+def test_spgemm_esc_batching(monkeypatch):
+    """Force tiny ESC batches: batched expansion must agree with scipy on
+    a dense-product multiply (rows with >1024-column products)."""
+    import scipy.sparse as sps
+
+    from sparse import csr_array, kernels
+
+    monkeypatch.setattr(kernels, "_ESC_LIMIT", 500)
+    rng = np.random.default_rng(44)
+    a = sps.random(60, 50, 0.4, random_state=45).tocsr()
+    b = sps.random(50, 2000, 0.6, random_state=46).tocsr()
+    C = csr_array(a) @ csr_array(b)
+    ref = (a @ b).tocsr()
+    assert C.nnz == ref.nnz
+    assert np.allclose(np.asarray(C.todense()), ref.toarray(), rtol=1e-10)
