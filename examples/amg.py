@@ -337,12 +337,13 @@ def main():
                         conv_test_iters=5,
                         callback=lambda _x: it_count.__setitem__(0, it_count[0] + 1))
     solve_ms = timer.stop()
+    r = b - A.dot(x)  # collective: all ranks participate
+    rn = float(r.norm().item())  # all-reduce: all ranks
     if comm.rank() == 0:
-        r = b - A.dot(x)
         print(f"levels={len(levels)} opcx={op_complexity:.2f} "
               f"setup={setup_ms:.1f}ms solve={solve_ms:.1f}ms iters={it_count[0]} "
               f"({it_count[0] / max(solve_ms, 1e-9) * 1000.0:.2f} iters/s) "
-              f"residual={float(r.norm().item()):.3e} info={info}")
+              f"residual={rn:.3e} info={info}")
 
 
 if __name__ == "__main__":

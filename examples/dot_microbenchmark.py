@@ -42,9 +42,10 @@ if use_sparse:
     for _ in range(args.iters):
         A.dot(x, out=y)
     ms = timer.stop()
+    nnz = A.nnz  # collective: all ranks
     if comm.rank() == 0:
         kk = args.k if args.op == "spmm" else 1
-        gflops = 2.0 * A.nnz * kk * args.iters / (ms / 1000.0) / 1e9
+        gflops = 2.0 * nnz * kk * args.iters / (ms / 1000.0) / 1e9
         print(f"{args.iters} {args.op}s in {ms:.1f} ms "
               f"({args.iters / (ms / 1000.0):.2f} iters/s, {gflops:.1f} GFLOP/s)")
 else:

@@ -252,12 +252,13 @@ def main():
                         maxiter=args.maxiter, conv_test_iters=5,
                         callback=lambda _x: it_count.__setitem__(0, it_count[0] + 1))
     solve_ms = t0.stop()
+    r = b - A.dot(x)  # collective: all ranks participate
+    rn = float(r.norm().item())  # all-reduce: all ranks
     if comm.rank() == 0:
-        r = b - A.dot(x)
         print(f"levels={len(mg.levels)} setup={setup_ms:.1f}ms "
               f"solve={solve_ms:.1f}ms iters={it_count[0]} "
               f"({it_count[0] / (solve_ms / 1000.0):.2f} iters/s) "
-              f"residual={float(r.norm().item()):.3e} info={info}")
+              f"residual={rn:.3e} info={info}")
 
 
 if __name__ == "__main__":

@@ -58,11 +58,12 @@ if use_sparse:
                              callback=lambda _x: it_count.__setitem__(0, it_count[0] + 1))
         iters = it_count[0]
     ms = timer.stop()
+    r = b - A.dot(xs)  # collective: all ranks participate
+    rn = float(r.norm().item())
     if comm.rank() == 0:
         print(f"Solve finished: {iters} iterations in {ms:.1f} ms "
               f"({iters / (ms / 1000.0):.2f} iters/s)")
-        r = b - A.dot(xs)
-        print(f"residual norm: {float(r.norm().item()):.3e}")
+        print(f"residual norm: {rn:.3e}")
 else:
     import scipy.sparse as sps
 

@@ -56,13 +56,19 @@ def main():
                               rtol=args.rtol, atol=1e-9)
     ms = timer.stop()
     state = res.y[:, -1]
+    # cost/optimum touch the Hamiltonian's gathered data (collectives):
+    # compute on ALL ranks, print on rank 0
+    cost = mis.cost_function(state)
+    opt = mis.optimum
+    ovl = mis.optimum_overlap(state)
+    ar = mis.approximation_ratio(state)
     if comm.rank() == 0:
         print(f"evolved in {ms:.1f} ms, {len(res.t) - 1} steps, "
               f"|psi|={np.linalg.norm(state):.6f}")
-        print(f"cost <C> = {mis.cost_function(state):.4f} "
-              f"(optimum {mis.optimum:.1f}), "
-              f"optimum overlap = {mis.optimum_overlap(state):.4f}, "
-              f"approx ratio = {mis.approximation_ratio(state):.4f}")
+        print(f"cost <C> = {cost:.4f} "
+              f"(optimum {opt:.1f}), "
+              f"optimum overlap = {ovl:.4f}, "
+              f"approx ratio = {ar:.4f}")
 
 
 if __name__ == "__main__":

@@ -40,6 +40,7 @@ for _ in range(args.iters):
     Ac = P.T @ (A @ P)
 ms_rap = timer.stop() / args.iters
 
+nnz_a, nnz_c, nnz_ac = A.nnz, C.nnz, Ac.nnz  # collectives: all ranks
 if comm.rank() == 0:
-    print(f"A@A:  {ms_aa:.2f} ms/op  (A nnz={A.nnz}, C nnz={C.nnz})")
-    print(f"R@A@P: {ms_rap:.2f} ms/op (Ac nnz={Ac.nnz})")
+    print(f"A@A:  {ms_aa:.2f} ms/op  (A nnz={nnz_a}, C nnz={nnz_c})")
+    print(f"R@A@P: {ms_rap:.2f} ms/op (Ac nnz={nnz_ac})")

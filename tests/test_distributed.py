@@ -40,3 +40,38 @@ def test_distributed_battery(nproc):
         ok = ok and p.returncode == 0
     assert ok, "\n".join(outs[-2:])
     assert any("DIST_ALL_OK" in o for o in outs), outs[0][-2000:]
+
+
+def test_examples_multirank():
+    """The examples themselves must be SPMD-clean (no rank-0-only
+    collectives): run the main ones at ws=2 on gloo."""
+    ex = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                      "examples")
+    cases = [
+        ("pde.py", ["-nx", "120", "-ny", "120", "-throughput", "-max_iter", "30"],
+         "residual norm"),
+        ("gmg.py", ["-N", "31"], "info=0"),
+        ("amg.py", ["-n", "1024"], "info=0"),
+        ("dot_microbenchmark.py", ["-n", "50000", "-iters", "3", "-warmup", "1"],
+         "spmvs"),
+        ("spgemm_microbenchmark.py", ["-nx", "63", "-iters", "2"], "R@A@P"),
+        ("quantum_mis.py", ["-l", "3", "-T", "1.0"], "approx ratio"),
+    ]
+    for i, (script, argv, needle) in enumerate(cases):
+        env = dict(os.environ)
+        env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(29930 + i),
+                    "GLOO_SOCKET_IFNAME": env.get("GLOO_SOCKET_IFNAME", "lo"),
+                    "WORLD_SIZE": "2"})
+        procs = []
+        for r in range(2):
+            e = dict(env)
+            e.update({"RANK": str(r), "LOCAL_RANK": str(r)})
+            procs.append(subprocess.Popen(
+                [sys.executable, os.path.join(ex, script), *argv], env=e,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        outs = []
+        for p in procs:
+            out, _ = p.communicate(timeout=300)
+            outs.append(out.decode(errors="replace"))
+            assert p.returncode == 0, f"{script}: {outs[-1][-1200:]}"
+        assert needle in outs[0], f"{script}: {outs[0][-400:]}"
