@@ -137,9 +137,11 @@ def elem_mult_csr(A, B, vdt):
 def spgemm_csr(A, B, a_col_lo, vdt):
     """Two-phase size-binned Gustavson SpGEMM.
 
-    Rows are binned by product upper bound (sum of touched B-row sizes) into
-    LDS hash sizes 64/256/1024; rows with ub > 512 go through the vectorized
-    expand-sort-reduce path and are scattered into the shared output."""
+    Rows are binned by product upper bound (sum of touched B-row sizes)
+    into LDS hash sizes 64/256/1024/2048; rows beyond that try a CHECKED
+    4096-entry table (dense-product rows usually have few distinct
+    columns) and only true overflows take the expand-sort-reduce
+    fallback."""
     from ..ops.local import LocalCSR
 
     dev = A.device
@@ -166,7 +168,21 @@ def spgemm_csr(A, B, a_col_lo, vdt):
         if rl.numel():
             ext().spgemm_nnz(A.indptr, A.indices, B.indptr, B.indices, rl,
                              counts, int(a_col_lo), H)
-    esc_rows = (ub > 1024).nonzero(as_tuple=False).flatten()
+    # checked speculative bin: dense-PRODUCT rows usually have far fewer
+    # distinct columns than products; try the 4096-entry LDS table first
+    # (overflow -> counts[r] = -1 -> ESC for just those rows).  8-byte
+    # value types only (complex accs exceed the 160 KB LDS budget).
+    big = (ub > 1024).nonzero(as_tuple=False).flatten()
+    checked_ok = torch.zeros(0, dtype=torch.int64, device=dev)
+    if big.numel() and vdt.itemsize <= 8:
+        ext().spgemm_nnz(A.indptr, A.indices, B.indptr, B.indices, big,
+                         counts, int(a_col_lo), 4096)
+        over = counts[big] < 0
+        esc_rows = big[over]
+        checked_ok = big[~over]
+        counts[esc_rows] = 0
+    else:
+        esc_rows = big
     esc_sub = None
     if esc_rows.numel():
         sc = acounts[esc_rows]
@@ -189,6 +205,10 @@ def spgemm_csr(A, B, a_col_lo, vdt):
             ext().spgemm_compute(A.indptr, A.indices, Av, B.indptr, B.indices,
                                  Bv, rl, indptr, indices, values,
                                  int(a_col_lo), H)
+    if checked_ok.numel():
+        ext().spgemm_compute(A.indptr, A.indices, Av, B.indptr, B.indices,
+                             Bv, checked_ok, indptr, indices, values,
+                             int(a_col_lo), 4096)
     if esc_sub is not None and esc_sub.nnz:
         ec = counts[esc_rows]
         eoff = torch.zeros(esc_rows.numel(), dtype=torch.int64, device=dev)

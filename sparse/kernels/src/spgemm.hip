@@ -23,7 +23,12 @@ __device__ __forceinline__ uint32_t hash_mul(int64_t c) {
 
 // LPR = lanes per row (16 or 64); WPB = waves per block;
 // rows per block = WPB*(WAVE/LPR)
-template <typename index_t, int HASH, int LPR, int WPB = WAVES_PER_BLOCK>
+// CHECKED: rows are binned by PRODUCT upper bound, but dense-product rows
+// often have far fewer distinct columns than products; the checked bin
+// admits such rows speculatively and flags table overflow (nnz_out = -1)
+// so the wrapper reroutes only true overflows to the ESC fallback.
+template <typename index_t, int HASH, int LPR, int WPB = WAVES_PER_BLOCK,
+          bool CHECKED = false>
 __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
     const int64_t* __restrict__ aip, const index_t* __restrict__ aix,
     const int64_t* __restrict__ bip, const index_t* __restrict__ bix,
@@ -32,6 +37,7 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
   constexpr int RPB = WPB * (WAVE / LPR);
   __shared__ int64_t keys[RPB][HASH];
   __shared__ int counts[2 * RPB];  // [0,RPB): distinct-col counts; [RPB,2RPB): scan totals
+  __shared__ int ovf[RPB];
   const int slot_id = threadIdx.x / LPR;  // row slot within block
   const int sl = threadIdx.x % LPR;       // lane within row
   const int64_t li = (int64_t)blockIdx.x * RPB + slot_id;
@@ -39,6 +45,7 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
   const int64_t r = rowlist[li];
   for (int i = sl; i < HASH; i += LPR) keys[slot_id][i] = -1;
   if (sl == 0) counts[slot_id] = 0;
+  if (sl == 0) ovf[slot_id] = 0;
   __builtin_amdgcn_wave_barrier();
   __threadfence_block();
   // product-per-lane: tile the A row by LPR entries; an LDS inclusive scan
@@ -48,6 +55,7 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
   __shared__ int bcnt_s[RPB][LPR + 1];
   const int64_t as = aip[r], ae = aip[r + 1];
   for (int64_t t0 = as; t0 < ae; t0 += LPR) {
+    if (CHECKED && ovf[slot_id]) break;
     const int na = (int)min((int64_t)LPR, ae - t0);
     if (sl < na) {
       const int64_t brow = (int64_t)aix[t0 + sl] - a_col_lo;
@@ -69,12 +77,14 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
     __threadfence_block();
     const int total = counts[slot_id + RPB];
     for (int t = sl; t < total; t += LPR) {
+      if (CHECKED && ovf[slot_id]) break;
       // find j: largest with prefix[j] <= t (linear ok: na <= LPR)
       int j = 0;
       for (int j2 = 1; j2 < na; ++j2) j += (bcnt_s[slot_id][j2] <= t);
       const int64_t q = bstart_s[slot_id][j] + (t - bcnt_s[slot_id][j]);
       int64_t c = (int64_t)bix[q];
       uint32_t h = hash_mul(c) & (HASH - 1);
+      int probes = 0;
       while (true) {
         int64_t old = atomicCAS((unsigned long long*)&keys[slot_id][h],
                                 (unsigned long long)(-1ll),
@@ -82,12 +92,18 @@ __global__ __launch_bounds__(256) void spgemm_nnz_kernel(
         if (old == -1ll) { atomicAdd(&counts[slot_id], 1); break; }
         if (old == c) break;
         h = (h + 1) & (HASH - 1);
+        if (CHECKED && ++probes >= HASH) { ovf[slot_id] = 1; break; }
       }
     }
     __builtin_amdgcn_wave_barrier();
     __threadfence_block();
   }
-  if (sl == 0) nnz_out[r] = (int64_t)counts[slot_id];
+  // distinct count must leave sort headroom in the compute phase
+  // (compaction arrays hold HASH/2 entries)
+  if (sl == 0) {
+    const bool over = CHECKED && (ovf[slot_id] || counts[slot_id] > HASH / 2);
+    nnz_out[r] = over ? -1 : (int64_t)counts[slot_id];
+  }
 }
 
 template <typename T, typename index_t, int HASH, int LPR,
@@ -232,7 +248,9 @@ void spgemm_nnz_hip(at::Tensor aip, at::Tensor aix, at::Tensor bip,
     if (hash_size <= 64) launch(spgemm_nnz_kernel<index_t, 64, 16>, 16, 256);
     else if (hash_size <= 256) launch(spgemm_nnz_kernel<index_t, 256, 64>, 4, 256);
     else if (hash_size <= 1024) launch(spgemm_nnz_kernel<index_t, 1024, 64>, 4, 256);
-    else launch(spgemm_nnz_kernel<index_t, 2048, 64, 2>, 2, 128);
+    else if (hash_size <= 2048) launch(spgemm_nnz_kernel<index_t, 2048, 64, 2>, 2, 128);
+    else  // checked speculative bin: overflow -> nnz_out = -1
+      launch(spgemm_nnz_kernel<index_t, 4096, 64, 1, true>, 1, 64);
   });
 }
 
@@ -256,6 +274,14 @@ void spgemm_compute_hip(at::Tensor aip, at::Tensor aix, at::Tensor av,
       };
       if (hash_size <= 64) {
         launch(spgemm_compute_kernel<T, index_t, 64, 16>, 16, 256);
+      } else if (hash_size > 2048) {
+        if constexpr (sizeof(T) <= 8) {
+          launch(spgemm_compute_kernel<T, index_t, 4096, 64, 1>, 1, 64);
+        } else {
+          TORCH_CHECK(false,
+                      "spgemm: the checked 4096 bin supports 8-byte value "
+                      "types only (wrapper routes complex to ESC)");
+        }
       } else if (hash_size <= 256) {
         launch(spgemm_compute_kernel<T, index_t, 256, 64>, 4, 256);
       } else if (hash_size <= 1024) {

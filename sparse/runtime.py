@@ -53,6 +53,12 @@ class Runtime:
         dist.init_process_group(backend=backend)
         self._adopt_dist()
         self._initialized_dist = True
+        # tear the process group down at interpreter exit: gloo/NCCL
+        # destructors racing interpreter shutdown otherwise abort with
+        # "terminate called without an active exception"
+        import atexit
+
+        atexit.register(self.finalize)
 
     def _adopt_dist(self) -> None:
         self.rank = dist.get_rank()

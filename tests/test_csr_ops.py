@@ -102,15 +102,13 @@ def test_dot_vector_matmul_operator():
     assert np.allclose(np.asarray(x @ csr_array(s)), x @ s)
 
 
-def test_spgemm_esc_batching(monkeypatch):
-    """Force tiny ESC batches: batched expansion must agree with scipy on
-    a dense-product multiply (rows with >1024-column products)."""
+def test_spgemm_dense_products_cpu():
+    """Dense-product multiply sanity on the CPU path (the GPU checked-bin
+    and ESC-batching variants live in test_gpu_kernels)."""
     import scipy.sparse as sps
 
-    from sparse import csr_array, kernels
+    from sparse import csr_array
 
-    monkeypatch.setattr(kernels, "_ESC_LIMIT", 500)
-    rng = np.random.default_rng(44)
     a = sps.random(60, 50, 0.4, random_state=45).tocsr()
     b = sps.random(50, 2000, 0.6, random_state=46).tocsr()
     C = csr_array(a) @ csr_array(b)

@@ -638,3 +638,32 @@ def test_integrate_banded_gpu():
     res = integrate.solve_ivp(f, (0, 1.0), y0, method="RK45", rtol=1e-8,
                               atol=1e-10)
     assert res.success
+
+
+def test_spgemm_checked_bin_gpu(monkeypatch):
+    """Rows with product upper bound > 1024 but few distinct columns take
+    the checked 4096-entry bin; true overflows (uniques > 2048) reroute
+    to the (row-batched) ESC fallback."""
+    from sparse import csr_array, kernels
+
+    rng = np.random.default_rng(47)
+    # checked-bin case: ~30K products/row, <= 1500 uniques
+    a = sps.random(40, 2000, 0.5, random_state=48).tocsr()
+    b = sps.random(2000, 1500, 0.01, random_state=49).tocsr()
+    C = csr_array(a) @ csr_array(b)
+    ref = (a @ b).tocsr()
+    assert C.nnz == ref.nnz
+    assert np.allclose(np.asarray(C.todense()), ref.toarray(), rtol=1e-10)
+    # overflow case: uniques ~ 5000 > 2048 -> ESC, with tiny forced batches
+    monkeypatch.setattr(kernels, "_ESC_LIMIT", 4000)
+    b2 = sps.random(2000, 5000, 0.02, random_state=50).tocsr()
+    C2 = csr_array(a) @ csr_array(b2)
+    ref2 = (a @ b2).tocsr()
+    assert C2.nnz == ref2.nnz
+    assert np.allclose(np.asarray(C2.todense()), ref2.toarray(), rtol=1e-10)
+    # complex stays on ESC (checked bin is 8-byte values only)
+    ac = csr_array(a.astype(np.complex128))
+    bc = csr_array(b.astype(np.complex128))
+    Cc = ac @ bc
+    refc = (a.astype(np.complex128) @ b.astype(np.complex128)).tocsr()
+    assert np.allclose(np.asarray(Cc.todense()), refc.toarray(), rtol=1e-10)
