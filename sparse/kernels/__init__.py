@@ -193,7 +193,9 @@ def spgemm_csr(A, B, a_col_lo, vdt):
                - torch.repeat_interleave(sub_ip[:-1], sc)
                + torch.repeat_interleave(A.indptr[esc_rows], sc))
         subA = LocalCSR(sub_ip, A.indices[pos], Av[pos], esc_rows.numel(), A.ncols)
-        esc_sub = _spgemm_esc(subA, B, a_col_lo, vdt)
+        esc_sub = _spgemm_dense_rows(subA, B, a_col_lo, vdt)
+        if esc_sub is None:
+            esc_sub = _spgemm_esc(subA, B, a_col_lo, vdt)
         counts[esc_rows] = esc_sub.indptr[1:] - esc_sub.indptr[:-1]
     indptr = torch.zeros(m + 1, dtype=torch.int64, device=dev)
     torch.cumsum(counts, 0, out=indptr[1:])
@@ -224,6 +226,58 @@ def spgemm_csr(A, B, a_col_lo, vdt):
 
 
 _ESC_LIMIT = 1 << 28  # max materialized products per ESC batch
+_DENSE_FALLBACK_BUDGET = 1 << 30  # max dense-accumulator elements
+
+
+def _spgemm_dense_rows(A, B, a_col_lo, vdt):
+    """Dense-workspace SpGEMM for a FEW very dense rows (> 2048 distinct
+    output columns): scatter-add products into an (nrows, ncols) dense
+    buffer, then compact.  The GPU analog of the reference's CPU dense
+    workspace (spgemm_csr_csr_csr.cc:27-85); returns None when the buffer
+    would exceed the budget (caller falls back to ESC)."""
+    from ..ops.local import LocalCSR
+
+    dev = A.device
+    if A.nrows * B.ncols > _DENSE_FALLBACK_BUDGET:
+        return None
+    acc = torch.zeros(A.nrows * B.ncols, dtype=vdt, device=dev)
+    touched = torch.zeros(A.nrows * B.ncols, dtype=torch.bool, device=dev)
+    acounts = A.indptr[1:] - A.indptr[:-1]
+    arows = torch.repeat_interleave(
+        torch.arange(A.nrows, dtype=torch.int64, device=dev), acounts)
+    acols = A.indices.long() - a_col_lo
+    bstart = B.indptr[acols]
+    bcounts = B.indptr[acols + 1] - bstart
+    CH = _ESC_LIMIT
+    total = int(bcounts.sum().item())
+    cum = torch.cumsum(bcounts, 0)
+    lo_nnz = 0
+    while lo_nnz < acols.numel():
+        base = int(cum[lo_nnz - 1].item()) if lo_nnz else 0
+        hi_nnz = int(torch.searchsorted(cum, base + CH, right=True).item())
+        hi_nnz = max(hi_nnz, lo_nnz + 1)
+        bc = bcounts[lo_nnz:hi_nnz]
+        tot = int(bc.sum().item())
+        offs = torch.zeros(bc.numel(), dtype=torch.int64, device=dev)
+        torch.cumsum(bc[:-1], 0, out=offs[1:])
+        pos = (torch.arange(tot, dtype=torch.int64, device=dev)
+               - torch.repeat_interleave(offs, bc)
+               + torch.repeat_interleave(bstart[lo_nnz:hi_nnz], bc))
+        er = torch.repeat_interleave(arows[lo_nnz:hi_nnz], bc)
+        ev = (torch.repeat_interleave(A.values[lo_nnz:hi_nnz].to(vdt), bc)
+              * B.values[pos].to(vdt))
+        cell = er * B.ncols + B.indices[pos].long()
+        acc.index_add_(0, cell, ev)
+        touched[cell] = True  # structural nnz (cancellation keeps the entry)
+        lo_nnz = hi_nnz
+    acc2 = acc.view(A.nrows, B.ncols)
+    nzmask = touched.view(A.nrows, B.ncols)
+    counts = nzmask.sum(dim=1)
+    indptr = torch.zeros(A.nrows + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    nz = nzmask.nonzero(as_tuple=True)
+    return LocalCSR(indptr, nz[1].to(A.indices.dtype), acc2[nz],
+                    A.nrows, B.ncols)
 
 
 def _spgemm_esc(A, B, a_col_lo, vdt):
