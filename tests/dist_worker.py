@@ -122,6 +122,7 @@ def main():
     extra_spgemm_2d_check()
     extra_domain_part_spmv_check()
     extra_banded_overlap_check()
+    extra_complex_check()
 
     if rank == 0:
         print("DIST_ALL_OK")
@@ -198,6 +199,29 @@ def extra_samplesort_check():
     allv = _c.all_gather_rows(v, [257] * dist.get_world_size()).numpy()
     assert sorted(allk.tolist()) == gk.tolist(), "samplesort permutation"
     assert np.isclose(gv.sum(), allv.sum()), "payload preserved"
+
+
+def extra_complex_check():
+    """Complex dtypes across the wire at ws>1 (view-as-real collectives):
+    SpMV, elementwise, and a bicg solve."""
+    import scipy.sparse as sps7
+
+    from sparse import csr_array as _csr
+    from sparse import linalg as _lin
+
+    rng = np.random.default_rng(71)
+    n = 24
+    s = sps7.random(n, n, 0.3, random_state=72).astype(np.complex128)
+    s.data = s.data + 1j * rng.random(len(s.data))
+    s = (s + n * sps7.eye(n)).tocsr()
+    A = _csr(s)
+    x = rng.random(n) + 1j * rng.random(n)
+    assert np.allclose(np.asarray(A @ x), s @ x), "dist complex spmv"
+    assert np.allclose(np.asarray((A + A).todense()), (s + s).toarray()), \
+        "dist complex add"
+    b = rng.random(n) + 1j * rng.random(n)
+    xs, info = _lin.bicg(A, b, tol=1e-10, maxiter=400, conv_test_iters=2)
+    assert np.allclose(s @ np.asarray(xs), b, atol=1e-5), "dist complex bicg"
 
 
 def extra_banded_overlap_check():
