@@ -20,7 +20,7 @@ def _rand(m, n, density, seed, dt):
     return s.astype(dt)
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=25, deadline=None, derandomize=True)
 @given(m=st.integers(1, 40), n=st.integers(1, 40),
        density=st.floats(0.0, 0.5), seed=st.integers(0, 10**6), dt=DT)
 def test_spmv_matches_scipy(m, n, density, seed, dt):
@@ -31,7 +31,7 @@ def test_spmv_matches_scipy(m, n, density, seed, dt):
     assert np.allclose(got, s @ x, rtol=rtol, atol=1e-6)
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=20, deadline=None, derandomize=True)
 @given(m=st.integers(1, 30), n=st.integers(1, 30), k=st.integers(1, 30),
        d1=st.floats(0.05, 0.5), d2=st.floats(0.05, 0.5),
        seed=st.integers(0, 10**6))
@@ -45,7 +45,7 @@ def test_spgemm_matches_scipy(m, n, k, d1, d2, seed):
                        rtol=1e-10, atol=1e-12)
 
 
-@settings(max_examples=20, deadline=None)
+@settings(max_examples=20, deadline=None, derandomize=True)
 @given(m=st.integers(1, 30), n=st.integers(1, 30),
        d1=st.floats(0.0, 0.5), d2=st.floats(0.0, 0.5),
        seed=st.integers(0, 10**6))
@@ -58,7 +58,7 @@ def test_add_mult_match_scipy(m, n, d1, d2, seed):
     assert np.allclose(np.asarray(M.todense()), a.multiply(b).toarray())
 
 
-@settings(max_examples=15, deadline=None)
+@settings(max_examples=15, deadline=None, derandomize=True)
 @given(m=st.integers(1, 30), n=st.integers(1, 30),
        density=st.floats(0.0, 0.5), seed=st.integers(0, 10**6))
 def test_conversion_roundtrips(m, n, density, seed):
