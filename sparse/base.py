@@ -93,6 +93,34 @@ class CompressedBase:
         s = self.sum(axis=axis, dtype=dtype, out=out)
         return s / denom
 
+    # structures are always column-sorted and duplicate-free here (owner
+    # shuffles sort and sum on construction): scipy API compat no-ops
+    has_sorted_indices = True
+    has_canonical_format = True
+
+    def sort_indices(self):
+        return None
+
+    def sorted_indices(self):
+        return self.copy()
+
+    def sum_duplicates(self):
+        return None
+
+    def eliminate_zeros(self):
+        """Drop stored zeros (in place), scipy semantics."""
+        v = self._values_tensor()
+        if v.numel() == 0:
+            return
+        keep = v != 0
+        if bool(keep.all()):
+            return
+        self._eliminate_zeros_impl(keep)
+
+    def _eliminate_zeros_impl(self, keep):
+        raise NotImplementedError(
+            f"eliminate_zeros not supported for {type(self).__name__}")
+
     def _values_tensor(self) -> torch.Tensor:
         raise NotImplementedError
 

@@ -225,6 +225,31 @@ class csr_array(CompressedBase, DenseSparseBase):
                                       int(self._indices.max().item()) + 1)
         return self._window_cache
 
+    def _eliminate_zeros_impl(self, keep):
+        lc = self.local
+        counts = lc.indptr[1:] - lc.indptr[:-1]
+        rows = torch.repeat_interleave(
+            torch.arange(lc.nrows, dtype=torch.int64, device=keep.device),
+            counts)[keep]
+        self._indices = self._indices[keep]
+        self._values = self._values[keep]
+        newip = torch.zeros(lc.nrows + 1, dtype=torch.int64,
+                            device=self._indptr.device)
+        if rows.numel():
+            torch.cumsum(torch.bincount(rows, minlength=lc.nrows), 0,
+                         out=newip[1:])
+        self._indptr = newip
+        self._invalidate_caches()
+
+    def _invalidate_caches(self):
+        self._ell_cache = None
+        self._dia_cache = None
+        self._csc_cache = None
+        self._maxrow_cache = None
+        self._nnz_cache = None
+        self._window_cache = None
+        self._plan_cache = {}
+
     def _dia_interior(self, dm, plan):
         """Even row bounds [a, b) of the interior — rows whose whole
         window falls inside the own piece, safe to compute while the halo

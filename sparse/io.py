@@ -87,3 +87,29 @@ def mmwrite(path, A, comment: str = "") -> None:
                 f.write(f"{r + 1} {cc + 1} {v.real:.17g} {v.imag:.17g}\n")
             else:
                 f.write(f"{r + 1} {cc + 1} {v:.17g}\n")
+
+
+def save_npz(file, matrix, compressed=True):
+    """Save a sparse array in scipy .npz format (scipy.sparse.save_npz
+    parity — checkpoint/restore superset of the reference, which has
+    read-only mmread).  Gathers to the host on rank 0."""
+    import scipy.sparse as _sp
+
+    from .parallel import comm as _c
+
+    s = matrix.tocsr().to_scipy_sparse_csr()
+    if _c.rank() == 0:
+        _sp.save_npz(file, s, compressed=compressed)
+    if _c.initialized():
+        import torch.distributed as dist
+
+        dist.barrier()
+
+
+def load_npz(file):
+    """Load a scipy .npz sparse file as a distributed csr_array."""
+    import scipy.sparse as _sp
+
+    from .csr import csr_array
+
+    return csr_array(_sp.load_npz(file).tocsr())

@@ -31,7 +31,30 @@ from .parallel import comm
 __all__ = [
     "LinearOperator", "IdentityOperator", "aslinearoperator", "cg", "cgs",
     "bicg", "bicgstab", "gmres", "lsqr", "eigsh", "spsolve", "cg_axpby",
+    "norm",
 ]
+
+
+def norm(A, ord="fro", axis=None):
+    """Matrix norms of a sparse array (scipy.sparse.linalg.norm parity —
+    a superset of the reference, which exposes no norm).  Supports
+    'fro' (default), 1, inf, and elementwise axis=None variants."""
+    if axis is not None:
+        raise NotImplementedError("axis norms are not implemented")
+    v = A._values_tensor()
+    if ord in ("fro", None):
+        local = (v.abs() ** 2).sum() if v.numel() else torch.zeros(
+            (), dtype=torch.float64)
+        local = local.real.double() if local.is_complex() else local.double()
+        comm.all_reduce_(local)
+        return float(local.item()) ** 0.5
+    if ord == 1:  # max column sum of |A|
+        col = abs(A).sum(axis=0)
+        return float(np.max(np.asarray(col)))
+    if ord in (np.inf, float("inf"), "inf"):
+        row = abs(A).sum(axis=1)
+        return float(np.max(np.asarray(row)))
+    raise ValueError(f"norm ord {ord!r} not supported")
 
 
 # -- fused axpby --------------------------------------------------------------

@@ -129,3 +129,34 @@ def test_abs_dunder():
     s = s.tocsr()
     A = csr_array(s)
     assert np.allclose(np.asarray(abs(A).todense()), abs(s).toarray())
+
+
+def test_norm_npz_eliminate():
+    """linalg.norm, save_npz/load_npz round-trip, eliminate_zeros and the
+    sorted-indices compat no-ops (scipy-API supersets)."""
+    import os
+    import tempfile
+
+    import sparse
+    import sparse.linalg as sl
+
+    s = sps.random(20, 30, 0.3, random_state=1, format="csr")
+    A = sparse.csr_array(s)
+    assert np.isclose(sl.norm(A), sps.linalg.norm(s))
+    assert np.isclose(sl.norm(A, 1), sps.linalg.norm(s, 1))
+    assert np.isclose(sl.norm(A, np.inf), sps.linalg.norm(s, np.inf))
+    with tempfile.TemporaryDirectory() as d:
+        p = os.path.join(d, "x.npz")
+        sparse.io.save_npz(p, A)
+        B = sparse.io.load_npz(p)
+        assert np.allclose(np.asarray(B.todense()), s.toarray())
+    z = s.copy()
+    z.data[::3] = 0.0
+    Az = sparse.csr_array(z)
+    Az.eliminate_zeros()
+    zc = z.copy()
+    zc.eliminate_zeros()
+    assert Az.nnz == zc.nnz
+    assert np.allclose(np.asarray(Az.todense()), zc.toarray())
+    assert A.has_sorted_indices and A.sort_indices() is None
+    assert A.sum_duplicates() is None
