@@ -225,6 +225,48 @@ class csr_array(CompressedBase, DenseSparseBase):
                                       int(self._indices.max().item()) + 1)
         return self._window_cache
 
+    def __getitem__(self, key):
+        """Minimal indexing (scipy-API superset; the reference supports no
+        indexing): A[i] -> 1-row csr, A[i:j] -> row-slice csr (step 1),
+        A[i, j] -> scalar.  Row slices repartition collectively."""
+        from .parallel.shuffle import gather_csr_rows
+
+        if isinstance(key, tuple) and len(key) == 2 and all(
+                isinstance(k, (int, np.integer)) for k in key):
+            i, j = int(key[0]), int(key[1])
+            if i < 0:
+                i += self.shape[0]
+            if j < 0:
+                j += self.shape[1]
+            row = self[i]
+            idx = row._indices
+            vals = row._values
+            hit = (idx.long() == j).nonzero(as_tuple=True)[0]
+            v = vals[hit[0]].item() if hit.numel() else self.dtype.type(0)
+            # replicated row: every rank returns the same scalar
+            return self.dtype.type(v)
+        if isinstance(key, (int, np.integer)):
+            i = int(key)
+            if i < 0:
+                i += self.shape[0]
+            if not 0 <= i < self.shape[0]:
+                raise IndexError(f"row {key} out of range")
+            key = slice(i, i + 1)
+        if isinstance(key, slice):
+            start, stop, step = key.indices(self.shape[0])
+            if step != 1:
+                raise NotImplementedError("only step-1 row slices")
+            nrows = max(0, stop - start)
+            part = RowPartition.equal(nrows, comm.world_size())
+            me = comm.rank()
+            lo = start + part.start(me)
+            hi = start + part.stop(me)
+            ip, ix, vs = gather_csr_rows(self._indptr, self._indices,
+                                         self._values, self.partition, lo, hi)
+            return csr_array.from_local(ip, ix, vs, part,
+                                        (nrows, self.shape[1]))
+        raise NotImplementedError(f"indexing with {key!r} is not supported")
+
     def _eliminate_zeros_impl(self, keep):
         lc = self.local
         counts = lc.indptr[1:] - lc.indptr[:-1]

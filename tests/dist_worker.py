@@ -123,6 +123,7 @@ def main():
     extra_domain_part_spmv_check()
     extra_banded_overlap_check()
     extra_complex_check()
+    extra_indexing_check()
 
     if rank == 0:
         print("DIST_ALL_OK")
@@ -199,6 +200,20 @@ def extra_samplesort_check():
     allv = _c.all_gather_rows(v, [257] * dist.get_world_size()).numpy()
     assert sorted(allk.tolist()) == gk.tolist(), "samplesort permutation"
     assert np.isclose(gv.sum(), allv.sum()), "payload preserved"
+
+
+def extra_indexing_check():
+    """Row-slice indexing repartitions collectively at ws>1."""
+    import scipy.sparse as sps8
+
+    from sparse import csr_array as _csr
+
+    s = sps8.random(23, 17, 0.3, random_state=81, format="csr")
+    A = _csr(s)
+    assert np.allclose(np.asarray(A[4:19].todense()), s[4:19].toarray()), \
+        "dist row slice"
+    assert np.allclose(np.asarray(A[7].todense()), s[[7]].toarray()), \
+        "dist row"
 
 
 def extra_complex_check():

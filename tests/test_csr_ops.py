@@ -115,3 +115,17 @@ def test_spgemm_dense_products_cpu():
     ref = (a @ b).tocsr()
     assert C.nnz == ref.nnz
     assert np.allclose(np.asarray(C.todense()), ref.toarray(), rtol=1e-10)
+
+
+def test_indexing():
+    """A[i], A[i:j], A[i, j] (superset: the reference supports no
+    indexing)."""
+    s = sample_csr(12, 9, 0.4, seed=55)
+    A = csr_array(s)
+    assert np.allclose(np.asarray(A[3].todense()), s[[3]].toarray())
+    assert np.allclose(np.asarray(A[2:9].todense()), s[2:9].toarray())
+    assert np.allclose(np.asarray(A[-1].todense()), s[[-1]].toarray())
+    assert np.isclose(A[3, 4], s[3, 4])
+    assert A[5:5].shape == (0, 9)
+    with pytest.raises(IndexError):
+        A[99]
