@@ -249,7 +249,6 @@ def _spgemm_dense_rows(A, B, a_col_lo, vdt):
     bstart = B.indptr[acols]
     bcounts = B.indptr[acols + 1] - bstart
     CH = _ESC_LIMIT
-    total = int(bcounts.sum().item())
     cum = torch.cumsum(bcounts, 0)
     lo_nnz = 0
     while lo_nnz < acols.numel():
