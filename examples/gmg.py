@@ -248,9 +248,16 @@ def main():
     Mop.matvec(b)
     it_count = [0]
     t0.start()
-    x, info = linalg.cg(A, b, M=Mop, tol=args.tol,
-                        maxiter=args.maxiter, conv_test_iters=5,
-                        callback=lambda _x: it_count.__setitem__(0, it_count[0] + 1))
+    if args.throughput:
+        # fixed-iteration throughput protocol (like pde.py -throughput)
+        x, info = linalg.cg(A, b, M=Mop, tol=0.0, atol=0.0,
+                            maxiter=args.maxiter, conv_test_iters=None)
+        it_count[0] = args.maxiter
+    else:
+        x, info = linalg.cg(A, b, M=Mop, tol=args.tol,
+                            maxiter=args.maxiter, conv_test_iters=5,
+                            callback=lambda _x: it_count.__setitem__(
+                                0, it_count[0] + 1))
     solve_ms = t0.stop()
     r = b - A.dot(x)  # collective: all ranks participate
     rn = float(r.norm().item())  # all-reduce: all ranks

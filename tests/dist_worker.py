@@ -83,8 +83,21 @@ def main():
     skew = (skew + sps.csr_matrix((np.ones(30), (np.zeros(30, dtype=int),
                                                  np.arange(30))), (40, 40))).tocsr()
     Ask = csr_array(skew)
+    part_before = Ask.partition
     Ask.balance()
     assert np.allclose(np.asarray(Ask.todense()), skew.toarray()), "balance"
+    # white-box (reference test_csr_spmm.py:79-105 checks the partitioner
+    # actually picked a balanced strategy): slabs must change for a skewed
+    # matrix and per-rank nnz must be near-equal
+    assert Ask.partition != part_before, "balance changed nothing"
+    from sparse.parallel import comm as _cb
+
+    me_n = Ask._values.numel()
+    tot = torch.zeros(1, dtype=torch.int64)
+    tot[0] = me_n
+    _cb.all_reduce_(tot)
+    avg = float(tot.item()) / dist.get_world_size()
+    assert me_n <= 2.0 * avg + 64, "balance left a rank overloaded"
     xb = rng.random(40)
     assert np.allclose(np.asarray(Ask @ xb), skew @ xb), "balanced spmv"
 
