@@ -25,3 +25,13 @@ python examples/spgemm_microbenchmark.py -nx 2047 -iters 10
 
 # quantum MIS demo
 python examples/quantum_mis.py -l 4 -T 3.0
+
+# capacity: 604M rows / 3.02B nnz on ONE GPU (288 GB sizing)
+python bench.py --nx 24576 --steps 30 --warmup 5
+
+# fp32 option
+python bench.py --nx 8192 --dtype fp32 --steps $STEPS --warmup 10
+
+# 8-GPU strong scaling (run on an 8-GPU node; the driver does this too)
+# python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+#     --master-addr 127.0.0.1 bench.py --gpus 8 --steps $STEPS --warmup 15
