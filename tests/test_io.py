@@ -30,3 +30,23 @@ def test_mmwrite_roundtrip(tmp_path):
     sio_ours.mmwrite(p, A)
     back = sio_ours.mmread(p)
     assert np.allclose(np.asarray(back.todense()), np.asarray(A.todense()))
+
+
+def test_mmread_integer_field(tmp_path):
+    p = tmp_path / "i.mtx"
+    p.write_text("%%MatrixMarket matrix coordinate integer general\n"
+                 "3 3 3\n1 1 5\n2 3 -2\n3 2 7\n")
+    A = sio_ours.mmread(str(p))
+    ref = np.zeros((3, 3))
+    ref[0, 0], ref[1, 2], ref[2, 1] = 5, -2, 7
+    assert np.allclose(np.asarray(A.todense()), ref)
+
+
+def test_mmread_skew_symmetric(tmp_path):
+    import scipy.io as sio
+
+    p = tmp_path / "s.mtx"
+    p.write_text("%%MatrixMarket matrix coordinate real skew-symmetric\n"
+                 "3 3 2\n2 1 1.5\n3 1 -2.0\n")
+    A = sio_ours.mmread(str(p))
+    assert np.allclose(np.asarray(A.todense()), sio.mmread(str(p)).toarray())
