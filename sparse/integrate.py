@@ -263,12 +263,15 @@ def _dop853_err_norm(self_obj, est, scale, n):
     s3 = _sq(err3 / scale)
     t = torch.stack([s5, s3])
     comm.all_reduce_(t)
-    e5 = float(t[0].item()) / n
-    e3 = float(t[1].item()) / n
-    denom = e5 + 0.01 * e3
+    # scipy's 8(5,3) norm: |h| * s5 / sqrt((s5 + 0.01*s3) * n) on the
+    # UN-normalized global squared sums (reference integrate.py:1117-1130);
+    # dividing by n first made step control ~sqrt(n) looser than rtol.
+    s5 = float(t[0].item())
+    s3 = float(t[1].item())
+    denom = s5 + 0.01 * s3
     if denom <= 0:
         return 0.0
-    return abs(h) * e5 / math.sqrt(denom * n)
+    return abs(h) * s5 / math.sqrt(denom * n)
 
 
 # patch the norm dispatch: RungeKutta.step calls _err_norm on locals; DOP853

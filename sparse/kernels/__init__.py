@@ -58,6 +58,10 @@ def spmv_dot(A, x, y, p, dot_out, col_lo: int):
 def axpby_norm2(y, x, a, b, isalpha, negate):
     """Fused axpby; returns sum(y_new^2) (real dtypes) as a 0-dim tensor."""
     n = y.numel()
+    if n == 0:
+        # the HIP launcher returns early on n==0 without writing `partial`;
+        # an empty slab must contribute an exact 0 to the all-reduced dot
+        return torch.zeros((), dtype=y.dtype, device=y.device)
     ept = max(1, 16 // y.element_size())
     blocks = (n // ept + 255) // 256 + 1
     partial = torch.empty(blocks, dtype=y.dtype, device=y.device)
@@ -474,6 +478,8 @@ def cg_xr_norm2(x, p, r, q, a, b):
     """Fused CG K2: x += (a/b)p; r -= (a/b)q; returns local sum(r_new^2)
     as a device 0-dim tensor."""
     n = x.numel()
+    if n == 0:
+        return torch.zeros((), dtype=x.dtype, device=x.device)
     ept = max(1, 16 // x.element_size())
     blocks = (n // ept + 255) // 256 + 1
     partial = torch.empty(blocks, dtype=x.dtype, device=x.device)

@@ -244,8 +244,18 @@ def cg(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
     fused = (ident_M and isinstance(A, _SparseMatrixLinearOperator)
              and getattr(A.A, "_format", None) == "csr"
              and b.local.is_cuda and not b.local.is_complex())
-    if (fused and A.A._dia() is not None
-            and os.environ.get("SPARSE_CG2") == "1"):
+    use_cg2 = fused and os.environ.get("SPARSE_CG2") == "1"
+    if use_cg2 and comm.world_size() > 1:
+        # the DIA mirror can build on some ranks and not others (memory
+        # headroom, non-uniform slabs); a rank-divergent branch choice means
+        # divergent collective sequences → deadlock.  All-reduce(min) the flag.
+        flag = torch.tensor(
+            [1.0 if A.A._dia() is not None else 0.0], device=b.local.device)
+        comm.all_reduce_(flag, op="min")
+        use_cg2 = bool(flag.item() > 0.5)
+    elif use_cg2:
+        use_cg2 = A.A._dia() is not None
+    if use_cg2:
         # two-kernel CG iteration on the DIA fast path: K1 folds the
         # p-update into the SpMV (p = r + beta p, q = Ap, p.q), K2 fuses
         # x += alpha p, r -= alpha q and |r|^2.  MEASURED SLOWER than the

@@ -191,6 +191,18 @@ def kron(A, B, format=None):
     return out.asformat(format)
 
 
+def _sample_flat_dedup(rng, mn: int, nnz: int) -> np.ndarray:
+    """nnz distinct flat indices in [0, mn) drawn with replacement +
+    dedup — memory-safe for huge mn.  Redraws until collisions leave at
+    least nnz distinct samples (a fixed margin under-fills at high
+    density, yielding rows/cols shorter than vals)."""
+    flat = np.unique(rng.integers(0, mn, size=int(nnz * 1.05) + 16))
+    while flat.size < nnz:
+        extra = rng.integers(0, mn, size=int(nnz * 0.25) + 16)
+        flat = np.unique(np.concatenate([flat, extra]))
+    return rng.permutation(flat)[:nnz]
+
+
 def random(m, n, density=0.01, format="coo", dtype=np.float64, random_state=None,
            data_rvs=None):
     """Seeded random sparse matrix (reference module.py:360-510).
@@ -209,10 +221,10 @@ def random(m, n, density=0.01, format="coo", dtype=np.float64, random_state=None
     elif m * n <= 100_000_000:
         flat = rng.choice(m * n, size=nnz, replace=False)
     else:
-        flat = np.unique(rng.integers(0, m * n, size=int(nnz * 1.05) + 16))
-        flat = rng.permutation(flat)[:nnz]
+        flat = _sample_flat_dedup(rng, m * n, nnz)
     rows = flat // n
     cols = flat % n
+    assert rows.shape[0] == nnz
     if data_rvs is None:
         vals = rng.random(nnz)
     else:

@@ -43,6 +43,22 @@ def test_matches_scipy_trajectory(method):
     assert np.allclose(ours.y[:, -1], ref.y[:, -1], rtol=1e-6, atol=1e-9)
 
 
+def test_dop853_step_count_matches_scipy():
+    """The 8(5,3) error norm must use the un-normalized squared sums
+    (|h|*s5/sqrt((s5+0.01*s3)*n)); a 1/sqrt(n)-loose norm accepts far
+    fewer/larger steps than scipy at the same rtol."""
+    n = 400
+    y0 = np.linspace(1.0, 2.0, n)
+    ours = integrate.solve_ivp(exp_decay, (0.0, 10.0), y0, method="DOP853",
+                               rtol=1e-10, atol=1e-12)
+    ref = sp_solve_ivp(lambda t, y: -0.5 * y, (0.0, 10.0), y0,
+                       method="DOP853", rtol=1e-10, atol=1e-12)
+    assert ours.success and ref.success
+    # same step controller => step counts within 25% of each other
+    assert abs(len(ours.t) - len(ref.t)) <= max(3, 0.25 * len(ref.t))
+    assert np.allclose(ours.y[:, -1], ref.y[:, -1], rtol=1e-8)
+
+
 def test_t_eval_and_dense_output():
     y0 = np.array([1.0])
     t_eval = np.linspace(0, 3, 7)

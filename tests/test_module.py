@@ -65,6 +65,33 @@ def test_random():
     assert (d != 0).sum() == r.nnz
 
 
+def test_random_dedup_sampler_high_density():
+    """Huge-matrix sampler must return exactly nnz distinct indices even
+    when collisions exceed the first-draw margin (high density)."""
+    from sparse.module import _sample_flat_dedup
+
+    rng = np.random.default_rng(7)
+    mn, nnz = 1000, 950  # 95% density: one 5%-margin draw WILL collide short
+    flat = _sample_flat_dedup(rng, mn, nnz)
+    assert flat.shape[0] == nnz
+    assert np.unique(flat).shape[0] == nnz
+    assert flat.min() >= 0 and flat.max() < mn
+
+
+def test_fused_norm_wrappers_empty_slab():
+    """axpby_norm2 / cg_xr_norm2 must contribute an exact 0 for an empty
+    local slab (uninitialized partial buffer corrupted CG)."""
+    import torch
+    from sparse import kernels
+
+    e = torch.zeros(0, dtype=torch.float64)
+    z = kernels.axpby_norm2(e, e, torch.tensor(1.0), torch.tensor(1.0),
+                            True, False)
+    assert float(z) == 0.0
+    z2 = kernels.cg_xr_norm2(e, e, e, e, torch.tensor(1.0), torch.tensor(1.0))
+    assert float(z2) == 0.0
+
+
 def test_issparse_predicates():
     A = sparse.eye(3)
     assert sparse.issparse(A)
