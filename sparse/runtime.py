@@ -1,11 +1,19 @@
-"""Runtime singleton: device selection, distributed context, HIP streams.
+"""Runtime singleton: device selection, distributed context.
 
 This replaces the Legion/Legate runtime of the reference
 (sparse/runtime.py:75-126): instead of a dynamic partitioning solver we run
 SPMD — one process per GPU (torch.distributed, backend "nccl" == RCCL on
 ROCm, "gloo" on CPU) — and every distributed op issues its own explicit
-collectives.  A side HIP stream is kept for overlapping halo exchange with
-interior compute.
+collectives.
+
+Overlap model (RCCL): torch.distributed's NCCL process group launches every
+collective/p2p on its OWN internal HIP stream per device; posting
+`batch_isend_irecv` returns immediately, compute kernels enqueued afterwards
+on the default stream run concurrently with the RCCL kernels over xGMI, and
+`work.wait()` only makes the CURRENT stream wait on the comm-stream event
+(no host block).  Halo/interior overlap (gather_halos_begin → interior
+kernels → gather_halos_end → boundary kernels) therefore needs no extra
+application-managed stream; we keep none.
 """
 from __future__ import annotations
 
@@ -34,11 +42,8 @@ class Runtime:
                 "LOCAL_RANK", self.rank)) % max(1, torch.cuda.device_count())
             torch.cuda.set_device(local_rank)
             self.device = torch.device("cuda", local_rank)
-            # Side stream for halo/collective overlap with interior compute.
-            self.comm_stream = torch.cuda.Stream()
         else:
             self.device = torch.device("cpu")
-            self.comm_stream = None
         self.num_gpus = torch.cuda.device_count() if self.use_gpu else 0
 
     # -- distributed bring-up -------------------------------------------------

@@ -16,7 +16,10 @@ import torch.distributed as dist
 
 
 def main():
-    dist.init_process_group("gloo")
+    backend = os.environ.get("SPARSE_DIST_BACKEND", "gloo")
+    if backend == "nccl":
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    dist.init_process_group(backend)
     import sparse
     from sparse import coo_array, csc_array, csr_array, darray, gallery, linalg
     from sparse.parallel import comm
