@@ -66,7 +66,8 @@ class WindowGatherPlan:
         ws = comm.world_size(self.group)
         me = comm.rank(self.group)
         xs, xe = self.xpart.start(me), self.xpart.stop(me)
-        own_a, own_b = max(self.lo, xs), min(self.hi, xe)
+        own_a = max(self.lo, xs)
+        own_b = max(own_a, min(self.hi, xe))  # clamp: window may miss my slab
         send_views = []
         for p in range(ws):
             if p == me:

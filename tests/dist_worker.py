@@ -132,6 +132,7 @@ def main():
 
     extra_samplesort_check()
     extra_halo_check()
+    extra_plan_stress_check()
     extra_solver_checks()
     extra_num_procs_check()
     extra_precise_images_check()
@@ -190,6 +191,80 @@ def extra_halo_check():
         hlo, own, hhi = plan.gather_halos(x)
         assert torch.equal(torch.cat([hlo, own, hhi]), full2), (rank, lo, hi, "ctx")
         x.mul_(0.5)
+
+
+def extra_plan_stress_check():
+    """plan math at the edges — windows spanning 0/1/all
+    ranks, empty slabs (n < ws), empty windows, complex halos, and
+    reduce-scatter — all against dense oracles."""
+    import torch
+    from sparse.parallel import comm as _c
+    from sparse.parallel.gather import (PreciseGatherPlan, ReduceScatterPlan,
+                                        WindowGatherPlan)
+    from sparse.parallel.partition import RowPartition
+
+    ws = dist.get_world_size()
+    rank = dist.get_rank()
+
+    for n, dtype in [(3, torch.float64),            # empty slabs: n < ws
+                     (ws, torch.float64),           # exactly one row each
+                     (57, torch.complex128),        # complex over the wire
+                     (57, torch.float64)]:
+        part = RowPartition.equal(n, ws)
+        s0, s1 = part.start(rank), part.stop(rank)
+        glob = (torch.arange(1, n + 1, dtype=torch.float64)
+                .to(dtype))
+        if dtype.is_complex:
+            glob = glob + 1j * torch.arange(n, dtype=torch.float64)
+        x = glob[s0:s1].clone()
+        windows = [(0, n),                       # all ranks
+                   (0, 0), (n, n),               # empty windows
+                   (s0, s1),                     # my slab only
+                   (min(1, n), n - min(1, n))]   # shrunk, may skip ranks
+        if n > 8:
+            windows += [(2, 5), (n - 4, n - 1)]  # single-rank windows
+        for lo, hi in windows:
+            lo, hi = min(lo, hi), max(lo, hi)
+            plan = WindowGatherPlan(lo, hi, part)
+            full = plan.gather(x)
+            assert torch.equal(full, glob[lo:hi]), (rank, n, lo, hi, "gather")
+            hlo, own, hhi = plan.gather_halos(x)
+            assert torch.equal(torch.cat([hlo, own, hhi]), glob[lo:hi]), \
+                (rank, n, lo, hi, "halos")
+            # reduce-scatter inverse: every rank contributes ones over its
+            # (possibly rank-dependent) window; owners must accumulate one
+            # contribution per covering rank
+            y = torch.zeros(s1 - s0, dtype=dtype)
+            rs = ReduceScatterPlan(lo, hi, part)
+            rs.scatter_add(torch.ones(hi - lo, dtype=dtype), y)
+            cov = torch.zeros(n, dtype=torch.float64)
+            cov[lo:hi] = 1.0
+            _c.all_reduce_(cov)
+            expect = cov.to(dtype)
+            assert torch.equal(y, expect[s0:s1]), (rank, n, lo, hi, "rs")
+        # precise plan with rank-dependent (incl. empty) request sets
+        idx = torch.arange(rank % (n + 1), dtype=torch.int64) % max(1, n)
+        pp = PreciseGatherPlan(idx, part)
+        got = pp.gather(x)
+        assert torch.equal(got, glob[pp.cols]), (rank, n, "precise")
+
+    # 2-D operand through a window plan (SpMM-shaped gather)
+    part = RowPartition.equal(11, ws)
+    s0, s1 = part.start(rank), part.stop(rank)
+    G = torch.arange(33, dtype=torch.float64).reshape(11, 3)
+    plan = WindowGatherPlan(2, 9, part)
+    assert torch.equal(plan.gather(G[s0:s1].clone()), G[2:9]), "2d window"
+
+    # tiny-n CG with empty slabs: empty ranks must contribute exact zeros
+    # to every all-reduced scalar (ADVICE r1 fused-norm guard, CPU analog)
+    from sparse import csr_array as _csr, linalg as _lin
+    import scipy.sparse as spsx
+
+    a = spsx.csr_matrix(np.array([[4.0, 1.0], [1.0, 3.0]]))
+    xs, info = _lin.cg(_csr(a), np.array([1.0, 2.0]), tol=1e-12,
+                       conv_test_iters=1)
+    assert info == 0 and np.allclose(a @ np.asarray(xs), [1.0, 2.0],
+                                     atol=1e-9), "tiny cg"
 
 
 def extra_samplesort_check():
