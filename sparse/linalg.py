@@ -340,11 +340,14 @@ def spsolve(A, b, **kwargs):
 @track_provenance(nested=True)
 def cgs(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
         conv_test_iters=25):
-    """Conjugate gradient squared (reference linalg.py:570-615)."""
+    """Preconditioned conjugate gradient squared (reference
+    linalg.py:570-615; M semantics per scipy.sparse.linalg.cgs)."""
     A = aslinearoperator(A)
     b = _vec(b, A.dtype)
     n = b.shape[0]
     maxiter = maxiter or n * 10
+    ident_M = M is None or isinstance(M, IdentityOperator)
+    Mop = aslinearoperator(M) if not ident_M else None
     x = _vec(x0, A.dtype).copy() if x0 is not None else darray.zeros((n,), dtype=A.dtype)
     r = b - A.matvec(x)
     rtilde = r.copy()
@@ -362,13 +365,15 @@ def cgs(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
             u = r + q_ * beta
             p = u + (q_ + p * beta) * beta
         rho = rho_new
-        vhat = A.matvec(p)
+        phat = p if ident_M else Mop.matvec(p)
+        vhat = A.matvec(phat)
         sigma = rtilde.dot(vhat)
         alpha = rho / sigma
         q_ = u - alpha * vhat
         uq = u + q_
-        x += uq * alpha
-        r -= A.matvec(uq) * alpha
+        uqhat = uq if ident_M else Mop.matvec(uq)
+        x += uqhat * alpha
+        r -= A.matvec(uqhat) * alpha
         if conv_test_iters and (i % conv_test_iters == 0 or i == maxiter - 1):
             if float(r.norm().item()) < threshold:
                 info = 0
@@ -384,17 +389,23 @@ def cgs(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
 @track_provenance(nested=True)
 def bicg(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
          conv_test_iters=25):
-    """Biconjugate gradients (reference linalg.py:620-665)."""
+    """Preconditioned biconjugate gradients (reference linalg.py:620-665;
+    M semantics per scipy.sparse.linalg.bicg — z = M r, ztilde = M^H
+    rtilde)."""
     A = aslinearoperator(A)
     b = _vec(b, A.dtype)
     n = b.shape[0]
     maxiter = maxiter or n * 10
+    ident_M = M is None or isinstance(M, IdentityOperator)
+    Mop = aslinearoperator(M) if not ident_M else None
     x = _vec(x0, A.dtype).copy() if x0 is not None else darray.zeros((n,), dtype=A.dtype)
     r = b - A.matvec(x)
     rtilde = r.copy()
-    p = r.copy()
-    ptilde = rtilde.copy()
-    rho = rtilde.dot(r)
+    z = r if ident_M else Mop.matvec(r)
+    ztilde = rtilde if ident_M else Mop.rmatvec(rtilde)
+    p = z.copy()
+    ptilde = ztilde.copy()
+    rho = rtilde.dot(z)
     bnorm = float(b.norm().item()) or 1.0
     threshold = _tols(bnorm, tol, atol)
     info = maxiter
@@ -409,11 +420,13 @@ def bicg(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None
             if float(r.norm().item()) < threshold:
                 info = 0
                 break
-        rho_new = rtilde.dot(r)
+        z = r if ident_M else Mop.matvec(r)
+        ztilde = rtilde if ident_M else Mop.rmatvec(rtilde)
+        rho_new = rtilde.dot(z)
         beta = rho_new / rho
         rho = rho_new
-        p = r + p * beta
-        ptilde = ptilde * beta.conj() + rtilde
+        p = z + p * beta
+        ptilde = ptilde * beta.conj() + ztilde
         if callback is not None:
             callback(x)
     if info != 0 and float(r.norm().item()) < threshold:
@@ -426,11 +439,14 @@ def bicg(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None
 def bicgstab(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None,
              atol=None, conv_test_iters=25):
     """BiCGSTAB (the reference's is commented 'Doesnt work',
-    linalg.py:795-...; this one follows the standard Van der Vorst form)."""
+    linalg.py:795-...; this follows the standard Van der Vorst form with a
+    right preconditioner M per scipy: x-updates use the M-hatted vectors)."""
     A = aslinearoperator(A)
     b = _vec(b, A.dtype)
     n = b.shape[0]
     maxiter = maxiter or n * 10
+    ident_M = M is None or isinstance(M, IdentityOperator)
+    Mop = aslinearoperator(M) if not ident_M else None
     x = _vec(x0, A.dtype).copy() if x0 is not None else darray.zeros((n,), dtype=A.dtype)
     r = b - A.matvec(x)
     rhat = r.copy()
@@ -447,12 +463,14 @@ def bicgstab(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None,
             beta = (rho_new / rho) * (alpha / omega)
             p = r + (p - v * omega) * beta
         rho = rho_new
-        v = A.matvec(p)
+        phat = p if ident_M else Mop.matvec(p)
+        v = A.matvec(phat)
         alpha = rho / rhat.dot(v)
         s = r - v * alpha
-        t = A.matvec(s)
+        shat = s if ident_M else Mop.matvec(s)
+        t = A.matvec(shat)
         omega = t.dot(s) / t.dot(t)
-        x += p * alpha + s * omega
+        x += phat * alpha + shat * omega
         r = s - t * omega
         if conv_test_iters and (i % conv_test_iters == 0 or i == maxiter - 1):
             if float(r.norm().item()) < threshold:
