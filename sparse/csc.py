@@ -336,20 +336,31 @@ class csc_array(CompressedBase, DenseSparseBase):
     def sddmm(self, C, D):
         """vals'[i,j] = vals[i,j] * (C[i,:] @ D[:,j]) on CSC structure
         (reference csc.py:495,556-...)."""
+        from .parallel.gather import ColBlockGatherPlan
+
         C = asdistarray(C)
         D = asdistarray(D)
         me = comm.rank()
         c0, c1 = self.partition.start(me), self.partition.stop(me)
-        Dg = D.gather()  # (k, n)
-        Cg = C.gather()  # (m, k)
+        # operand-block gathers (reference csr.py:1244-1312 mirrored):
+        # D columns for MY col slab; C rows for my min/max ROW window only
+        key = ("sddmm_d", D.partition.starts)
+        if key not in self._plan_cache:
+            self._plan_cache[key] = ColBlockGatherPlan(c0, c1, D.partition)
+        Dblk = self._plan_cache[key].gather(D.local)  # (k, c1-c0)
+        rlo, rhi = self._row_window()
+        key = ("sddmm_c", C.partition.starts)
+        if key not in self._plan_cache:
+            self._plan_cache[key] = WindowGatherPlan(rlo, rhi, C.partition)
+        Cblk = self._plan_cache[key].gather(C.local)  # (rhi-rlo, k)
         vdt = common_value_dtype(self._values.dtype,
-                                 common_value_dtype(Cg.dtype, Dg.dtype))
+                                 common_value_dtype(Cblk.dtype, Dblk.dtype))
         # local is CSR of A^T: rows = my cols j, entries at global rows i.
-        # out[nz at (j,i)] = vals * (D^T[j,:] @ C^T[:,i])
+        # out[nz at (j,i)] = vals * (D^T[j,:] @ C^T[:,i - rlo])
         lc = ops.LocalCSR(self._colptr, self._indices, self._values.to(vdt),
                           c1 - c0, self.shape[0])
-        out = ops.sddmm(lc, Dg.T[c0:c1].contiguous().to(vdt),
-                        Cg.T.contiguous().to(vdt))
+        out = ops.sddmm(lc, Dblk.T.contiguous().to(vdt),
+                        Cblk.T.contiguous().to(vdt), col_lo=rlo)
         return csc_array.from_local(self._colptr, self._indices, out,
                                     self.partition, self.shape)
 

@@ -830,19 +830,38 @@ class csr_array(CompressedBase, DenseSparseBase):
                                     part, (m, n))
 
     def sddmm(self, C, D) -> "csr_array":
-        """vals'[i,j] = vals[i,j] * (C[i,:] @ D[:,j]) (reference csr.py:693-705)."""
+        """vals'[i,j] = vals[i,j] * (C[i,:] @ D[:,j]) (reference csr.py:693-705).
+
+        Operand-block gathers (reference csr.py:1244-1312): C rows are
+        fetched only for MY row slab (window gather over C's partition);
+        D is fetched as the COLUMN block of my min/max column window
+        (MinMaxImage proj dim 1) — per-rank D traffic = k x window, not
+        k x n."""
+        from .parallel.gather import ColBlockGatherPlan
+
         C = asdistarray(C)
         D = asdistarray(D)
         r = comm.rank()
-        Clocal = C.local if C.partition == self.partition else C.gather()[
-            self.partition.start(r): self.partition.stop(r)]
-        Dg = D.gather()
+        if C.partition == self.partition:
+            Clocal = C.local
+        else:
+            key = ("sddmm_c", C.partition.starts)
+            if key not in self._plan_cache:
+                self._plan_cache[key] = WindowGatherPlan(
+                    self.partition.start(r), self.partition.stop(r),
+                    C.partition)
+            Clocal = self._plan_cache[key].gather(C.local)
+        lo, hi = self._col_window()
+        key = ("sddmm_d", D.partition.starts)
+        if key not in self._plan_cache:
+            self._plan_cache[key] = ColBlockGatherPlan(lo, hi, D.partition)
+        Dblk = self._plan_cache[key].gather(D.local)
         vdt = common_value_dtype(self._values.dtype,
-                                 common_value_dtype(Clocal.dtype, Dg.dtype))
+                                 common_value_dtype(Clocal.dtype, Dblk.dtype))
         lc = self.local
         out = ops.sddmm(ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt),
                                      lc.nrows, lc.ncols),
-                        Clocal.to(vdt), Dg.to(vdt))
+                        Clocal.to(vdt), Dblk.to(vdt), col_lo=lo)
         return csr_array.from_local(self._indptr, self._indices, out,
                                     self.partition, self.shape)
 

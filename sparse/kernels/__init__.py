@@ -77,8 +77,8 @@ def rspmm(B, A_dense, C):
     ext().rspmm(B.indptr, B.indices, B.values, A_dense, C)
 
 
-def sddmm(A, C, D, out):
-    ext().sddmm(A.indptr, A.indices, A.values, C, D, out)
+def sddmm(A, C, D, out, col_lo: int = 0):
+    ext().sddmm(A.indptr, A.indices, A.values, C, D, out, int(col_lo))
 
 
 def mult_dense(A, D, out):

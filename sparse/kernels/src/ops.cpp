@@ -60,7 +60,7 @@ void spmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
               int64_t);
 void rspmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void sddmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-               at::Tensor);
+               at::Tensor, int64_t);
 void csr_to_dense_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void csr_diagonal_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t);
 void csc_spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
@@ -134,7 +134,7 @@ TORCH_LIBRARY(sparse_hip, m) {
   m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "
         "Tensor(a!) C) -> ()");
   m.def("sddmm(Tensor indptr, Tensor indices, Tensor vals, Tensor C, "
-        "Tensor D, Tensor(a!) out) -> ()");
+        "Tensor D, Tensor(a!) out, int col_lo) -> ()");
   m.def("csr_to_dense(Tensor indptr, Tensor indices, Tensor vals, "
         "Tensor(a!) out) -> ()");
   m.def("csr_diagonal(Tensor indptr, Tensor indices, Tensor vals, "

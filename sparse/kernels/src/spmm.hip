@@ -71,7 +71,9 @@ __global__ void rspmm_kernel(const int64_t* __restrict__ indptr,
   }
 }
 
-// out[p] = vals[p] * sum_t Cm[row,t] * D[t, col]
+// out[p] = vals[p] * sum_t Cm[row,t] * D[t, col-col_lo]; D is the gathered
+// COLUMN BLOCK [col_lo, col_lo+n) of the global operand (reference
+// MinMaxImage on proj dim 1, csr.py:1244-1312)
 template <typename T, typename index_t>
 __global__ void sddmm_kernel(const int64_t* __restrict__ indptr,
                              const index_t* __restrict__ indices,
@@ -79,11 +81,11 @@ __global__ void sddmm_kernel(const int64_t* __restrict__ indptr,
                              const T* __restrict__ Cm,  // (m, kd) row-major
                              const T* __restrict__ D,   // (kd, n) row-major
                              T* __restrict__ out, int64_t m, int64_t n,
-                             int64_t kd, int64_t nnz) {
+                             int64_t kd, int64_t nnz, int64_t col_lo) {
   int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (p >= nnz) return;
   int64_t r = ub_i64(indptr, m + 1, p) - 1;
-  int64_t c = (int64_t)indices[p];
+  int64_t c = (int64_t)indices[p] - col_lo;
   T acc = ZeroOf<T>::value();
   for (int64_t t = 0; t < kd; ++t) acc += Cm[r * kd + t] * D[t * n + c];
   out[p] = vals[p] * acc;
@@ -142,7 +144,7 @@ void rspmm_hip(at::Tensor indptr, at::Tensor indices, at::Tensor vals,
 }
 
 void sddmm_hip(at::Tensor indptr, at::Tensor indices, at::Tensor vals,
-               at::Tensor Cm, at::Tensor D, at::Tensor out) {
+               at::Tensor Cm, at::Tensor D, at::Tensor out, int64_t col_lo) {
   int64_t nnz = vals.numel();
   if (nnz == 0) return;
   int64_t m = indptr.numel() - 1;
@@ -153,7 +155,7 @@ void sddmm_hip(at::Tensor indptr, at::Tensor indices, at::Tensor vals,
                          dim3(256), 0, cur_stream(), indptr.data_ptr<int64_t>(),
                          indices.data_ptr<index_t>(), vals.data_ptr<T>(),
                          Cm.data_ptr<T>(), D.data_ptr<T>(), out.data_ptr<T>(),
-                         m, D.size(1), Cm.size(1), nnz);
+                         m, D.size(1), Cm.size(1), nnz, col_lo);
     });
   });
 }

@@ -150,17 +150,20 @@ def rspmm(A_dense: torch.Tensor, B: LocalCSR) -> torch.Tensor:
     return torch.as_tensor(np.ascontiguousarray(r), device=B.device).to(vdt)
 
 
-def sddmm(A: LocalCSR, C: torch.Tensor, D: torch.Tensor) -> torch.Tensor:
-    """out_vals[nz at (i,j)] = A.values[nz] * (C[i,:] @ D[:,j]).
-    Reference: CSR_SDDMM (sddmm.cu:25-85)."""
+def sddmm(A: LocalCSR, C: torch.Tensor, D: torch.Tensor,
+          col_lo: int = 0) -> torch.Tensor:
+    """out_vals[nz at (i,j)] = A.values[nz] * (C[i,:] @ D[:,j - col_lo]).
+    D is the gathered column block [col_lo, col_lo + D.shape[1]) of the
+    global operand.  Reference: CSR_SDDMM (sddmm.cu:25-85) with the
+    MinMaxImage proj-dim-1 D gather (csr.py:1244-1312)."""
     if is_gpu(A.values):
         out = torch.empty_like(A.values)
-        hip().sddmm(A, C.contiguous(), D.contiguous(), out)
+        hip().sddmm(A, C.contiguous(), D.contiguous(), out, col_lo)
         return out
     m = A.to_scipy().tocoo()
     Cn = C.detach().cpu().numpy()
     Dn = D.detach().cpu().numpy()
-    vals = m.data * np.einsum("ij,ji->i", Cn[m.row], Dn[:, m.col])
+    vals = m.data * np.einsum("ij,ji->i", Cn[m.row], Dn[:, m.col - col_lo])
     return torch.as_tensor(vals, device=A.device)
 
 

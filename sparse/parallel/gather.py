@@ -247,6 +247,52 @@ class PreciseGatherPlan:
         return out.reshape(self.hi, *tail) if tail else out
 
 
+class ColBlockGatherPlan:
+    """Gather the COLUMN block D[:, lo:hi) of a dim-0-partitioned 2-D
+    operand — the reference's MinMaxImage on projection dim 1 for the
+    SDDMM D operand (csr.py:1244-1312, sddmm.cu:25-85).  Per-rank wire
+    bytes = own_rows x my_window (each owner slices its rows to each
+    peer's window), not rows x n as a full gather would move."""
+
+    def __init__(self, lo: int, hi: int, dpart: RowPartition, group=None):
+        self.lo = int(lo)
+        self.hi = int(hi)
+        self.dpart = dpart
+        self.group = group
+        ws = comm.world_size(group)
+        if ws == 1:
+            self.windows = [(self.lo, self.hi)]
+            return
+        w = torch.tensor([self.lo, self.hi], dtype=torch.int64)
+        outs = [torch.zeros(2, dtype=torch.int64) for _ in range(ws)]
+        import torch.distributed as dist
+
+        if dist.get_backend(group) == "nccl":
+            from ..runtime import runtime
+
+            w = w.to(runtime().device)
+            outs = [o.to(runtime().device) for o in outs]
+        dist.all_gather(outs, w, group=group)
+        self.windows = [(int(o[0].item()), int(o[1].item())) for o in outs]
+
+    def gather(self, dlocal: torch.Tensor) -> torch.Tensor:
+        """dlocal: my (rows_p, n) slab; returns the full-height column
+        block (rows, hi-lo), contiguous."""
+        ws = comm.world_size(self.group)
+        me = comm.rank(self.group)
+        if ws == 1:
+            return dlocal[:, self.lo: self.hi].contiguous()
+        send = []
+        for p in range(ws):
+            plo, phi = self.windows[p]
+            send.append(dlocal[:, plo:phi].contiguous().reshape(-1))
+        w = self.hi - self.lo
+        rc = [self.dpart.count(p) * w for p in range(ws)]
+        recv = comm.all_to_all_v(send, group=self.group, recv_counts=rc)
+        pieces = [recv[p].reshape(self.dpart.count(p), w) for p in range(ws)]
+        return torch.cat(pieces, dim=0)
+
+
 class ReduceScatterPlan:
     """Inverse of the window gather: each rank holds partial contributions to
     y[lo:hi); owners receive and sum them.
