@@ -41,6 +41,18 @@ for _ in range(args.iters):
 ms_rap = timer.stop() / args.iters
 
 nnz_a, nnz_c, nnz_ac = A.nnz, C.nnz, Ac.nnz  # collectives: all ranks
+
+# per-rank comm volume for ONE timed R@A@P (VERDICT r1 #2 evidence)
+comm.reset_stats()
+Ac = P.T @ (A @ P)
+per_rank = [f"r{comm.rank()}: {comm.stats['a2a_send_bytes']/1e6:.2f} MB"]
+if comm.world_size() > 1:
+    gathered = [None] * comm.world_size()
+    import torch.distributed as dist
+
+    dist.all_gather_object(gathered, per_rank[0])
+    per_rank = gathered
 if comm.rank() == 0:
     print(f"A@A:  {ms_aa:.2f} ms/op  (A nnz={nnz_a}, C nnz={nnz_c})")
     print(f"R@A@P: {ms_rap:.2f} ms/op (Ac nnz={nnz_ac})")
+    print("R@A@P per-rank alltoallv send bytes:", "; ".join(per_rank))

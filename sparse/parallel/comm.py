@@ -110,6 +110,17 @@ def bcast_(t: torch.Tensor, src: int = 0, group=None) -> None:
         tt.copy_(wire.to(tt.device))
 
 
+# wire-byte accounting (per-rank, monotonically increasing): lets
+# benchmarks log per-rank comm volume (VERDICT r1 #2 "per-rank bytes
+# logged").  Counts alltoallv payload bytes actually sent to PEERS.
+stats = {"a2a_send_bytes": 0, "a2a_calls": 0}
+
+
+def reset_stats() -> None:
+    stats["a2a_send_bytes"] = 0
+    stats["a2a_calls"] = 0
+
+
 def all_to_all_v(send: List[torch.Tensor], group=None,
                  recv_counts: Optional[List[int]] = None) -> List[torch.Tensor]:
     """Exchange send[r] -> rank r (1-D tensors); returns recv list indexed by
@@ -147,8 +158,10 @@ def all_to_all_v(send: List[torch.Tensor], group=None,
     # and RCCL (xGMI neighbor exchange); zero-size guards per the
     # reference's NCCL hang workaround, sort.cu:259-263.
     p2p = []
+    stats["a2a_calls"] += 1
     for peer in range(ws):
         if peer != me and wire[peer].numel() > 0:
+            stats["a2a_send_bytes"] += wire[peer].numel() * wire[peer].element_size()
             p2p.append(dist.P2POp(dist.isend, wire[peer], peer, group=group))
         if peer != me and recv[peer].numel() > 0:
             p2p.append(dist.P2POp(dist.irecv, recv[peer], peer, group=group))

@@ -75,6 +75,66 @@ def gather_csr_rows(indptr, indices, values, mypart: RowPartition, lo: int, hi: 
     return out_indptr, idx, vals
 
 
+def gather_csr_rows_precise(indptr, indices, values, mypart: RowPartition,
+                            rows: torch.Tensor, group=None
+                            ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Ship ONLY the requested rows: `rows` is this rank's sorted distinct
+    global row-id list; returns (indptr, indices, values) over those rows
+    in order.  The row analog of PreciseGatherPlan: for a
+    scattered SpGEMM operand per-rank traffic is O(nnz of referenced
+    rows), not O(nnz of the min/max window).  Every rank calls with its
+    own request list (collective)."""
+    ws = comm.world_size(group)
+    me = comm.rank(group)
+    dev = indices.device
+    rows = rows.to(torch.int64)
+    if ws == 1:
+        counts = indptr[rows + 1] - indptr[rows]
+        out_indptr = torch.zeros(rows.numel() + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(counts, 0, out=out_indptr[1:])
+        tot = int(out_indptr[-1].item())
+        pos = (torch.arange(tot, dtype=torch.int64, device=dev)
+               - torch.repeat_interleave(out_indptr[:-1], counts)
+               + torch.repeat_interleave(indptr[rows], counts))
+        return out_indptr, indices[pos], values[pos]
+    # 1. split my request list by owning rank (rows sorted => contiguous)
+    starts = torch.tensor(mypart.starts, dtype=torch.int64, device=rows.device)
+    cuts = torch.searchsorted(rows, starts)
+    reqs = [rows[cuts[p]: cuts[p + 1]].contiguous() for p in range(ws)]
+    # 2. owners receive the row ids each peer wants
+    got = comm.all_to_all_v(reqs, group=group)
+    s0 = mypart.start(me)
+    send_counts, send_idx, send_val = [], [], []
+    for g in got:
+        if g.numel() == 0:
+            send_counts.append(torch.zeros(0, dtype=torch.int64, device=dev))
+            send_idx.append(indices[:0])
+            send_val.append(values[:0])
+            continue
+        r = g.to(dev) - s0
+        c = (indptr[r + 1] - indptr[r]).to(torch.int64)
+        tot = int(c.sum().item())
+        off = torch.zeros(r.numel(), dtype=torch.int64, device=dev)
+        torch.cumsum(c[:-1], 0, out=off[1:])
+        pos = (torch.arange(tot, dtype=torch.int64, device=dev)
+               - torch.repeat_interleave(off, c)
+               + torch.repeat_interleave(indptr[r], c))
+        send_counts.append(c)
+        send_idx.append(indices[pos])
+        send_val.append(values[pos])
+    # 3. ship counts + payloads back (row order within each peer's reply
+    # matches its request order; concatenation over ascending owners
+    # matches the sorted `rows` order)
+    rc = comm.all_to_all_v(send_counts, group=group)
+    ri = comm.all_to_all_v(send_idx, group=group)
+    rv = comm.all_to_all_v(send_val, group=group)
+    counts = torch.cat(rc)
+    assert counts.numel() == rows.numel()
+    out_indptr = torch.zeros(rows.numel() + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=out_indptr[1:])
+    return out_indptr, torch.cat(ri), torch.cat(rv)
+
+
 def shuffle_to_owner(key: torch.Tensor, part: RowPartition, *payload, group=None):
     """Send each element to the rank owning key[t] under part.  Returns the
     concatenated (key, *payload) received, unsorted across sources."""
