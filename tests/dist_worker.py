@@ -139,6 +139,7 @@ def main():
     extra_spgemm_2d_check()
     extra_sddmm_block_gather_check()
     extra_spgemm_precise_rows_check()
+    extra_mmread_parallel_check()
     extra_domain_part_spmv_check()
     extra_banded_overlap_check()
     extra_complex_check()
@@ -389,6 +390,61 @@ def extra_domain_part_spmv_check():
     y = A.dot(x, spmv_domain_part=True)
     assert np.allclose(np.asarray(y), s @ x), "domain-part spmv"
     assert A._csc_cache is not None
+
+
+def extra_mmread_parallel_check():
+    """Byte-range-parallel mmread: each rank parses only
+    its chunk; the union must equal the scipy oracle exactly — including
+    tiny files (more ranks than lines), boundary-straddling lines, no
+    trailing newline, and symmetric expansion."""
+    import tempfile
+
+    import scipy.io as spio
+    import scipy.sparse as spsB
+
+    from sparse import io as _sio
+
+    ws = dist.get_world_size()
+    rank = dist.get_rank()
+    rng = np.random.default_rng(33)
+    path = os.path.join(tempfile.gettempdir(), f"dist_mm_{ws}.mtx")
+    for case, (mat, strip_nl) in enumerate([
+        (spsB.random(40, 31, 0.2, random_state=34), False),
+        (spsB.random(3, 3, 0.4, random_state=35), False),      # tiny file
+        (spsB.random(25, 25, 0.15, random_state=36), True),    # no trailing \n
+        (None, False),                                          # symmetric
+    ]):
+        if rank == 0:
+            if mat is None:
+                b = spsB.random(20, 20, 0.2, random_state=37)
+                mat = b + b.T  # scipy mmwrite emits symmetric format
+            spio.mmwrite(path, mat)
+            if strip_nl:
+                with open(path, "rb+") as fh:
+                    fh.seek(-1, 2)
+                    if fh.read(1) == b"\n":
+                        fh.seek(-1, 2)
+                        fh.truncate()
+        dist.barrier()
+        ours = _sio.mmread(path)
+        ref = spio.mmread(path).tocsr()
+        ref.sort_indices()
+        got = ours.tocsr().to_scipy_sparse_csr()
+        assert got.shape == ref.shape, (case, got.shape, ref.shape)
+        assert np.allclose(got.toarray(), ref.toarray()), f"mmread case {case}"
+        # per-rank chunking really happened: local counts must sum to nnz
+        loc = torch.zeros(ws, dtype=torch.int64)
+        loc[rank] = ours._vals.numel()
+        from sparse.parallel import comm as _cc
+
+        _cc.all_reduce_(loc)
+        assert int(loc.sum()) == ref.nnz, (case, loc.tolist(), ref.nnz)
+        if ws > 1 and ref.nnz >= 4 * ws:
+            assert int(loc.max()) < ref.nnz, f"one rank parsed all (case {case})"
+        dist.barrier()
+    if rank == 0:
+        os.remove(path)
+    _ = rng  # keep deterministic-seed convention
 
 
 def extra_spgemm_precise_rows_check():
