@@ -192,6 +192,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         self._plan_cache = {}
         self._ell_cache = None
         self._dia_cache = None
+        self._bsr_cache = None
         self._csc_cache = None
         self._maxrow_cache = None
 
@@ -286,6 +287,7 @@ class csr_array(CompressedBase, DenseSparseBase):
     def _invalidate_caches(self):
         self._ell_cache = None
         self._dia_cache = None
+        self._bsr_cache = None
         self._csc_cache = None
         self._maxrow_cache = None
         self._nnz_cache = None
@@ -318,6 +320,19 @@ class csr_array(CompressedBase, DenseSparseBase):
             dm = kernels.build_dia(self.local, self.partition.start(comm.rank()))
             self._dia_cache = dm or "no"
         return None if self._dia_cache == "no" else self._dia_cache
+
+    def _bsr(self):
+        """Cached 16x16-block MFMA mirror for multi-vector SpMM
+        (kernels.build_bsr; measured win region in profiles/MFMA_r02.md)."""
+        if not self._values.is_cuda or self._bsr_cache == "no":
+            return None
+        if self._bsr_cache is None:
+            from . import kernels
+
+            kernels.require()
+            bm = kernels.build_bsr(self.local)
+            self._bsr_cache = bm or "no"
+        return None if self._bsr_cache == "no" else self._bsr_cache
 
     def _ell(self):
         """Cached padded-ELL mirror (GPU fast SpMV; kernels.build_ell)."""
@@ -370,6 +385,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         self._values = torch.as_tensor(mine, device=self._values.device).to(self._values.dtype)
         self._ell_cache = None
         self._dia_cache = None
+        self._bsr_cache = None
         self._csc_cache = None
 
     @property
@@ -562,6 +578,34 @@ class csr_array(CompressedBase, DenseSparseBase):
             raise ValueError(f"dimension mismatch {self.shape} @ {B.shape}")
         plan = self._xplan(B.partition)
         vdt = self._out_dtype(B.local.dtype)
+        k = int(B.local.shape[1]) if B.local.dim() == 2 else 1
+        if (self._values.is_cuda and isinstance(plan, WindowGatherPlan)
+                and k >= 16 and vdt == self._values.dtype
+                and vdt in (torch.float64, torch.float32)):
+            from . import kernels
+
+            bm = self._bsr()
+            if kernels.bsr_profitable(bm, k):
+                # MFMA path reads whole 16-row blocks of B: use a
+                # block-aligned window (profiles/MFMA_r02.md win region)
+                lo, hi = self._col_window()
+                lo16 = (lo >> 4) << 4
+                hi16 = min(self.shape[1], ((hi + 15) >> 4) << 4)
+                key = ("x16", B.partition.starts)
+                if key not in self._plan_cache:
+                    self._plan_cache[key] = WindowGatherPlan(
+                        lo16, hi16, B.partition)
+                aplan = self._plan_cache[key]
+                Bw = aplan.gather(B.local.to(vdt)).contiguous()
+                Clocal = torch.empty(
+                    (self.partition.count(comm.rank()), k), dtype=vdt,
+                    device=self._values.device)
+                kernels.bsr_spmm(bm, Bw, Clocal, aplan.lo)
+                if out is not None:
+                    out.local.copy_(Clocal.to(out.local.dtype))
+                    return out
+                return DistArray.from_local(Clocal, self.partition,
+                                            (self.shape[0], k))
         Bw = plan.gather(B.local.to(vdt))
         lc = self.local
         if not isinstance(plan, WindowGatherPlan):

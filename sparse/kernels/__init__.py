@@ -505,6 +505,70 @@ def dia_jacobi(dm: DiaMirror, pieces, xloc, b, dinv, omega, xout,
                      int(rbase), int(rhi))
 
 
+# -- BSR (MFMA) fast path -----------------------------------------------------
+class BsrMirror:
+    """16x16 dense-block mirror of a CSR slab for the MFMA SpMM path
+    (profiles/MFMA_r02.md): bvals holds row-major 16x16 blocks, bcol the
+    GLOBAL block-column ids.  Local block rows cover the slab rows 0..m."""
+
+    __slots__ = ("bptr", "bcol", "bvals", "nbrows", "m", "fill")
+
+    def __init__(self, bptr, bcol, bvals, nbrows, m, fill):
+        self.bptr = bptr
+        self.bcol = bcol
+        self.bvals = bvals
+        self.nbrows = nbrows
+        self.m = m
+        self.fill = fill
+
+
+def build_bsr(A, min_fill=0.05):
+    """Build the 16x16 BSR mirror of a LocalCSR (fp32/fp64 only), or None
+    when the block fill ratio / memory budget make MFMA unprofitable."""
+    m = A.nrows
+    if m == 0 or A.nnz == 0 or A.values.dtype not in (torch.float32,
+                                                      torch.float64):
+        return None
+    dev = A.device
+    counts = A.indptr[1:] - A.indptr[:-1]
+    lrows = torch.repeat_interleave(
+        torch.arange(m, dtype=torch.int64, device=dev), counts)
+    brow = lrows >> 4
+    bcol_nnz = A.indices.long() >> 4
+    nbc = (A.ncols + 15) // 16
+    key = brow * nbc + bcol_nnz
+    ukey = torch.unique(key)
+    nblocks = int(ukey.numel())
+    fill = A.nnz / (256.0 * nblocks)
+    if fill < min_fill:
+        return None
+    need = nblocks * 256 * A.values.element_size()
+    free, _t = torch.cuda.mem_get_info(dev) if A.values.is_cuda else (1 << 62, 0)
+    if need > 0.25 * free:
+        return None
+    nbrows = (m + 15) // 16
+    blk = torch.searchsorted(ukey, key)
+    bvals = torch.zeros(nblocks * 256, dtype=A.values.dtype, device=dev)
+    bvals[blk * 256 + (lrows & 15) * 16 + (A.indices.long() & 15)] = A.values
+    bptr = torch.zeros(nbrows + 1, dtype=torch.int64, device=dev)
+    bptr[1:] = torch.cumsum(
+        torch.bincount(ukey // nbc, minlength=nbrows), dim=0)
+    bcol = (ukey - (ukey // nbc) * nbc).to(torch.int32)
+    return BsrMirror(bptr, bcol, bvals, nbrows, m, fill)
+
+
+def bsr_spmm(bm: BsrMirror, Bw, C, col_lo: int):
+    """C[m, k] = BSR @ Bw (gathered window rows, global cols - col_lo)."""
+    ext().bsr_spmm(bm.bptr, bm.bcol, bm.bvals, Bw, C, int(col_lo))
+
+
+def bsr_profitable(bm: BsrMirror, k: int) -> bool:
+    """Measured win region (profiles/MFMA_r02.md): at k<=32 MFMA wins from
+    ~6% fill; by k=64 the lane-tiled kernel catches up below ~20% fill."""
+    return bm is not None and (k <= 32 and bm.fill >= 0.06
+                               or bm.fill >= 0.2)
+
+
 # -- ELL fast path ------------------------------------------------------------
 class EllMirror:
     """Column-major padded-ELL copy of a row-uniform CSR slab (fast SpMV)."""

@@ -59,6 +59,8 @@ void axpby_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool, bool);
 void spmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
               int64_t);
 void rspmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void bsr_spmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                  int64_t);
 void sddmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, int64_t);
 void csr_to_dense_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor);
@@ -133,6 +135,8 @@ TORCH_LIBRARY(sparse_hip, m) {
         "Tensor(a!) C, int col_lo) -> ()");
   m.def("rspmm(Tensor indptr, Tensor indices, Tensor vals, Tensor A, "
         "Tensor(a!) C) -> ()");
+  m.def("bsr_spmm(Tensor bptr, Tensor bcol, Tensor bvals, Tensor B, "
+        "Tensor(a!) C, int col_lo) -> ()");
   m.def("sddmm(Tensor indptr, Tensor indices, Tensor vals, Tensor C, "
         "Tensor D, Tensor(a!) out, int col_lo) -> ()");
   m.def("csr_to_dense(Tensor indptr, Tensor indices, Tensor vals, "
@@ -177,6 +181,7 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("dia_jacobi", dia_jacobi_hip);
   m.impl("spmm", spmm_hip);
   m.impl("rspmm", rspmm_hip);
+  m.impl("bsr_spmm", bsr_spmm_hip);
   m.impl("sddmm", sddmm_hip);
   m.impl("csr_to_dense", csr_to_dense_hip);
   m.impl("csr_diagonal", csr_diagonal_hip);
