@@ -99,7 +99,8 @@ __global__ void sddmm_kernel(const int64_t* __restrict__ indptr,
 // the lane-tiled kernel on the 5-diag band at k=32, 9-17x on dense-block
 // structure.  Lane maps measured with tools/mfma_spmm_bench.hip --probe:
 //   A[i][kk]: i=lane%16, kk=lane/16;  B[kk][j]: j=lane%16, kk=lane/16;
-//   D[i][j]:  j=lane%16, i=lane/16 + 4*reg.
+//   D[i][j]:  j=lane%16, i = lane/16 + 4*reg (f64) / 4*(lane/16) + reg (f32)
+//   — the two families pack D rows DIFFERENTLY (both probed on gfx950).
 typedef double mfma_d4 __attribute__((ext_vector_type(4)));
 typedef float mfma_f4 __attribute__((ext_vector_type(4)));
 
@@ -143,7 +144,8 @@ __global__ __launch_bounds__(256) void bsr_mfma_spmm_kernel(
   if (j >= k) return;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    int64_t i = brow * 16 + lk + 4 * r;
+    int64_t i = brow * 16 + (std::is_same_v<T, double> ? lk + 4 * r
+                                                       : 4 * lk + r);
     if (i < mrows) C[i * k + j] = acc[r];
   }
 }

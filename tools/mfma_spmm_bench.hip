@@ -107,6 +107,63 @@ __global__ void map_probe(const double* A16x4, const double* B4x16,
   for (int r = 0; r < 4; ++r) out[l * 4 + r] = acc[r];
 }
 
+typedef float f4v __attribute__((ext_vector_type(4)));
+__global__ void map_probe_f32(const float* A16x4, const float* B4x16,
+                              float* out /*64*4*/, int amap, int bmap) {
+  int l = threadIdx.x;
+  int i = amap ? (l >> 4) : (l & 15);
+  int ka = amap ? (l & 15) : (l >> 4);
+  float a = (ka < 4) ? A16x4[i * 4 + ka] : 0.0f;
+  int j = bmap ? (l >> 4) : (l & 15);
+  int kb = bmap ? (l & 15) : (l >> 4);
+  float b = (kb < 4) ? B4x16[kb * 16 + j] : 0.0f;
+  f4v acc = {0.0f, 0.0f, 0.0f, 0.0f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  for (int r = 0; r < 4; ++r) out[l * 4 + r] = acc[r];
+}
+
+void probe_mapping_f32() {
+  std::vector<float> A(64), B(64);
+  for (int i = 0; i < 64; ++i) {
+    A[i] = 0.1f * i + 0.3f;
+    B[i] = 0.05f * i - 1.1f;
+  }
+  float *dA, *dB, *dO;
+  HIP_CHECK(hipMalloc(&dA, 64 * 4));
+  HIP_CHECK(hipMalloc(&dB, 64 * 4));
+  HIP_CHECK(hipMalloc(&dO, 256 * 4));
+  HIP_CHECK(hipMemcpy(dA, A.data(), 64 * 4, hipMemcpyDefault));
+  HIP_CHECK(hipMemcpy(dB, B.data(), 64 * 4, hipMemcpyDefault));
+  float ref[16][16];
+  for (int i = 0; i < 16; ++i)
+    for (int j = 0; j < 16; ++j) {
+      ref[i][j] = 0.0f;
+      for (int k = 0; k < 4; ++k) ref[i][j] += A[i * 4 + k] * B[k * 16 + j];
+    }
+  for (int amap = 0; amap < 2; ++amap)
+    for (int bmap = 0; bmap < 2; ++bmap) {
+      hipLaunchKernelGGL(map_probe_f32, dim3(1), dim3(64), 0, 0, dA, dB, dO,
+                         amap, bmap);
+      HIP_CHECK(hipDeviceSynchronize());
+      std::vector<float> O(256);
+      HIP_CHECK(hipMemcpy(O.data(), dO, 256 * 4, hipMemcpyDefault));
+      for (int dmap = 0; dmap < 2; ++dmap) {
+        double err = 0.0;
+        for (int l = 0; l < 64; ++l)
+          for (int r = 0; r < 4; ++r) {
+            int j = l & 15;
+            int i = dmap ? ((l >> 4) + 4 * r) : (4 * (l >> 4) + r);
+            err = std::max(err, (double)std::fabs(O[l * 4 + r] - ref[i][j]));
+          }
+        printf("f32 probe amap=%d bmap=%d dmap=%d err=%.3e%s\n", amap, bmap,
+               dmap, err, err < 1e-4 ? "  <-- MATCH" : "");
+      }
+    }
+  HIP_CHECK(hipFree(dA));
+  HIP_CHECK(hipFree(dB));
+  HIP_CHECK(hipFree(dO));
+}
+
 void probe_mapping() {
   std::vector<double> A(64), B(64);
   for (int i = 0; i < 64; ++i) {
@@ -369,6 +426,7 @@ int run_case(const char* name, const Csr& A, long k, int iters) {
 int main(int argc, char** argv) {
   if (argc > 1 && std::string(argv[1]) == "--probe") {
     probe_mapping();
+    probe_mapping_f32();
     return 0;
   }
   long m = argc > 1 ? atol(argv[1]) : (1L << 20);
