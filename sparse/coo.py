@@ -215,20 +215,10 @@ class coo_array(CompressedBase):
         me = comm.rank()
         r0 = part.start(me)
         mloc = part.count(me)
-        # local sort by (row, col); sum duplicates (scipy semantics)
-        key = (i - r0) * max(1, self.shape[1]) + j
-        key, order = torch.sort(key)
-        v = v[order]
-        ukey, inv = torch.unique_consecutive(key, return_inverse=True)
-        if ukey.numel() != key.numel():
-            vs = torch.zeros(ukey.numel(), dtype=v.dtype, device=v.device)
-            vs.index_add_(0, inv, v)
-            v = vs
-            key = ukey
-        rows = torch.div(key, max(1, self.shape[1]), rounding_mode="floor")
-        cols = key - rows * max(1, self.shape[1])
-        indptr = ops.coords_to_indptr(rows, mloc)
+        # segmented scatter+sort (GPU kernel) with dup-summing fallback
         idt = index_dtype_for(self.shape)
+        indptr, cols, v = ops.local_coo_to_csr(
+            i - r0, j.to(idt), v, mloc, self.shape[1])
         return csr_array.from_local(indptr, cols.to(idt), v, part, self.shape)
 
     def tocsc(self, copy=False):

@@ -891,17 +891,10 @@ class csr_array(CompressedBase, DenseSparseBase):
         i, j, v = shuffle_to_owner(trows, part, tcols, C_l.values)
         me = comm.rank()
         mloc = part.count(me)
-        key = (i - part.start(me)) * max(1, n) + j
-        key, order = torch.sort(key)
-        v = v[order]
-        rows_l = torch.div(key, max(1, n), rounding_mode="floor")
-        cols_l = key - rows_l * max(1, n)
-        indptr = torch.zeros(mloc + 1, dtype=torch.int64, device=v.device)
-        if rows_l.numel():
-            indptr[1:] = torch.cumsum(
-                torch.bincount(rows_l, minlength=mloc), dim=0)
         idt = torch.int32 if n < 2**31 - 1 else torch.int64
-        return csr_array.from_local(indptr, cols_l.to(idt), v.to(vdt),
+        indptr, cols_l, v = ops.local_coo_to_csr(
+            i - part.start(me), j.to(idt), v.to(vdt), mloc, n)
+        return csr_array.from_local(indptr, cols_l.to(idt), v,
                                     part, (m, n))
 
     def sddmm(self, C, D) -> "csr_array":
@@ -976,18 +969,13 @@ class csr_array(CompressedBase, DenseSparseBase):
                                              row_offset=self.partition.start(comm.rank()))
         cols, rows2, vals = shuffle_to_owner(self._indices.to(torch.int64), cpart,
                                              rows.to(torch.int64), self._values)
-        # local sort by (col, row)
+        # local (col, row) grouping via the segmented scatter+sort kernel
         me = comm.rank()
         c0 = cpart.start(me)
         ncl = cpart.count(me)
-        key = (cols - c0) * max(1, self.shape[0]) + rows2
-        order = torch.argsort(key)
-        cols = cols[order] - c0
-        rows2 = rows2[order]
-        vals = vals[order]
-        colptr = torch.zeros(ncl + 1, dtype=torch.int64, device=vals.device)
-        torch.cumsum(torch.bincount(cols, minlength=ncl), 0, out=colptr[1:])
         idt = index_dtype_for(self.shape)
+        colptr, rows2, vals = ops.local_coo_to_csr(
+            cols - c0, rows2.to(idt), vals, ncl, self.shape[0])
         return csc_array.from_local(colptr, rows2.to(idt), vals, cpart, self.shape)
 
     def todia(self, copy=False):

@@ -89,6 +89,13 @@ def csr_to_dense(A, out):
     ext().csr_to_dense(A.indptr, A.indices, A.values, out)
 
 
+def coo_to_csr(rows, cols, vals, cursor, indptr, out_idx, out_vals, flags):
+    """Segmented COO->CSR: atomic scatter + per-row LDS sort; flags[0]=row
+    overflow, flags[1]=duplicate columns (caller falls back)."""
+    ext().coo_to_csr(rows, cols, vals, cursor, indptr, out_idx, out_vals,
+                     flags)
+
+
 def csr_diagonal(A, out, row_offset: int):
     ext().csr_diagonal(A.indptr, A.indices, A.values, out, int(row_offset))
 

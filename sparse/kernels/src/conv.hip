@@ -145,3 +145,100 @@ void csc_spmm_hip(at::Tensor colptr, at::Tensor rowidx, at::Tensor vals,
     });
   });
 }
+
+// -- segmented COO -> CSR (de-torch the conversion path) --
+// Replaces the global radix sort (rocprim onesweep was ~25% of bench GPU
+// time in profiles/cg_nx16384_nt_r01.md) with: per-row atomic scatter into
+// precomputed indptr slots + a per-row odd-even LDS sort.  Rows longer
+// than COO2CSR_MAXROW or containing duplicate columns raise a flag and
+// the caller falls back to the torch sort path (rare).
+namespace {
+
+constexpr int COO2CSR_MAXROW = 1024;
+
+template <typename T, typename index_t>
+__global__ void coo_scatter_kernel(const int64_t* __restrict__ rows,
+                                   const index_t* __restrict__ cols,
+                                   const T* __restrict__ vals,
+                                   int64_t* __restrict__ cursor,
+                                   index_t* __restrict__ out_idx,
+                                   T* __restrict__ out_vals, int64_t nnz) {
+  int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (e >= nnz) return;
+  int64_t pos = atomicAdd(
+      reinterpret_cast<unsigned long long*>(&cursor[rows[e]]), 1ull);
+  out_idx[pos] = cols[e];
+  out_vals[pos] = vals[e];
+}
+
+// one WAVE per block per row: LDS odd-even transposition sort of
+// (indices, vals); block=64 so early exit never strands a barrier
+template <typename T, typename index_t>
+__global__ __launch_bounds__(64) void row_sort_kernel(
+    const int64_t* __restrict__ indptr, index_t* __restrict__ indices,
+    T* __restrict__ vals, int64_t m, int* __restrict__ flags) {
+  __shared__ index_t sc[COO2CSR_MAXROW];
+  // raw storage: c10::complex has a ctor, which __shared__ forbids
+  __shared__ __align__(16) unsigned char sv_raw[COO2CSR_MAXROW * sizeof(T)];
+  T* sv = reinterpret_cast<T*>(sv_raw);
+  int64_t row = blockIdx.x;
+  if (row >= m) return;
+  const int64_t s = indptr[row];
+  const int len = (int)(indptr[row + 1] - s);
+  if (len <= 1) return;
+  if (len > COO2CSR_MAXROW) {
+    if (threadIdx.x == 0) atomicOr(&flags[0], 1);
+    return;
+  }
+  const int lane = threadIdx.x;
+  for (int i = lane; i < len; i += 64) {
+    sc[i] = indices[s + i];
+    sv[i] = vals[s + i];
+  }
+  __syncthreads();
+  for (int pass = 0; pass < len; ++pass) {
+    const int start = pass & 1;
+    for (int t = lane; 2 * t + start + 1 < len; t += 64) {
+      const int a = 2 * t + start;
+      if (sc[a] > sc[a + 1]) {
+        index_t tc = sc[a]; sc[a] = sc[a + 1]; sc[a + 1] = tc;
+        T tv = sv[a]; sv[a] = sv[a + 1]; sv[a + 1] = tv;
+      }
+    }
+    __syncthreads();
+  }
+  int dup = 0;
+  for (int i = lane; i + 1 < len; i += 64) dup |= (sc[i] == sc[i + 1]);
+  if (dup) atomicOr(&flags[1], 1);
+  for (int i = lane; i < len; i += 64) {
+    indices[s + i] = sc[i];
+    vals[s + i] = sv[i];
+  }
+}
+
+}  // namespace
+
+void coo_to_csr_hip(at::Tensor rows, at::Tensor cols, at::Tensor vals,
+                    at::Tensor cursor, at::Tensor indptr, at::Tensor out_idx,
+                    at::Tensor out_vals, at::Tensor flags) {
+  int64_t nnz = vals.numel();
+  int64_t m = indptr.numel() - 1;
+  if (m == 0) return;
+  DISPATCH_VALUES(vals.scalar_type(), "coo_to_csr", [&] {
+    using T = scalar_t;
+    DISPATCH_INDEX(cols.scalar_type(), "coo_to_csr_idx", [&] {
+      if (nnz) {
+        hipLaunchKernelGGL((coo_scatter_kernel<T, index_t>),
+                           dim3((nnz + 255) / 256), dim3(256), 0, cur_stream(),
+                           rows.data_ptr<int64_t>(), cols.data_ptr<index_t>(),
+                           vals.data_ptr<T>(), cursor.data_ptr<int64_t>(),
+                           out_idx.data_ptr<index_t>(), out_vals.data_ptr<T>(),
+                           nnz);
+      }
+      hipLaunchKernelGGL((row_sort_kernel<T, index_t>), dim3(m), dim3(64), 0,
+                         cur_stream(), indptr.data_ptr<int64_t>(),
+                         out_idx.data_ptr<index_t>(), out_vals.data_ptr<T>(),
+                         m, flags.data_ptr<int>());
+    });
+  });
+}

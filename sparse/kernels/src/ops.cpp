@@ -64,6 +64,8 @@ void bsr_spmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
 void sddmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                at::Tensor, int64_t);
 void csr_to_dense_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void coo_to_csr_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                    at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void csr_diagonal_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t);
 void csc_spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                   int64_t);
@@ -141,6 +143,9 @@ TORCH_LIBRARY(sparse_hip, m) {
         "Tensor D, Tensor(a!) out, int col_lo) -> ()");
   m.def("csr_to_dense(Tensor indptr, Tensor indices, Tensor vals, "
         "Tensor(a!) out) -> ()");
+  m.def("coo_to_csr(Tensor rows, Tensor cols, Tensor vals, Tensor(a!) cursor, "
+        "Tensor indptr, Tensor(b!) out_idx, Tensor(c!) out_vals, "
+        "Tensor(d!) flags) -> ()");
   m.def("csr_diagonal(Tensor indptr, Tensor indices, Tensor vals, "
         "Tensor(a!) out, int row_offset) -> ()");
   m.def("csc_spmv(Tensor colptr, Tensor rowidx, Tensor vals, Tensor x, "
@@ -184,6 +189,7 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("bsr_spmm", bsr_spmm_hip);
   m.impl("sddmm", sddmm_hip);
   m.impl("csr_to_dense", csr_to_dense_hip);
+  m.impl("coo_to_csr", coo_to_csr_hip);
   m.impl("csr_diagonal", csr_diagonal_hip);
   m.impl("csc_spmv", csc_spmv_hip);
   m.impl("csc_spmm", csc_spmm_hip);

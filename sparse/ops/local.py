@@ -313,6 +313,44 @@ def coords_to_indptr(sorted_rows: torch.Tensor, nrows: int, row_offset: int = 0)
     return indptr
 
 
+def local_coo_to_csr(rows_local: torch.Tensor, cols: torch.Tensor,
+                     vals: torch.Tensor, mloc: int, ncols: int):
+    """Unordered local COO triples (0-based rows) -> sorted, deduped local
+    CSR (indptr int64, indices as given, values).  GPU: per-row atomic
+    scatter + per-row LDS sort (VERDICT r1 #9 — replaces the global radix
+    sort that was ~25% of bench GPU time); rows >1024 nnz or duplicate
+    (i,j) pairs fall back to the torch sort path.  CPU: torch sort."""
+    dev = vals.device
+    key_w = max(1, ncols)
+    if is_gpu(vals) and rows_local.numel():
+        counts = torch.bincount(rows_local, minlength=mloc)
+        indptr = torch.zeros(mloc + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(counts, 0, out=indptr[1:])
+        cursor = indptr[:-1].contiguous()
+        out_idx = torch.empty_like(cols)
+        out_vals = torch.empty_like(vals)
+        flags = torch.zeros(2, dtype=torch.int32, device=dev)
+        hip().coo_to_csr(rows_local, cols, vals, cursor, indptr,
+                         out_idx, out_vals, flags)
+        f = flags.cpu()
+        if int(f[0]) == 0 and int(f[1]) == 0:
+            return indptr, out_idx, out_vals
+        # overflow / duplicates: torch fallback (exact former behavior)
+    key = rows_local * key_w + cols.long()
+    key, order = torch.sort(key)
+    v = vals[order]
+    ukey, inv = torch.unique_consecutive(key, return_inverse=True)
+    if ukey.numel() != key.numel():
+        vs = torch.zeros(ukey.numel(), dtype=v.dtype, device=dev)
+        vs.index_add_(0, inv, v)
+        v = vs
+        key = ukey
+    rws = torch.div(key, key_w, rounding_mode="floor")
+    cls = (key - rws * key_w).to(cols.dtype)
+    indptr = coords_to_indptr(rws, mloc)
+    return indptr, cls, v
+
+
 # -- CSC col-split ops --------------------------------------------------------
 def csc_spmv(colptr: torch.Tensor, rowidx: torch.Tensor, values: torch.Tensor,
              x_cols: torch.Tensor, rlo: int, rhi: int) -> torch.Tensor:
