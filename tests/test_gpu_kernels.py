@@ -73,6 +73,47 @@ def test_spmm_gpu(dt, k):
     assert np.allclose(np.asarray(csr_array(s) @ B), s @ B, **tol(dt))
 
 
+def test_segmented_coo_to_csr_gpu():
+    """The scatter + per-row LDS sort conversion kernel:
+    scattered input, duplicate (i,j) pairs (summed via fallback), a
+    >1024-nnz row (overflow fallback), complex dtype, and tocsc."""
+    import scipy.sparse as sps
+
+    from sparse import coo_array, csr_array
+
+    rng = np.random.default_rng(31)
+    n = 400
+    s = sps.random(n, n, 0.05, random_state=32, format="coo")
+    perm = rng.permutation(s.nnz)
+    A = coo_array((s.data[perm], (s.row[perm], s.col[perm])), shape=s.shape)
+    ref = s.tocsr()
+    ref.sort_indices()
+    got = A.tocsr().to_scipy_sparse_csr()
+    assert (got.indptr == ref.indptr).all()
+    assert (got.indices == ref.indices).all()
+    assert np.allclose(got.data, ref.data)
+    # duplicates -> summed (scipy semantics), kernel flags + falls back
+    i2 = np.array([0, 0, 1, 0, 2]); j2 = np.array([3, 3, 1, 3, 2])
+    v2 = np.array([1.0, 2.0, 5.0, 4.0, 6.0])
+    got2 = coo_array((v2, (i2, j2)), shape=(4, 4)).tocsr()
+    ref2 = sps.coo_matrix((v2, (i2, j2)), shape=(4, 4)).tocsr()
+    assert np.allclose(np.asarray(got2.todense()), ref2.toarray())
+    # one huge row (> 1024 nnz) -> overflow fallback
+    jj = rng.permutation(3000)[:2000]
+    big = sps.coo_matrix((rng.random(2000), (np.zeros(2000, int), jj)),
+                         shape=(4, 3000))
+    gotb = coo_array((big.data, (big.row, big.col)), shape=big.shape).tocsr()
+    refb = big.tocsr(); refb.sort_indices()
+    gb = gotb.to_scipy_sparse_csr()
+    assert (gb.indices == refb.indices).all() and np.allclose(gb.data, refb.data)
+    # complex + tocsc
+    sc = sps.random(120, 90, 0.08, random_state=33).astype(np.complex128)
+    sc.data = sc.data + 1j * rng.random(sc.nnz)
+    gc = csr_array(sc.tocsr()).tocsc()
+    refc = sc.tocsc()
+    assert np.allclose(np.asarray(gc.todense()), refc.toarray())
+
+
 @pytest.mark.parametrize("dt", [np.float64, np.float32])
 @pytest.mark.parametrize("k", [16, 32, 40])
 def test_bsr_mfma_spmm_gpu(dt, k):
