@@ -326,7 +326,7 @@ def local_coo_to_csr(rows_local: torch.Tensor, cols: torch.Tensor,
         counts = torch.bincount(rows_local, minlength=mloc)
         indptr = torch.zeros(mloc + 1, dtype=torch.int64, device=dev)
         torch.cumsum(counts, 0, out=indptr[1:])
-        cursor = indptr[:-1].contiguous()
+        cursor = indptr[:-1].clone()  # clone: the scatter mutates cursors
         out_idx = torch.empty_like(cols)
         out_vals = torch.empty_like(vals)
         flags = torch.zeros(2, dtype=torch.int32, device=dev)
