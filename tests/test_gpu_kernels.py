@@ -73,10 +73,12 @@ def test_spmm_gpu(dt, k):
     assert np.allclose(np.asarray(csr_array(s) @ B), s @ B, **tol(dt))
 
 
-def test_segmented_coo_to_csr_gpu():
-    """The scatter + per-row LDS sort conversion kernel:
-    scattered input, duplicate (i,j) pairs (summed via fallback), a
-    >1024-nnz row (overflow fallback), complex dtype, and tocsc."""
+def test_segmented_coo_to_csr_gpu(monkeypatch):
+    """The scatter + per-row LDS sort conversion kernel (VERDICT r1 #9;
+    opt-in after the measured A/B in profiles/CONV_r02.md): scattered
+    input, duplicate (i,j) pairs (summed via fallback), a >1024-nnz row
+    (overflow fallback), complex dtype, and tocsc."""
+    monkeypatch.setenv("SPARSE_SEGMENTED_CONV", "1")
     import scipy.sparse as sps
 
     from sparse import coo_array, csr_array
