@@ -71,32 +71,26 @@ def all_reduce_(t: torch.Tensor, op: str = "sum", group=None, async_op: bool = F
 def all_gather_rows(local: torch.Tensor, counts: Sequence[int], group=None) -> torch.Tensor:
     """All-gather variable-size row slabs (dim 0) into one global tensor.
 
-    Pads to the max slab so plain equal-shape all_gather works on both
-    backends, then slices.  Fine for the near-equal slabs we use.
-    """
+    Exact-size wire: batched p2p of each slab to every peer (all_to_all_v
+    with identical sends) — no pad-to-max inflation, which for skewed
+    slabs (post-balance() reads, .indptr of a skewed matrix) cost up to
+    ws x the bytes in the padded-all_gather formulation."""
     ws = world_size(group)
     if not initialized() or ws == 1:
         return local
-    local_c = local.contiguous()
-    cplx = local_c.is_complex()
-    wire = torch.view_as_real(local_c) if cplx else local_c
-    orig_device = wire.device
-    wire = _to_wire_device(wire, group)
-    maxc = max(counts)
-    if maxc == 0:
+    if max(counts) == 0:
         return local
-    tail = wire.shape[1:]
-    pad = torch.empty((maxc, *tail), dtype=wire.dtype, device=wire.device)
-    if wire.shape[0] > 0:
-        pad[: wire.shape[0]].copy_(wire)
-    outs = [torch.empty_like(pad) for _ in range(ws)]
-    dist.all_gather(outs, pad, group=group)
-    pieces = [outs[r][: counts[r]] for r in range(ws)]
-    out = torch.cat(pieces, dim=0)
-    if out.device != orig_device:
-        out = out.to(orig_device)
-    if cplx:
-        out = torch.view_as_complex(out)
+    local_c = local.contiguous()
+    tail = local_c.shape[1:]
+    k = 1
+    for t in tail:
+        k *= t
+    flat = local_c.reshape(-1)
+    recv = all_to_all_v([flat] * ws, group=group,
+                        recv_counts=[c * k for c in counts])
+    out = torch.cat(recv, dim=0)
+    if tail:
+        out = out.reshape(-1, *tail)
     return out
 
 
