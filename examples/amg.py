@@ -246,10 +246,45 @@ def build_hierarchy(A: csr_array, theta: float, max_coarse: int):
     return levels
 
 
-def vcycle(levels, li, b):
+class _Shim:
+    pass
+
+
+def build_repl_tail(levels, repl_threshold=1 << 16, smooth_iters=2):
+    """Replicated coarse tail at ws>1 (sparse.multigrid; same latency plan
+    as gmg.py): returns (ReplicatedCoarseCycle | None, cut index)."""
+    from sparse.multigrid import ReplicatedCoarseCycle
+
+    ri = next((i for i, l in enumerate(levels)
+               if l["A"].shape[0] <= repl_threshold), len(levels))
+    if comm.world_size() == 1 or ri >= len(levels):
+        return None, len(levels)
+    shims = []
+    for l in levels[ri:]:
+        sh = _Shim()
+        sh.A = l["A"]
+        if "dinv" in l:
+            sh.dinv = l["dinv"]
+            sh.omega = l["omega"]
+        if "R" in l:
+            sh.Rdown = l["R"]
+            sh.Pdown = l["P"]
+        shims.append(sh)
+    return ReplicatedCoarseCycle(shims, levels[-1]["coarse_inv"],
+                                 smooth_iters), ri
+
+
+def vcycle(levels, li, b, repl=None, ri=None):
     from sparse.parallel import comm as _comm
 
     lvl = levels[li]
+    if repl is not None and li == ri:
+        bf = b.gather() if _comm.world_size() > 1 else b.local
+        xf = repl.apply(bf)
+        me = _comm.rank()
+        return darray.DistArray.from_local(
+            xf[b.partition.start(me): b.partition.stop(me)].clone(),
+            b.partition, b.gshape)
     if "coarse_inv" in lvl:
         bg = b.gather() if _comm.world_size() > 1 else b.local
         return darray.asdistarray(lvl["coarse_inv"] @ bg)
@@ -259,7 +294,7 @@ def vcycle(levels, li, b):
     x = A.jacobi_smooth(x, b, dinv, omega)
     r = A.residual(x, b)
     rc = lvl["R"].dot(r)
-    xc = vcycle(levels, li + 1, rc)
+    xc = vcycle(levels, li + 1, rc, repl, ri)
     x += lvl["P"].dot(xc)
     for _ in range(2):
         x = A.jacobi_smooth(x, b, dinv, omega)
@@ -271,6 +306,7 @@ class _GraphedVcycle:
 
     def __init__(self, levels):
         self.levels = levels
+        self.repl, self.ri = build_repl_tail(levels)
         self.graph = None
         self.tried = False
 
@@ -291,11 +327,11 @@ class _GraphedVcycle:
                     side.wait_stream(torch.cuda.current_stream())
                     with torch.cuda.stream(side):
                         for _ in range(2):
-                            vcycle(self.levels, 0, rin)
+                            vcycle(self.levels, 0, rin, self.repl, self.ri)
                     torch.cuda.current_stream().wait_stream(side)
                     g = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(g):
-                        z = vcycle(self.levels, 0, rin)
+                        z = vcycle(self.levels, 0, rin, self.repl, self.ri)
                     self.graph = g
                     self.gout = z.local
                 except Exception as e:
@@ -308,7 +344,7 @@ class _GraphedVcycle:
                 out.local.copy_(self.gout)
                 return out
             return darray.DistArray.from_local(self.gout.clone(), r.partition, r.gshape)
-        z = vcycle(self.levels, 0, r)
+        z = vcycle(self.levels, 0, r, self.repl, self.ri)
         if out is not None:
             out.local.copy_(z.local.to(out.local.dtype))
             return out

@@ -79,99 +79,6 @@ class Level:
         self.omega = (4.0 / 3.0) / estimate_rho_dinv_a(A, self.dinv)
 
 
-class _ReplCycle:
-    """Replicated coarse sub-hierarchy: below a size
-    threshold every rank holds the FULL coarse matrices and runs the
-    identical local V-cycle — zero collectives per coarse level instead of
-    a latency-bound halo exchange each, and on GPU the whole sub-cycle is
-    one hipGraph replay.  This is the MI355X realization of the
-    reference's machine-scoping (gmg.py:212-218), extended from
-    'fewer ranks' to 'all ranks, replicated' because xGMI makes the one
-    transition all-gather cheap while eager launch latency is what killed
-    the reference at scale (BASELINE.md 37.2 -> 1.5 it/s).
-
-    Smoothing in the replicated tail is weighted Jacobi (the rbgs/symgs
-    options apply to the distributed fine levels above it)."""
-
-    def __init__(self, levels, coarse_inv_t, smooth_iters):
-        from sparse import ops
-
-        self.smooth_iters = smooth_iters
-        self.coarse_inv_t = coarse_inv_t
-        dev = coarse_inv_t.device
-        self.on_gpu = dev.type == "cuda"
-        self.lv = []
-        for lvl in levels:
-            ent = {
-                "dinv": torch.as_tensor(np.asarray(lvl.dinv), device=dev),
-                "omega": float(lvl.omega),
-            }
-            A_sp = lvl.A.to_scipy_sparse_csr()  # collective gather
-            if self.on_gpu:
-                ent["A"] = ops.LocalCSR.from_scipy(A_sp, dev)
-            else:
-                ent["A"] = A_sp
-            if hasattr(lvl, "Rdown"):
-                R_sp = lvl.Rdown.to_scipy_sparse_csr()
-                P_sp = lvl.Pdown.to_scipy_sparse_csr()
-                if self.on_gpu:
-                    ent["R"] = ops.LocalCSR.from_scipy(R_sp, dev)
-                    ent["P"] = ops.LocalCSR.from_scipy(P_sp, dev)
-                else:
-                    ent["R"] = R_sp
-                    ent["P"] = P_sp
-            self.lv.append(ent)
-        self._graph = None
-        self._graph_tried = False
-
-    def _spmv(self, M, x):
-        if self.on_gpu:
-            from sparse import ops
-
-            return ops.spmv(M, x)
-        return torch.as_tensor(M @ x.numpy())
-
-    def _vcycle(self, i, b):
-        lv = self.lv[i]
-        if i == len(self.lv) - 1:
-            return self.coarse_inv_t @ b
-        x = b * lv["dinv"] * lv["omega"]
-        for _ in range(self.smooth_iters - 1):
-            x = x + lv["omega"] * lv["dinv"] * (b - self._spmv(lv["A"], x))
-        r = b - self._spmv(lv["A"], x)
-        xc = self._vcycle(i + 1, self._spmv(lv["R"], r))
-        x = x + self._spmv(lv["P"], xc)
-        for _ in range(self.smooth_iters):
-            x = x + lv["omega"] * lv["dinv"] * (b - self._spmv(lv["A"], x))
-        return x
-
-    def apply(self, b_full: torch.Tensor) -> torch.Tensor:
-        if self.on_gpu and not self._graph_tried:
-            self._graph_tried = True
-            if not os.environ.get("SPARSE_NO_HIPGRAPH"):
-                try:
-                    self._gin = b_full.clone()
-                    side = torch.cuda.Stream()
-                    side.wait_stream(torch.cuda.current_stream())
-                    with torch.cuda.stream(side):
-                        for _ in range(2):
-                            self._vcycle(0, self._gin)
-                    torch.cuda.current_stream().wait_stream(side)
-                    g = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(g):
-                        self._gout = self._vcycle(0, self._gin)
-                    self._graph = g
-                except Exception as e:
-                    print(f"[gmg] coarse-tail graph capture unavailable "
-                          f"({e}); eager")
-                    self._graph = None
-        if self._graph is not None:
-            self._gin.copy_(b_full)
-            self._graph.replay()
-            return self._gout
-        return self._vcycle(0, b_full)
-
-
 class GMG:
     """V-cycle preconditioner (2-D bilinear or 3-D trilinear transfers)."""
 
@@ -230,8 +137,10 @@ class GMG:
                     self.ri = i
                     break
             if self.ri < len(self.levels):
-                self.repl = _ReplCycle(self.levels[self.ri:],
-                                       self.coarse_inv_t, smooth_iters)
+                from sparse.multigrid import ReplicatedCoarseCycle
+
+                self.repl = ReplicatedCoarseCycle(
+                    self.levels[self.ri:], self.coarse_inv_t, smooth_iters)
         self._graph = None
         self._graph_tried = False
 
