@@ -82,6 +82,40 @@ def test_gloo_on_gpu_battery():
 
 
 @pytest.mark.gpu
+def test_gloo_on_gpu_bench_ws2():
+    """bench.py at ws=2 with both ranks on one GPU (gloo wire): the exact
+    multi-rank bench code path (DIA interior/boundary split, halo
+    exchange, max-over-ranks timing) with device tensors — guards the
+    first contact with a real 8-GPU node."""
+    if _ngpus() < 1:
+        pytest.skip("needs a GPU")
+    import json
+
+    argv = [sys.executable, os.path.join(ROOT, "bench.py"), "--gpus", "2",
+            "--steps", "5", "--warmup", "2", "--nx", "1024", "--weak",
+            "--no-matched"]
+    env = dict(os.environ)
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29748",
+                "WORLD_SIZE": "2", "SPARSE_DIST_BACKEND": "gloo",
+                "GLOO_SOCKET_IFNAME": env.get("GLOO_SOCKET_IFNAME", "lo")})
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e.update({"RANK": str(r), "LOCAL_RANK": str(r)})
+        procs.append(subprocess.Popen(argv, env=e, stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT, cwd=ROOT))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=600)
+        outs.append(out.decode(errors="replace"))
+        assert p.returncode == 0, outs[-1][-2500:]
+    line = [ln for ln in outs[0].splitlines() if ln.startswith("{")]
+    assert len(line) == 1, outs[0][-800:]
+    rec = json.loads(line[0])
+    assert rec["n_gpus"] == 2 and rec["value"] > 0
+
+
+@pytest.mark.gpu
 @pytest.mark.parametrize("nproc", [2, 8])
 def test_rccl_battery(nproc):
     if _ngpus() < nproc:
