@@ -290,10 +290,16 @@ def main():
     r = b - A.dot(x)  # collective: all ranks participate
     rn = float(r.norm().item())  # all-reduce: all ranks
     if comm.rank() == 0:
+        res = f"residual={rn:.3e}"
+        if args.throughput and not math.isfinite(rn):
+            # forced fixed-iteration runs iterate PAST convergence (GMG-CG
+            # converges in ~5 iters): rz underflows to 0 and beta = 0/0;
+            # timing is unaffected — flag rather than alarm
+            res = "residual=n/a (throughput mode iterated past convergence)"
         print(f"levels={len(mg.levels)} setup={setup_ms:.1f}ms "
               f"solve={solve_ms:.1f}ms iters={it_count[0]} "
               f"({it_count[0] / (solve_ms / 1000.0):.2f} iters/s) "
-              f"residual={rn:.3e} info={info}")
+              f"{res} info={info}")
 
 
 if __name__ == "__main__":
