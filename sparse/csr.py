@@ -373,8 +373,27 @@ class csr_array(CompressedBase, DenseSparseBase):
     def dtype(self):
         return to_numpy_dtype(self._values.dtype)
 
+    _PROP_WARN_NNZ = 50_000_000
+
+    def _warn_global_read(self, what: str) -> None:
+        # O(global) host materialization: cheap correctness for tests and
+        # small matrices, a trap at scale — warn once
+        # per matrix per property past a size threshold
+        if self.nnz >= self._PROP_WARN_NNZ and what not in getattr(
+                self, "_prop_warned", set()):
+            from .utils import perf_warning
+
+            if not hasattr(self, "_prop_warned"):
+                self._prop_warned = set()
+            self._prop_warned.add(what)
+            perf_warning(
+                f".{what} gathers the ENTIRE distributed array to host "
+                f"({self.nnz} nnz); use .local / to_scipy_sparse_csr() on "
+                f"a subset, or keep computation in sparse ops")
+
     @property
     def data(self) -> np.ndarray:
+        self._warn_global_read("data")
         return comm.all_gather_rows(self._values, self._nnz_counts()).cpu().numpy()
 
     @data.setter
@@ -390,6 +409,7 @@ class csr_array(CompressedBase, DenseSparseBase):
 
     @property
     def indices(self) -> np.ndarray:
+        self._warn_global_read("indices")
         return comm.all_gather_rows(self._indices, self._nnz_counts()).cpu().numpy()
 
     @property
@@ -397,6 +417,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         # reassemble the global indptr from per-rank local ones
         if comm.world_size() == 1:
             return self._indptr.cpu().numpy()
+        self._warn_global_read("indptr")
         counts = self._nnz_counts()
         local = (self._indptr[1:]).cpu()
         glob = comm.all_gather_rows(local, self.partition.counts()).numpy().astype(np.int64)

@@ -42,21 +42,25 @@ def test_distributed_battery(nproc):
     assert any("DIST_ALL_OK" in o for o in outs), outs[0][-2000:]
 
 
-@pytest.mark.parametrize("extra", [[], ["--weak"]])
-def test_bench_multirank(extra):
-    """bench.py --gpus 2 must be a tested code path:
-    run it at ws=2 on gloo/CPU and require the one-line JSON contract."""
+@pytest.mark.parametrize("nproc,extra", [(2, []), (2, ["--weak"]), (8, [])])
+def test_bench_multirank(nproc, extra):
+    """bench.py --gpus N must be a tested code path:
+    run it at ws=2 and ws=8 (the driver's SCALE shape) on gloo/CPU and
+    require the one-line JSON contract."""
     import json
 
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     env = dict(os.environ)
-    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(29810 + len(extra)),
+    env.update({"MASTER_ADDR": "127.0.0.1",
+                "MASTER_PORT": str(29810 + nproc + len(extra)),
                 "GLOO_SOCKET_IFNAME": env.get("GLOO_SOCKET_IFNAME", "lo"),
-                "WORLD_SIZE": "2"})
-    argv = [sys.executable, os.path.join(root, "bench.py"), "--gpus", "2",
-            "--steps", "4", "--warmup", "1", "--nx", "128", *extra]
+                "WORLD_SIZE": str(nproc)})
+    nx = "128" if nproc == 2 else "64"
+    argv = [sys.executable, os.path.join(root, "bench.py"),
+            "--gpus", str(nproc), "--steps", "4", "--warmup", "1",
+            "--nx", nx, *extra]
     procs = []
-    for r in range(2):
+    for r in range(nproc):
         e = dict(env)
         e.update({"RANK": str(r), "LOCAL_RANK": str(r)})
         procs.append(subprocess.Popen(argv, env=e, stdout=subprocess.PIPE,
@@ -69,12 +73,12 @@ def test_bench_multirank(extra):
     line = [ln for ln in outs[0].splitlines() if ln.startswith("{")]
     assert len(line) == 1, outs[0][-800:]
     rec = json.loads(line[0])
-    assert rec["metric"] == "cg_iters_per_sec" and rec["n_gpus"] == 2
+    assert rec["metric"] == "cg_iters_per_sec" and rec["n_gpus"] == nproc
     assert rec["scaling"] == ("weak" if extra else "strong")
-    if extra:  # sqrt(2)-scaled grid edge
-        assert rec["config"]["nx"] == int(round(128 * 2 ** 0.5))
-    # rank 1 stays silent (one JSON line per job)
-    assert not any(ln.startswith("{") for ln in outs[1].splitlines())
+    if extra:  # sqrt(ws)-scaled grid edge
+        assert rec["config"]["nx"] == int(round(int(nx) * nproc ** 0.5))
+    # other ranks stay silent (one JSON line per job)
+    assert not any(ln.startswith("{") for o in outs[1:] for ln in o.splitlines())
 
 
 def test_examples_multirank():
