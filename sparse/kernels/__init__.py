@@ -89,6 +89,12 @@ def csr_to_dense(A, out):
     ext().csr_to_dense(A.indptr, A.indices, A.values, out)
 
 
+def dense_to_csr(D, indptr_or_counts, indices, vals, fill: bool):
+    """Two-phase dense->CSR: fill=False counts nnz/row, fill=True emits
+    ordered (indices, vals) via ballot compaction (dense_to_csr.cu parity)."""
+    ext().dense_to_csr(D, indptr_or_counts, indices, vals, bool(fill))
+
+
 def coo_to_csr(rows, cols, vals, cursor, indptr, out_idx, out_vals, flags):
     """Segmented COO->CSR: atomic scatter + per-row LDS sort; flags[0]=row
     overflow, flags[1]=duplicate columns (caller falls back)."""

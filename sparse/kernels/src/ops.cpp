@@ -66,6 +66,7 @@ void sddmm_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
 void csr_to_dense_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor);
 void coo_to_csr_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, at::Tensor, at::Tensor, at::Tensor);
+void dense_to_csr_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool);
 void csr_diagonal_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, int64_t);
 void csc_spmv_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                   int64_t);
@@ -146,6 +147,8 @@ TORCH_LIBRARY(sparse_hip, m) {
   m.def("coo_to_csr(Tensor rows, Tensor cols, Tensor vals, Tensor(a!) cursor, "
         "Tensor indptr, Tensor(b!) out_idx, Tensor(c!) out_vals, "
         "Tensor(d!) flags) -> ()");
+  m.def("dense_to_csr(Tensor D, Tensor(a!) indptr_or_counts, "
+        "Tensor(b!) indices, Tensor(c!) vals, bool fill) -> ()");
   m.def("csr_diagonal(Tensor indptr, Tensor indices, Tensor vals, "
         "Tensor(a!) out, int row_offset) -> ()");
   m.def("csc_spmv(Tensor colptr, Tensor rowidx, Tensor vals, Tensor x, "
@@ -190,6 +193,7 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("sddmm", sddmm_hip);
   m.impl("csr_to_dense", csr_to_dense_hip);
   m.impl("coo_to_csr", coo_to_csr_hip);
+  m.impl("dense_to_csr", dense_to_csr_hip);
   m.impl("csr_diagonal", csr_diagonal_hip);
   m.impl("csc_spmv", csc_spmv_hip);
   m.impl("csc_spmm", csc_spmm_hip);

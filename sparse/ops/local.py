@@ -241,16 +241,29 @@ def csr_to_dense(A: LocalCSR) -> torch.Tensor:
 
 def dense_to_csr(D: torch.Tensor, ncols: Optional[int] = None) -> LocalCSR:
     """Reference: DENSE_TO_CSR_NNZ / DENSE_TO_CSR (dense_to_csr.cu).
-    Torch-vectorized on both devices (mask + nonzero are HIP kernels under
-    torch; a dedicated kernel buys nothing — conversion is not hot)."""
+    GPU: two-phase HIP kernel (wave per row, ballot-compaction — ordered,
+    atomic-free).  CPU: torch mask + nonzero."""
     ncols = D.shape[1] if ncols is None else ncols
     D = D.contiguous()
+    idt = index_dtype_for((D.shape[0], ncols))
+    if is_gpu(D) and D.shape[0] > 0:
+        k = hip()
+        counts = torch.empty(D.shape[0], dtype=torch.int64, device=D.device)
+        empty_i = torch.empty(0, dtype=idt, device=D.device)
+        empty_v = torch.empty(0, dtype=D.dtype, device=D.device)
+        k.dense_to_csr(D, counts, empty_i, empty_v, False)
+        indptr = torch.zeros(D.shape[0] + 1, dtype=torch.int64, device=D.device)
+        torch.cumsum(counts, 0, out=indptr[1:])
+        nnz = int(indptr[-1].item())
+        indices = torch.empty(nnz, dtype=idt, device=D.device)
+        vals = torch.empty(nnz, dtype=D.dtype, device=D.device)
+        k.dense_to_csr(D, indptr, indices, vals, True)
+        return LocalCSR(indptr, indices, vals, D.shape[0], ncols)
     mask = D != 0
     nnz_per_row = mask.sum(dim=1)
     indptr = torch.zeros(D.shape[0] + 1, dtype=torch.int64, device=D.device)
     torch.cumsum(nnz_per_row, 0, out=indptr[1:])
     idx = mask.nonzero(as_tuple=False)
-    idt = index_dtype_for((D.shape[0], ncols))
     return LocalCSR(indptr, idx[:, 1].to(idt), D[mask], D.shape[0], ncols)
 
 

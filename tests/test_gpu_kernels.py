@@ -73,6 +73,37 @@ def test_spmm_gpu(dt, k):
     assert np.allclose(np.asarray(csr_array(s) @ B), s @ B, **tol(dt))
 
 
+@pytest.mark.parametrize("dt", types)
+def test_dense_to_csr_kernel_gpu(dt):
+    """Two-phase ballot-compaction dense->CSR kernel (reference
+    dense_to_csr.cu parity, VERDICT r1 partial closed): ordered columns,
+    all dtypes, wide rows (>64 cols), empty rows, and the csr_array
+    constructor path."""
+    import scipy.sparse as sps
+
+    from sparse import csr_array
+    from sparse.ops import local as L
+
+    rng = np.random.default_rng(41)
+    d = rng.random((37, 201))
+    d[d < 0.8] = 0.0
+    d[5, :] = 0.0  # empty row
+    d[6, :] = 1.0  # dense row
+    if np.dtype(dt).kind == "c":
+        d = d + 1j * (d != 0)
+    d = d.astype(dt)
+    dd = torch.as_tensor(d, device="cuda")
+    lc = L.dense_to_csr(dd)
+    ref = sps.csr_matrix(d)
+    ref.sort_indices()
+    assert (lc.indptr.cpu().numpy() == ref.indptr).all()
+    assert (lc.indices.cpu().numpy() == ref.indices).all()
+    assert np.allclose(lc.values.cpu().numpy(), ref.data)
+    # constructor end-to-end
+    A = csr_array(d)
+    assert np.allclose(np.asarray(A.todense()), d)
+
+
 def test_segmented_coo_to_csr_gpu(monkeypatch):
     """The scatter + per-row LDS sort conversion kernel (VERDICT r1 #9;
     opt-in after the measured A/B in profiles/CONV_r02.md): scattered
