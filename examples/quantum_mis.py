@@ -20,6 +20,11 @@ parser = argparse.ArgumentParser()
 parser.add_argument("-l", type=int, default=3, help="lattice edge (l x l grid graph)")
 parser.add_argument("-T", type=float, default=4.0, help="anneal time")
 parser.add_argument("-rtol", type=float, default=1e-6)
+parser.add_argument("-kmax", type=int, default=None,
+                    help="truncate the Hilbert space at this excitation "
+                         "level (sets of size <= kmax) — the reference "
+                         "rydberg benchmark's space-fraction knob; "
+                         "REQUIRED for large lattices (l >= 6)")
 args, _ = parser.parse_known_args()
 _, timer, npx, sparse_mod, linalg, use_sparse = parse_common_args()
 
@@ -31,8 +36,9 @@ from sparse.parallel import comm
 
 def main():
     G = nx.convert_node_labels_to_integers(nx.grid_2d_graph(args.l, args.l))
-    ip = quantum.independence_polynomial(G)
-    drv = quantum.HamiltonianDriver(graph=G, energies=(1,), dtype=np.complex128)
+    ip = quantum.independence_polynomial(G, kmax=args.kmax)
+    drv = quantum.HamiltonianDriver(graph=G, energies=(1,), dtype=np.complex128,
+                                    kmax=args.kmax)
     mis = quantum.HamiltonianMIS(graph=G, poly=ip, dtype=np.complex128)
     n = mis.nstates
     if comm.rank() == 0:

@@ -18,7 +18,7 @@ flip(repeat(levels, poly))).
 """
 from __future__ import annotations
 
-from typing import List
+from typing import List, Optional
 
 import numpy as np
 
@@ -172,17 +172,40 @@ def sets_to_sizes(queues, graph) -> np.ndarray:
     return np.array([bin(int(q)).count("1") for q in queues], dtype=np.int64)
 
 
-def independence_polynomial(graph) -> List[int]:
+# hard ceiling on enumerated states: grid-graph counts grow like 1.5^n —
+# an unbounded l=9 (81-node) enumeration is ~1e14 sets and takes the HOST
+# down with it.  Raise a clear error instead (override via kmax or
+# max_states).
+MAX_ENUM_STATES = 1 << 27
+
+
+def _check_enum_budget(total: int, k: int, max_states: int) -> None:
+    if total > max_states:
+        raise ValueError(
+            f"independence-set enumeration exceeded {max_states} states at "
+            f"size k={k} ({total} so far); this graph's Hilbert space is "
+            f"intractable to enumerate fully. Truncate with kmax= (the "
+            f"excitation-level cutoff; quantum_mis.py -kmax) or raise "
+            f"max_states explicitly.")
+
+
+def independence_polynomial(graph, kmax: Optional[int] = None,
+                            max_states: int = MAX_ENUM_STATES) -> List[int]:
     """ip[k] = number of independence sets of size k (reference
-    quantum.py:447-460)."""
+    quantum.py:447-460).  kmax truncates enumeration at that set size
+    (the Hilbert-space-fraction knob of the reference's external rydberg
+    benchmark)."""
     ip = [1]
     sets, nbrs = None, None
-    for k in range(1, graph.number_of_nodes() + 1):
+    hi = graph.number_of_nodes() if kmax is None else min(
+        kmax, graph.number_of_nodes())
+    for k in range(1, hi + 1):
         sets, nbrs = enumerate_independent_sets(graph, k, prevk_sets=sets,
                                                 prevk_queues=nbrs)
         if len(sets) == 0:
             break
         ip.append(len(sets))
+        _check_enum_budget(sum(ip), k, max_states)
         if isinstance(nbrs, np.ndarray):
             if not nbrs.any():
                 break
@@ -204,19 +227,23 @@ class HamiltonianDriver:
     every T = S \\ {v} (one fewer excitation); symmetric, entries
     energies[0] (reference quantum.py:27-300)."""
 
-    def __init__(self, energies: tuple = (1,), graph=None, dtype=np.complex64):
+    def __init__(self, energies: tuple = (1,), graph=None, dtype=np.complex64,
+                 kmax: Optional[int] = None,
+                 max_states: int = MAX_ENUM_STATES):
         self.energies = energies
         n = graph.number_of_nodes()
         self.ip = [1]
         all_sets: List[List[int]] = [[0]]
         sets, nbrs = None, None
-        for k in range(1, n + 1):
+        hi = n if kmax is None else min(kmax, n)
+        for k in range(1, hi + 1):
             sets, nbrs = enumerate_independent_sets(graph, k, prevk_sets=sets,
                                                     prevk_queues=nbrs)
             if len(sets) == 0:
                 break
             self.ip.append(len(sets))
             all_sets.append(sets)
+            _check_enum_budget(sum(self.ip), k, max_states)
             if isinstance(nbrs, np.ndarray):
                 if not nbrs.any():
                     break

@@ -140,3 +140,24 @@ def test_wide_graph_polynomial():
     assert len(ip) == 6
     for k in range(6):
         assert ip[k] == math.comb(5, k) * 14 ** k, k
+
+
+def test_enum_budget_guard_and_kmax():
+    """Unbounded enumeration of a large lattice must raise (a 9x9 grid has
+    ~1e14 independence sets — it took a GPU HOST down before this guard);
+    kmax truncation keeps it tractable and consistent."""
+    import networkx as nx
+
+    from sparse import quantum
+
+    G = nx.convert_node_labels_to_integers(nx.grid_2d_graph(7, 7))
+    with pytest.raises(ValueError, match="exceeded"):
+        quantum.independence_polynomial(G, max_states=50_000)
+    ip = quantum.independence_polynomial(G, kmax=3)
+    assert len(ip) == 4 and ip[0] == 1 and ip[1] == 49
+    drv = quantum.HamiltonianDriver(graph=G, kmax=3, dtype=np.complex128)
+    assert drv.nstates == sum(ip)
+    H = drv.hamiltonian
+    assert H.shape == (drv.nstates, drv.nstates)
+    # driver couples size-k sets to size-(k-1) subsets: nnz = 2 * sum k*ip[k]
+    assert H.nnz == 2 * sum(k * c for k, c in enumerate(ip))
