@@ -10,8 +10,10 @@ import sys
 import pytest
 
 
-@pytest.mark.parametrize("nproc", [2, 4])
+@pytest.mark.parametrize("nproc", [2, 3, 4])
 def test_distributed_battery(nproc):
+    """ws=3 covers odd world sizes: uneven slabs, 3x1 2-D SpGEMM grids,
+    asymmetric alltoallv counts."""
     env = dict(os.environ)
     env.update({
         "MASTER_ADDR": "127.0.0.1",
