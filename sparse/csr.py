@@ -602,7 +602,8 @@ class csr_array(CompressedBase, DenseSparseBase):
         k = int(B.local.shape[1]) if B.local.dim() == 2 else 1
         if (self._values.is_cuda and isinstance(plan, WindowGatherPlan)
                 and k >= 16 and vdt == self._values.dtype
-                and vdt in (torch.float64, torch.float32)):
+                and vdt in (torch.float64, torch.float32)
+                and not os.environ.get("SPARSE_NO_BSR")):
             from . import kernels
 
             bm = self._bsr()
