@@ -224,7 +224,11 @@ def main() -> None:
     del x, y
 
     # -- headline CG loop ----------------------------------------------------
-    allow_graph = on_gpu and ws == 1
+    # ws>1 capture (RCCL collectives inside the hipGraph) is opt-in until
+    # validated on a multi-GPU box: SPARSE_WS_HIPGRAPH=1; a failed capture
+    # falls back to eager inside make_cg_stepper
+    allow_graph = on_gpu and (
+        ws == 1 or os.environ.get("SPARSE_WS_HIPGRAPH") == "1")
     cg_step, _state = make_cg_stepper(A, b, n, dtype, on_gpu, allow_graph)
     elapsed = time_steps(cg_step, args.steps, args.warmup, sync)
     iters_per_sec = args.steps / elapsed
