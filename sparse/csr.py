@@ -308,6 +308,21 @@ class csr_array(CompressedBase, DenseSparseBase):
         b = max(a, b // 2 * 2)
         return a, b
 
+    def _ell_interior(self, ell, plan):
+        """Even row bounds [a, b) of the longest contiguous interior run —
+        rows whose whole column window lies in the own x piece (the ELL
+        analog of _dia_interior; (0, 0) disables the overlap split)."""
+        me = comm.rank()
+        key = ("ell_int", plan.lo, plan.hi, plan.xpart.starts)
+        if key not in self._plan_cache:
+            from . import kernels
+
+            xs, xe = plan.xpart.start(me), plan.xpart.stop(me)
+            own_a = max(plan.lo, xs)
+            own_b = max(own_a, min(plan.hi, xe))
+            self._plan_cache[key] = kernels.ell_interior(ell, own_a, own_b)
+        return self._plan_cache[key]
+
     def _dia(self):
         """Cached diagonal mirror (GPU banded fast SpMV; kernels.build_dia —
         values only, no index stream)."""
@@ -572,10 +587,21 @@ class csr_array(CompressedBase, DenseSparseBase):
         elif ell is not None:
             from . import kernels
 
-            pieces = plan.gather_halos(x.local.to(vdt))
             ylocal = torch.empty(self.partition.count(comm.rank()), dtype=vdt,
                                  device=self._values.device)
-            kernels.ell_spmv(ell, pieces, ylocal, plan.lo)
+            a = b = 0
+            if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
+                a, b = self._ell_interior(ell, plan)
+            if b > a:
+                h = plan.gather_halos_begin(x.local.to(vdt))
+                kernels.ell_spmv(ell, plan.handle_pieces(h), ylocal,
+                                 plan.lo, a, b)
+                pieces = plan.gather_halos_end(h)
+                kernels.ell_spmv(ell, pieces, ylocal, plan.lo, 0, a)
+                kernels.ell_spmv(ell, pieces, ylocal, plan.lo, b, -1)
+            else:
+                pieces = plan.gather_halos(x.local.to(vdt))
+                kernels.ell_spmv(ell, pieces, ylocal, plan.lo)
         else:
             xw = plan.gather(x.local.to(vdt))
             self._max_row_nnz()
@@ -760,8 +786,22 @@ class csr_array(CompressedBase, DenseSparseBase):
                 dot = kernels.dia_spmv_dot(dm, pieces, q.local, p.local,
                                            plan.lo, ws_)
         elif ell is not None:
-            pieces = plan.gather_halos(p.local)
-            dot = kernels.ell_spmv_dot(ell, pieces, q.local, p.local, plan.lo)
+            a = b = 0
+            if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
+                a, b = self._ell_interior(ell, plan)
+            if b > a:
+                h = plan.gather_halos_begin(p.local)
+                dot = kernels.ell_spmv_dot(ell, plan.handle_pieces(h),
+                                           q.local, p.local, plan.lo, a, b)
+                pieces = plan.gather_halos_end(h)
+                dot = dot + kernels.ell_spmv_dot(ell, pieces, q.local,
+                                                 p.local, plan.lo, 0, a)
+                dot = dot + kernels.ell_spmv_dot(ell, pieces, q.local,
+                                                 p.local, plan.lo, b, -1)
+            else:
+                pieces = plan.gather_halos(p.local)
+                dot = kernels.ell_spmv_dot(ell, pieces, q.local, p.local,
+                                           plan.lo)
         else:
             xw = plan.gather(p.local)
             dot = torch.zeros((), dtype=self._values.dtype, device=self._values.device)

@@ -584,15 +584,19 @@ def bsr_profitable(bm: BsrMirror, k: int) -> bool:
 
 # -- ELL fast path ------------------------------------------------------------
 class EllMirror:
-    """Column-major padded-ELL copy of a row-uniform CSR slab (fast SpMV)."""
+    """Column-major padded-ELL copy of a row-uniform CSR slab (fast SpMV).
+    rmin/rmax are per-row global column windows (for the ws>1
+    interior/boundary overlap split)."""
 
-    __slots__ = ("eidx", "evals", "W", "m")
+    __slots__ = ("eidx", "evals", "W", "m", "rmin", "rmax")
 
-    def __init__(self, eidx, evals, W, m):
+    def __init__(self, eidx, evals, W, m, rmin=None, rmax=None):
         self.eidx = eidx
         self.evals = evals
         self.W = W
         self.m = m
+        self.rmin = rmin
+        self.rmax = rmax
 
 
 def build_ell(A):
@@ -617,13 +621,45 @@ def build_ell(A):
     evals = torch.empty(W * mp, dtype=A.values.dtype, device=A.device)
     pad_idx = int(A.indices[0].item())
     ext().build_ell(A.indptr, A.indices, A.values, eidx, evals, W, pad_idx)
-    return EllMirror(eidx, evals, W, m)
+    # per-row global column windows (empty rows: [n, -1) -> never interior)
+    rows = torch.repeat_interleave(
+        torch.arange(m, dtype=torch.int64, device=A.device), counts)
+    idx = A.indices.long()
+    rmin = torch.full((m,), A.ncols, dtype=torch.int64, device=A.device)
+    rmax = torch.full((m,), -1, dtype=torch.int64, device=A.device)
+    rmin.scatter_reduce_(0, rows, idx, "amin")
+    rmax.scatter_reduce_(0, rows, idx, "amax")
+    return EllMirror(eidx, evals, W, m, rmin, rmax)
 
 
-def ell_spmv(ell: EllMirror, pieces, y, col_lo: int):
+def ell_interior(ell: EllMirror, own_a: int, own_b: int):
+    """Even row bounds [a, b) of the longest contiguous run of rows whose
+    whole column window lies in the own x piece [own_a, own_b) — the ELL
+    analog of the DIA interior split.  Returns (0, 0) when the interior
+    is too small to be worth a split kernel pair."""
+    mask = (ell.rmin >= own_a) & (ell.rmax < own_b)
+    pad = torch.zeros(1, dtype=torch.int8, device=mask.device)
+    d = torch.diff(torch.cat([pad, mask.to(torch.int8), pad]))
+    starts = (d == 1).nonzero(as_tuple=False).flatten()
+    ends = (d == -1).nonzero(as_tuple=False).flatten()
+    if starts.numel() == 0:
+        return 0, 0
+    lens = ends - starts
+    j = int(torch.argmax(lens).item())
+    a, b = int(starts[j].item()), int(ends[j].item())
+    a = (a + 1) // 2 * 2
+    b = max(a, b // 2 * 2)
+    if b - a < max(1024, ell.m // 4):
+        return 0, 0
+    return a, b
+
+
+def ell_spmv(ell: EllMirror, pieces, y, col_lo: int,
+             rbase: int = 0, rhi: int = -1):
     hlo, own, hhi = pieces
     ext().ell_spmv(ell.eidx, ell.evals, hlo.contiguous(), own.contiguous(),
-                   hhi.contiguous(), y, ell.W, ell.m, int(col_lo))
+                   hhi.contiguous(), y, ell.W, ell.m, int(col_lo),
+                   int(rbase), int(rhi))
 
 
 def ell_jacobi(ell: EllMirror, pieces, xloc, b, dinv, omega, xout, col_lo: int):
@@ -634,12 +670,16 @@ def ell_jacobi(ell: EllMirror, pieces, xloc, b, dinv, omega, xout, col_lo: int):
                      int(col_lo), float(omega))
 
 
-def ell_spmv_dot(ell: EllMirror, pieces, y, p, col_lo: int):
+def ell_spmv_dot(ell: EllMirror, pieces, y, p, col_lo: int,
+                 rbase: int = 0, rhi: int = -1):
     hlo, own, hhi = pieces
     mp = ell.evals.numel() // ell.W
-    nblocks = (mp // 2 + 255) // 256
-    partial = torch.empty(nblocks, dtype=ell.evals.dtype, device=ell.evals.device)
+    hi = mp if rhi < 0 else rhi
+    if hi <= rbase:
+        return torch.zeros((), dtype=ell.evals.dtype, device=ell.evals.device)
+    nblocks = ((hi - rbase) // 2 + 255) // 256 + 1
+    partial = torch.zeros(nblocks, dtype=ell.evals.dtype, device=ell.evals.device)
     ext().ell_spmv_dot(ell.eidx, ell.evals, hlo.contiguous(), own.contiguous(),
                        hhi.contiguous(), y, p, partial, ell.W, ell.m,
-                       int(col_lo))
+                       int(col_lo), int(rbase), int(hi))
     return partial.sum()

@@ -18,10 +18,11 @@ void axpby_norm2_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool,
 void build_ell_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                    int64_t, int64_t);
 void ell_spmv_plain_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                        at::Tensor, at::Tensor, int64_t, int64_t, int64_t);
+                        at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
+                        int64_t, int64_t);
 void ell_spmv_dot_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                       at::Tensor, at::Tensor, at::Tensor, at::Tensor,
-                      int64_t, int64_t, int64_t);
+                      int64_t, int64_t, int64_t, int64_t, int64_t);
 void ell_jacobi_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                     at::Tensor, int64_t, int64_t, int64_t, double);
@@ -106,10 +107,11 @@ TORCH_LIBRARY(sparse_hip, m) {
   m.def("build_ell(Tensor indptr, Tensor indices, Tensor values, "
         "Tensor(a!) eidx, Tensor(b!) evals, int W, int pad_idx) -> ()");
   m.def("ell_spmv(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
-        "Tensor hhi, Tensor(a!) y, int W, int m, int col_lo) -> ()");
+        "Tensor hhi, Tensor(a!) y, int W, int m, int col_lo, "
+        "int rbase, int rhi) -> ()");
   m.def("ell_spmv_dot(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor(a!) y, Tensor pvec, Tensor(b!) dot_partial, "
-        "int W, int m, int col_lo) -> ()");
+        "int W, int m, int col_lo, int rbase, int rhi) -> ()");
   m.def("ell_jacobi(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor xloc, Tensor b, Tensor dinv, Tensor(a!) xout, "
         "int W, int m, int col_lo, float omega) -> ()");

@@ -344,13 +344,17 @@ __global__ __launch_bounds__(BLK) void ell_spmv_kernel(
     const T* __restrict__ hlo, const T* __restrict__ own,
     const T* __restrict__ hhi, T* __restrict__ y,
     const T* __restrict__ pvec, T* __restrict__ dot_partial,
-    int64_t m, int64_t mp, int W, int64_t col_lo, int64_t nlo, int64_t nown) {
+    int64_t m, int64_t mp, int W, int64_t col_lo, int64_t nlo, int64_t nown,
+    int64_t rbase, int64_t rhi) {
+  // [rbase, rhi): row sub-range (rbase even) — the interior/boundary split
+  // that overlaps halo exchange with interior compute at ws>1 (same
+  // contract as dia_spmv_kernel)
   __shared__ __align__(16) char red_raw[BLK * sizeof(T)];
   T* red = reinterpret_cast<T*>(red_raw);
   const int64_t t = (int64_t)blockIdx.x * BLK + threadIdx.x;
-  const int64_t r0 = 2 * t;
+  const int64_t r0 = rbase + 2 * t;
   T a0 = ZeroOf<T>::value(), a1 = ZeroOf<T>::value();
-  if (r0 < mp) {
+  if (r0 < rhi) {
     for (int k = 0; k < W; ++k) {
       const int64_t base = (int64_t)k * mp + r0;
       // adjacent pair: one 2-element vector load per stream
@@ -371,7 +375,7 @@ __global__ __launch_bounds__(BLK) void ell_spmv_kernel(
         a1 += vv.b * xpiece((int64_t)ii.b - col_lo, hlo, nlo, own, nown, hhi);
       }
     }
-    if (r0 + 1 < m) {
+    if (r0 + 1 < min(m, rhi)) {
       struct alignas(2 * sizeof(T) <= 16 ? 2 * sizeof(T) : 16) TP { T a, b; };
       TP out{a0, a1};
       *reinterpret_cast<TP*>(&y[r0]) = out;
@@ -381,8 +385,8 @@ __global__ __launch_bounds__(BLK) void ell_spmv_kernel(
   }
   if (FUSE_DOT) {
     T d = ZeroOf<T>::value();
-    if (r0 < m) d += a0 * pvec[r0];
-    if (r0 + 1 < m) d += a1 * pvec[r0 + 1];
+    if (r0 < min(m, rhi)) d += a0 * pvec[r0];
+    if (r0 + 1 < min(m, rhi)) d += a1 * pvec[r0 + 1];
     red[threadIdx.x] = d;
     __syncthreads();
     for (int w = BLK / 2; w > 0; w >>= 1) {
@@ -416,10 +420,13 @@ void ell_spmv_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
                   at::Tensor own, at::Tensor hhi, at::Tensor y,
                   int64_t W, int64_t m, int64_t col_lo,
                   const c10::optional<at::Tensor>& pvec,
-                  const c10::optional<at::Tensor>& dot_partial) {
+                  const c10::optional<at::Tensor>& dot_partial,
+                  int64_t rbase, int64_t rhi) {
   const int64_t mp = evals.numel() / W;
+  if (rhi < 0) rhi = mp;
+  if (rhi <= rbase) return;
   const bool fuse = pvec.has_value();
-  const int64_t nblocks = (mp / 2 + BLK - 1) / BLK;
+  const int64_t nblocks = ((rhi - rbase) / 2 + BLK - 1) / BLK;
   const int64_t nlo = hlo.numel();
   const int64_t nown = own.numel();
   DISPATCH_VALUES(evals.scalar_type(), "ell_spmv", [&] {
@@ -432,7 +439,8 @@ void ell_spmv_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
         hipLaunchKernelGGL(kern, dim3(nblocks), dim3(BLK), 0, cur_stream(),
                            eidx.data_ptr<index_t>(), evals.data_ptr<T>(),
                            hlo_p, own.data_ptr<T>(), hhi_p, y.data_ptr<T>(),
-                           pv, dp, m, mp, (int)W, col_lo, nlo, nown);
+                           pv, dp, m, mp, (int)W, col_lo, nlo, nown,
+                           rbase, rhi);
       };
       if (fuse && single)
         launch(ell_spmv_kernel<T, index_t, true, true>, pvec->data_ptr<T>(),
@@ -450,16 +458,19 @@ void ell_spmv_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
 
 void ell_spmv_plain_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
                         at::Tensor own, at::Tensor hhi, at::Tensor y,
-                        int64_t W, int64_t m, int64_t col_lo) {
+                        int64_t W, int64_t m, int64_t col_lo,
+                        int64_t rbase, int64_t rhi) {
   ell_spmv_hip(eidx, evals, hlo, own, hhi, y, W, m, col_lo, c10::nullopt,
-               c10::nullopt);
+               c10::nullopt, rbase, rhi);
 }
 
 void ell_spmv_dot_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
                       at::Tensor own, at::Tensor hhi, at::Tensor y,
                       at::Tensor pvec, at::Tensor dot_partial,
-                      int64_t W, int64_t m, int64_t col_lo) {
-  ell_spmv_hip(eidx, evals, hlo, own, hhi, y, W, m, col_lo, pvec, dot_partial);
+                      int64_t W, int64_t m, int64_t col_lo,
+                      int64_t rbase, int64_t rhi) {
+  ell_spmv_hip(eidx, evals, hlo, own, hhi, y, W, m, col_lo, pvec, dot_partial,
+               rbase, rhi);
 }
 
 // ---------------------------------------------------------------------------

@@ -142,6 +142,7 @@ def main():
     extra_mmread_parallel_check()
     extra_domain_part_spmv_check()
     extra_banded_overlap_check()
+    extra_ell_overlap_check()
     extra_complex_check()
     extra_indexing_check()
 
@@ -331,6 +332,39 @@ def extra_complex_check():
     b = rng.random(n) + 1j * rng.random(n)
     xs, info = _lin.bicg(A, b, tol=1e-10, maxiter=400, conv_test_iters=2)
     assert np.allclose(s @ np.asarray(xs), b, atol=1e-5), "dist complex bicg"
+
+
+def extra_ell_overlap_check():
+    """Row-uniform SCATTERED matrix (ELL-eligible, not DIA) at ws>1: on a
+    GPU box this exercises the ELL interior/boundary halo-overlap split
+    (kernels.ell_interior + ranged ell_spmv/ell_spmv_dot); on CPU the same
+    call path runs the CSR fallback."""
+    import scipy.sparse as spsE
+
+    from sparse import csr_array as _csr, darray as _d
+
+    n = 6000
+    rng = np.random.default_rng(66)
+    W = 8
+    # mostly-local columns (so an interior run exists) + some far columns
+    rows = np.repeat(np.arange(n), W)
+    near = (rows + rng.integers(-40, 41, n * W)) % n
+    far = rng.integers(0, n, n * W)
+    use_far = rng.random(n * W) < 0.08
+    cols = np.where(use_far, far, near)
+    s = spsE.csr_matrix((rng.random(n * W), (rows, cols)), shape=(n, n))
+    s.sum_duplicates()
+    A = _csr(s)
+    x = _d.random((n,), seed=67)
+    y = A @ x
+    assert np.allclose(np.asarray(y), s @ np.asarray(x), rtol=1e-10), \
+        "ell overlap spmv"
+    q = _d.zeros((n,))
+    dv = A.spmv_dot(x, q)
+    assert np.allclose(np.asarray(q), s @ np.asarray(x), rtol=1e-10), \
+        "ell overlap spmv_dot q"
+    assert np.isclose(float(dv), float(np.asarray(x) @ (s @ np.asarray(x))),
+                      rtol=1e-8), "ell overlap dot"
 
 
 def extra_banded_overlap_check():

@@ -73,6 +73,53 @@ def test_spmm_gpu(dt, k):
     assert np.allclose(np.asarray(csr_array(s) @ B), s @ B, **tol(dt))
 
 
+def test_ell_ranged_split_gpu():
+    """Ranged ELL SpMV/dot (the ws>1 interior/boundary overlap split):
+    three range launches must reproduce the full kernel exactly, and
+    ell_interior must return a valid even interior run."""
+    import scipy.sparse as sps
+
+    from sparse import csr_array, kernels
+
+    n = 5000
+    rng = np.random.default_rng(51)
+    W = 6
+    rows = np.repeat(np.arange(n), W)
+    cols = (rows + rng.integers(-30, 31, n * W)) % n
+    s = sps.csr_matrix((rng.random(n * W), (rows, cols)), shape=(n, n))
+    s.sum_duplicates()
+    A = csr_array(s)
+    ell = A._ell()
+    assert ell is not None and ell.rmin is not None
+    x = torch.as_tensor(rng.random(n), device="cuda")
+    pieces = (x[:0], x, x[:0])
+    y_full = torch.empty(n, dtype=torch.float64, device="cuda")
+    kernels.ell_spmv(ell, pieces, y_full, 0)
+    # split at an arbitrary even cut pair
+    a, b = 1024, 3072
+    y_split = torch.empty(n, dtype=torch.float64, device="cuda")
+    kernels.ell_spmv(ell, pieces, y_split, 0, a, b)
+    kernels.ell_spmv(ell, pieces, y_split, 0, 0, a)
+    kernels.ell_spmv(ell, pieces, y_split, 0, b, -1)
+    assert torch.equal(y_full, y_split)
+    p = torch.as_tensor(rng.random(n), device="cuda")
+    q1 = torch.empty_like(y_full)
+    d_full = kernels.ell_spmv_dot(ell, pieces, q1, p, 0)
+    q2 = torch.empty_like(y_full)
+    d_split = (kernels.ell_spmv_dot(ell, pieces, q2, p, 0, a, b)
+               + kernels.ell_spmv_dot(ell, pieces, q2, p, 0, 0, a)
+               + kernels.ell_spmv_dot(ell, pieces, q2, p, 0, b, -1))
+    assert torch.equal(q1, q2)
+    assert torch.allclose(d_full, d_split, rtol=1e-12)
+    # interior run over a mid-window piece: valid even bounds, all rows'
+    # windows inside
+    ia, ib = kernels.ell_interior(ell, 1000, 4000)
+    if ib > ia:
+        assert ia % 2 == 0 and ib % 2 == 0
+        assert bool((ell.rmin[ia:ib] >= 1000).all())
+        assert bool((ell.rmax[ia:ib] < 4000).all())
+
+
 @pytest.mark.parametrize("dt", types)
 def test_dense_to_csr_kernel_gpu(dt):
     """Two-phase ballot-compaction dense->CSR kernel (reference
