@@ -69,3 +69,48 @@ def test_conversion_roundtrips(m, n, density, seed):
     assert np.allclose(np.asarray(A.T.todense()), s.T.toarray())
     d = np.asarray(A.diagonal())
     assert np.allclose(d, s.diagonal())
+
+
+@settings(max_examples=15, deadline=None, derandomize=True)
+@given(m=st.integers(1, 30), n=st.integers(1, 30), d=st.floats(0.0, 0.5),
+       seed=st.integers(0, 10**6), dt=st.sampled_from([np.float64, np.complex128]))
+def test_mmread_roundtrip_matches_scipy(m, n, d, seed, dt):
+    """scipy mmwrite -> byte-range-parallel mmread (random shapes incl.
+    empty matrices, complex fields)."""
+    import os
+    import tempfile
+
+    import scipy.io as spio
+
+    import sparse
+
+    s = sps.random(m, n, d, random_state=seed, format="coo").astype(dt)
+    if dt == np.complex128 and s.nnz:
+        s.data = s.data + 1j * np.random.default_rng(seed).random(s.nnz)
+    fd, path = tempfile.mkstemp(suffix=".mtx")
+    os.close(fd)
+    try:
+        spio.mmwrite(path, s)
+        back = sparse.io.mmread(path).tocsr().to_scipy_sparse_csr()
+    finally:
+        os.remove(path)
+    ref = s.tocsr()
+    ref.sort_indices()
+    assert back.shape == ref.shape
+    assert np.allclose(back.toarray(), ref.toarray())
+
+
+@settings(max_examples=15, deadline=None, derandomize=True)
+@given(m=st.integers(1, 25), n=st.integers(1, 25), k=st.integers(1, 6),
+       d=st.floats(0.05, 0.5), seed=st.integers(0, 10**6))
+def test_sddmm_matches_oracle(m, n, k, d, seed):
+    s = _rand(m, n, d, seed, np.float64)
+    rng = np.random.default_rng(seed)
+    C = rng.random((m, k))
+    D = rng.random((k, n))
+    out = csr_array(s).sddmm(C, D)
+    coo = s.tocoo()
+    exp = s.multiply(sps.coo_matrix(((C @ D)[coo.row, coo.col],
+                                     (coo.row, coo.col)), shape=s.shape))
+    assert np.allclose(np.asarray(out.todense()), exp.toarray(),
+                       rtol=1e-10, atol=1e-12)
