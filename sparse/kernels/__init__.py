@@ -170,10 +170,12 @@ def spgemm_csr(A, B, a_col_lo, vdt):
     acounts = A.indptr[1:] - A.indptr[:-1]
     acols = A.indices.long() - a_col_lo
     bcounts = B.indptr[1:] - B.indptr[:-1]
-    rows_of_nnz = torch.repeat_interleave(
-        torch.arange(m, dtype=torch.int64, device=dev), acounts)
-    ub = torch.zeros(m, dtype=torch.int64, device=dev)
-    ub.index_add_(0, rows_of_nnz, bcounts[acols])
+    # per-row product upper bound as a SEGMENT sum (cumsum + indptr
+    # gather): rows_of_nnz + index_add_ was an atomic scatter over nnz
+    # elements — 79 ms/call on the 931M-nnz 3-D GMG Galerkin setup
+    cs = torch.zeros(A.nnz + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(bcounts[acols], 0, out=cs[1:])
+    ub = cs[A.indptr[1:]] - cs[A.indptr[:-1]]
     counts = torch.zeros(m, dtype=torch.int64, device=dev)
     bins = [((ub <= 32), 64), ((ub > 32) & (ub <= 128), 256),
             ((ub > 128) & (ub <= 512), 1024),
