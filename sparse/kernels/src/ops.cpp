@@ -17,6 +17,8 @@ void axpby_norm2_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, bool,
                      bool, at::Tensor);
 void build_ell_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                    int64_t, int64_t);
+void build_dia_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor, at::Tensor,
+                   int64_t, int64_t);
 void ell_spmv_plain_hip(at::Tensor, at::Tensor, at::Tensor, at::Tensor,
                         at::Tensor, at::Tensor, int64_t, int64_t, int64_t,
                         int64_t, int64_t);
@@ -106,6 +108,8 @@ TORCH_LIBRARY(sparse_hip, m) {
         "bool negate, Tensor(b!) dot_out) -> ()");
   m.def("build_ell(Tensor indptr, Tensor indices, Tensor values, "
         "Tensor(a!) eidx, Tensor(b!) evals, int W, int pad_idx) -> ()");
+  m.def("build_dia(Tensor indptr, Tensor indices, Tensor values, "
+        "Tensor offs, Tensor(a!) dvals, int W, int row0) -> ()");
   m.def("ell_spmv(Tensor eidx, Tensor evals, Tensor hlo, Tensor own, "
         "Tensor hhi, Tensor(a!) y, int W, int m, int col_lo, "
         "int rbase, int rhi) -> ()");
@@ -180,6 +184,7 @@ TORCH_LIBRARY_IMPL(sparse_hip, CUDA, m) {
   m.impl("csr_row_spmv", csr_row_spmv_hip);
   m.impl("axpby_norm2", axpby_norm2_hip);
   m.impl("build_ell", build_ell_hip);
+  m.impl("build_dia", build_dia_hip);
   m.impl("ell_spmv", ell_spmv_plain_hip);
   m.impl("ell_spmv_dot", ell_spmv_dot_hip);
   m.impl("ell_jacobi", ell_jacobi_hip);

@@ -297,6 +297,32 @@ __global__ void build_ell_kernel(const int64_t* __restrict__ indptr,
   }
 }
 
+// scatter the CSR values into the padded diagonal planes (dvals assumed
+// pre-zeroed): one thread per ROW walking its short slice — replaces an
+// nnz-scale torch index_put_ in the mirror build (indexFuncLargeIndex was
+// ~79 ms/call on the 931M-nnz 3-D GMG fine level)
+template <typename T, typename index_t>
+__global__ void build_dia_kernel(const int64_t* __restrict__ indptr,
+                                 const index_t* __restrict__ indices,
+                                 const T* __restrict__ vals,
+                                 const int64_t* __restrict__ offs,
+                                 T* __restrict__ dvals, int64_t m, int64_t mp,
+                                 int W, int64_t row0) {
+  const int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (r >= m) return;
+  const int64_t e = indptr[r + 1];
+  for (int64_t p = indptr[r]; p < e; ++p) {
+    const int64_t diag = (int64_t)indices[p] - (row0 + r);
+    // offs sorted ascending, W <= 48: binary search
+    int lo = 0, hi = W;
+    while (lo < hi) {
+      const int mid = (lo + hi) >> 1;
+      if (offs[mid] < diag) lo = mid + 1; else hi = mid;
+    }
+    dvals[(int64_t)lo * mp + r] = vals[p];
+  }
+}
+
 // nt_load (common.h): measured on the 16384^2 5-pt DIA SpMV:
 // 2.580 -> 2.474 ms (tools/dia_nt_bench.hip); nt STORES measured slower,
 // so outputs use plain stores.
@@ -452,6 +478,25 @@ void ell_spmv_hip(at::Tensor eidx, at::Tensor evals, at::Tensor hlo,
         launch(ell_spmv_kernel<T, index_t, false, true>, nullptr, nullptr);
       else
         launch(ell_spmv_kernel<T, index_t, false, false>, nullptr, nullptr);
+    });
+  });
+}
+
+void build_dia_hip(at::Tensor indptr, at::Tensor indices, at::Tensor values,
+                   at::Tensor offs, at::Tensor dvals, int64_t W,
+                   int64_t row0) {
+  const int64_t m = indptr.numel() - 1;
+  const int64_t mp = dvals.numel() / W;
+  if (m == 0) return;
+  DISPATCH_VALUES(values.scalar_type(), "build_dia", [&] {
+    using T = scalar_t;
+    DISPATCH_INDEX(indices.scalar_type(), "build_dia_idx", [&] {
+      hipLaunchKernelGGL((build_dia_kernel<T, index_t>),
+                         dim3((m + 255) / 256), dim3(256), 0, cur_stream(),
+                         indptr.data_ptr<int64_t>(),
+                         indices.data_ptr<index_t>(), values.data_ptr<T>(),
+                         offs.data_ptr<int64_t>(), dvals.data_ptr<T>(), m, mp,
+                         (int)W, row0);
     });
   });
 }
