@@ -30,8 +30,8 @@ from .parallel import comm
 
 __all__ = [
     "LinearOperator", "IdentityOperator", "aslinearoperator", "cg", "cgs",
-    "bicg", "bicgstab", "gmres", "lsqr", "eigsh", "spsolve", "cg_axpby",
-    "norm",
+    "bicg", "bicgstab", "gmres", "minres", "lsqr", "eigsh", "spsolve",
+    "cg_axpby", "norm",
 ]
 
 
@@ -527,6 +527,86 @@ class _Basis:
             else w_local @ w_local
         comm.all_reduce_(s)
         return math.sqrt(max(float(s.item()), 0.0))
+
+
+# -- MINRES -------------------------------------------------------------------
+@track_provenance(nested=True)
+def minres(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None,
+           atol=None):
+    """Minimum-residual method for symmetric (possibly indefinite) systems
+    (Paige & Saunders Lanczos formulation; scipy.sparse.linalg.minres
+    surface).  API superset of the reference, which has no indefinite
+    solver — its CG diverges on saddle-point operators."""
+    A = aslinearoperator(A)
+    b = _vec(b, A.dtype)
+    n = b.shape[0]
+    maxiter = maxiter or 5 * n
+    ident_M = M is None or isinstance(M, IdentityOperator)
+    Mop = aslinearoperator(M) if not ident_M else None
+    x = _vec(x0, A.dtype).copy() if x0 is not None else darray.zeros(
+        (n,), dtype=A.dtype)
+    r1 = b - A.matvec(x)
+    y = r1 if ident_M else Mop.matvec(r1)
+    beta1 = float(r1.dot(y).item())
+    if beta1 < 0:
+        raise ValueError("minres: preconditioner M is not positive definite")
+    if beta1 == 0:
+        return x, 0
+    beta1 = math.sqrt(beta1)
+    bnorm = float(b.norm().item()) or 1.0
+    threshold = _tols(bnorm, tol, atol)
+    oldb = 0.0
+    beta = beta1
+    dbar = epsln = phi = 0.0
+    phibar = beta1
+    cs = -1.0
+    sn = 0.0
+    w = darray.zeros((n,), dtype=A.dtype)
+    w2 = darray.zeros((n,), dtype=A.dtype)
+    r2 = r1
+    info = maxiter
+    for itn in range(1, maxiter + 1):
+        s = 1.0 / beta
+        v = y * s
+        y = A.matvec(v)
+        if itn >= 2:
+            y = y - r1 * (beta / oldb)
+        alfa = float(v.dot(y).item())
+        y = y - r2 * (alfa / beta)
+        r1 = r2
+        r2 = y
+        y = r2 if ident_M else Mop.matvec(r2)
+        oldb = beta
+        beta = float(r2.dot(y).item())
+        if beta < 0:
+            raise ValueError("minres: M is not positive definite")
+        beta = math.sqrt(beta)
+        # QR update of the tridiagonal (plane rotations)
+        oldeps = epsln
+        delta = cs * dbar + sn * alfa
+        gbar = sn * dbar - cs * alfa
+        epsln = sn * beta
+        dbar = -cs * beta
+        gamma = math.sqrt(gbar * gbar + beta * beta) or np.finfo(float).eps
+        cs = gbar / gamma
+        sn = beta / gamma
+        phi = cs * phibar
+        phibar = sn * phibar
+        # update x
+        w1 = w2
+        w2 = w
+        w = (v - w1 * oldeps - w2 * delta) * (1.0 / gamma)
+        x += w * phi
+        if callback is not None:
+            callback(x)
+        if abs(phibar) < threshold:
+            info = 0
+            break
+    if info != 0:
+        r = b - A.matvec(x)
+        if float(r.norm().item()) < threshold:
+            info = 0
+    return x, info
 
 
 # -- GMRES --------------------------------------------------------------------

@@ -199,3 +199,29 @@ def test_lsqr_complex_rmatvec():
     x = rng.random(m) + 1j * rng.random(m)
     op = linalg.aslinearoperator(A)
     assert np.allclose(np.asarray(op.rmatvec(x)), s.conj().T @ x)
+
+
+def test_minres_indefinite():
+    """minres solves symmetric INDEFINITE systems where CG diverges
+    (API superset: the reference has no indefinite solver)."""
+    n = 60
+    rng = np.random.default_rng(29)
+    m = sps.random(n, n, 0.2, random_state=30)
+    a = (m + m.T).tocsr()
+    a = (a + sps.diags(np.where(np.arange(n) % 2 == 0, 8.0, -8.0))).tocsr()
+    b = rng.random(n)
+    x, info = linalg.minres(csr_array(a), b, tol=1e-10, maxiter=2000)
+    assert info == 0
+    assert np.allclose(a @ np.asarray(x), b, atol=1e-5)
+
+
+def test_minres_spd_matches_scipy():
+    import scipy.sparse.linalg as spla
+
+    n = 50
+    s = spd_csr(n, seed=31)
+    b = sample_dense(n, seed=32)
+    x, info = linalg.minres(csr_array(s), b, tol=1e-12)
+    xs, _ = spla.minres(s, b, rtol=1e-12)
+    assert info == 0
+    assert np.allclose(np.asarray(x), xs, atol=1e-6)
