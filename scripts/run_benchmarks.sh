@@ -1,5 +1,5 @@
 #!/bin/bash
-# Reproduce the numbers in profiles/RESULTS_r01.md on an MI355X box.
+# Reproduce the numbers in profiles/RESULTS_r02.md (and _r01) on an MI355X box.
 # Usage: bash scripts/run_benchmarks.sh [quick]
 set -x
 QUICK=${1:-}
@@ -23,8 +23,17 @@ python examples/gmg.py -N 511 -dim 3 -maxiter 60
 python examples/amg.py -n 1048576 -maxiter 200
 python examples/spgemm_microbenchmark.py -nx 2047 -iters 10
 
-# quantum MIS demo
-python examples/quantum_mis.py -l 4 -T 3.0
+# quantum MIS demo (6x6 = FULL 5.6M-state Hilbert space; 9x9 truncated)
+python examples/quantum_mis.py -l 6 -T 4.0
+python examples/quantum_mis.py -l 9 -T 4.0 -kmax 4
+
+# multi-vector SpMM: auto BSR-MFMA route vs lane-tiled (profiles/MFMA_r02.md)
+python examples/dot_microbenchmark.py -op spmm -k 32 -n 4000000 -iters 50 -warmup 10
+SPARSE_NO_BSR=1 python examples/dot_microbenchmark.py -op spmm -k 32 -n 4000000 -iters 50 -warmup 10
+
+# MFMA kernel A/B + lane-map probe (standalone)
+hipcc --offload-arch=gfx950 -O3 tools/mfma_spmm_bench.hip -o /tmp/mfma_bench
+/tmp/mfma_bench --probe && /tmp/mfma_bench 1048576 30
 
 # capacity: 604M rows / 3.02B nnz on ONE GPU (288 GB sizing)
 python bench.py --nx 24576 --steps 30 --warmup 5
