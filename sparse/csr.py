@@ -633,7 +633,20 @@ class csr_array(CompressedBase, DenseSparseBase):
             from . import kernels
 
             bm = self._bsr()
-            if kernels.bsr_profitable(bm, k):
+            use_bsr = kernels.bsr_profitable(bm, k)
+            if comm.world_size() > 1:
+                # block fill differs per slab: the aligned-window plan
+                # construction below is COLLECTIVE, so the route must be
+                # unanimous (all-reduce(min) vote, cached — profitability
+                # depends only on the fixed fill and the k<=32 bucket)
+                vkey = ("bsr_vote", k <= 32)
+                if vkey not in self._plan_cache:
+                    flag = torch.tensor([1.0 if use_bsr else 0.0],
+                                        device=self._values.device)
+                    comm.all_reduce_(flag, op="min")
+                    self._plan_cache[vkey] = bool(flag.item() > 0.5)
+                use_bsr = self._plan_cache[vkey]
+            if use_bsr:
                 # MFMA path reads whole 16-row blocks of B: use a
                 # block-aligned window (profiles/MFMA_r02.md win region)
                 lo, hi = self._col_window()

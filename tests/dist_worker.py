@@ -395,6 +395,12 @@ def extra_banded_overlap_check():
     rr = A.residual(x, _d.asdistarray(np.ones(n)))
     assert np.allclose(np.asarray(rr), 1.0 - sref @ np.asarray(x),
                        rtol=1e-10), "banded dist residual"
+    # multi-vector SpMM at ws>1: on a GPU box this takes the BSR-MFMA
+    # route behind the unanimous-vote gate (collective-safe); CPU runs
+    # the plain path — both must match the oracle
+    Bm = np.asarray(_d.random((n, 16), seed=58).gather())
+    got = np.asarray(A @ _d.asdistarray(Bm))
+    assert np.allclose(got, sref @ Bm, rtol=1e-10), "banded dist spmm k=16"
     bb = _d.random((n,), seed=56)
     dinv = _d.asdistarray(1.0 / sref.diagonal())
     outj = A.jacobi_smooth(x, bb, dinv, 0.7)
