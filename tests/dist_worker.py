@@ -398,7 +398,7 @@ def extra_banded_overlap_check():
     # multi-vector SpMM at ws>1: on a GPU box this takes the BSR-MFMA
     # route behind the unanimous-vote gate (collective-safe); CPU runs
     # the plain path — both must match the oracle
-    Bm = np.asarray(_d.random((n, 16), seed=58).gather())
+    Bm = _d.random((n, 16), seed=58).gather().cpu().numpy()
     got = np.asarray(A @ _d.asdistarray(Bm))
     assert np.allclose(got, sref @ Bm, rtol=1e-10), "banded dist spmm k=16"
     bb = _d.random((n,), seed=56)
