@@ -30,8 +30,8 @@ from .parallel import comm
 
 __all__ = [
     "LinearOperator", "IdentityOperator", "aslinearoperator", "cg", "cgs",
-    "bicg", "bicgstab", "gmres", "minres", "lsqr", "eigsh", "spsolve",
-    "cg_axpby", "norm",
+    "bicg", "bicgstab", "gmres", "minres", "lsqr", "eigsh", "svds",
+    "spsolve", "cg_axpby", "norm",
 ]
 
 
@@ -527,6 +527,33 @@ class _Basis:
             else w_local @ w_local
         comm.all_reduce_(s)
         return math.sqrt(max(float(s.item()), 0.0))
+
+
+def svds(A, k=6, tol=0, maxiter=None, return_singular_vectors=True):
+    """Largest-k singular triplets via thick-restart Lanczos on the Gram
+    operator A^H A (scipy.sparse.linalg.svds surface; API superset — the
+    reference has no SVD).  Returns (U, s, Vh) ascending in s like scipy."""
+    Aop = aslinearoperator(A)
+    m, n = Aop.shape
+    if k <= 0 or k >= min(m, n):
+        raise ValueError("k must be in (0, min(shape))")
+
+    def gram(x, out=None):
+        return Aop.rmatvec(Aop.matvec(x), out=out)
+
+    G = LinearOperator((n, n), matvec=gram, dtype=Aop.dtype)
+    w, V = eigsh(G, k=k, which="LA", tol=tol, maxiter=maxiter)
+    s = np.sqrt(np.maximum(w, 0.0))
+    order = np.argsort(s)  # scipy returns ascending
+    s = s[order]
+    V = V[:, order]
+    if not return_singular_vectors:
+        return s
+    U = np.empty((m, k), dtype=V.dtype)
+    for i in range(k):
+        av = np.asarray(Aop.matvec(V[:, i]))
+        U[:, i] = av / s[i] if s[i] > 0 else 0.0
+    return U, s, V.conj().T
 
 
 # -- MINRES -------------------------------------------------------------------

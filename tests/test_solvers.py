@@ -225,3 +225,19 @@ def test_minres_spd_matches_scipy():
     xs, _ = spla.minres(s, b, rtol=1e-12)
     assert info == 0
     assert np.allclose(np.asarray(x), xs, atol=1e-6)
+
+
+def test_svds_matches_scipy():
+    """svds (API superset): largest-k singular triplets vs scipy."""
+    import scipy.sparse.linalg as spla
+
+    m, n, k = 60, 45, 4
+    s = sps.random(m, n, 0.3, random_state=40, format="csr")
+    U, sv, Vh = linalg.svds(csr_array(s), k=k, tol=1e-10)
+    sv_ref = spla.svds(s, k=k, return_singular_vectors=False)
+    assert np.allclose(np.sort(sv), np.sort(sv_ref), atol=1e-6)
+    # triplet consistency: A ~ U diag(s) Vh on the captured subspace
+    assert np.allclose(s @ Vh.conj().T, U * sv, atol=1e-6)
+    # orthonormal factors
+    assert np.allclose(Vh @ Vh.conj().T, np.eye(k), atol=1e-8)
+    assert np.allclose(U.T @ U, np.eye(k), atol=1e-6)
