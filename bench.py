@@ -97,12 +97,22 @@ def make_cg_stepper(A, b, n, dtype, on_gpu, allow_graph):
 
     else:
 
-        def cg_step_gpu() -> None:  # fused 4-kernel path (see linalg.cg)
+        use3 = not os.environ.get("SPARSE_CG4")
+
+        def cg_step_gpu() -> None:  # fused 3-kernel path (see linalg.cg)
             # rz_buf is a persistent 0-dim buffer so the step is
             # hipGraph-capturable (fixed addresses across replays)
             pq = A.spmv_dot(pvec, q)
-            linalg.cg_axpby(xv, pvec, rz_buf, pq, isalpha=True, negate=False)
-            rz_new = linalg._axpby_norm2(r, q, rz_buf, pq, negate=True)
+            if use3:
+                # K2 fuses x += (rz/pq)p and r -= (rz/pq)q + |r|^2: same
+                # HBM bytes as the two axpbys it replaces, one kernel less
+                rz_new = kernels.cg_xr_norm2(xv.local, pvec.local, r.local,
+                                             q.local, rz_buf, pq)
+                comm.all_reduce_(rz_new)
+            else:  # SPARSE_CG4=1: the round-1 4-kernel formulation
+                linalg.cg_axpby(xv, pvec, rz_buf, pq, isalpha=True,
+                                negate=False)
+                rz_new = linalg._axpby_norm2(r, q, rz_buf, pq, negate=True)
             linalg.cg_axpby(pvec, r, rz_new, rz_buf, isalpha=False, negate=False)
             rz_buf.copy_(rz_new)
 

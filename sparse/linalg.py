@@ -294,9 +294,18 @@ def cg(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None, atol=None,
         return x, info
     for i in range(maxiter):
         if fused:
+            from . import kernels
+
+            # 3-kernel iteration: SpMV+p·Ap fused; x/r updates + |r|^2
+            # fused (cg_xr_norm2); p-update axpby.  Same HBM bytes as the
+            # 4-kernel form (measured parity at nx=16384), one launch and
+            # one Python dispatch fewer per iteration — which is what
+            # matters in the eager ws>1 loop.  SPARSE_CG4=1 in bench.py
+            # keeps the 4-kernel formulation for A/B.
             pq = A.A.spmv_dot(p, q)
-            cg_axpby(x, p, rz, pq, isalpha=True, negate=False)
-            rz_new = _axpby_norm2(r, q, rz, pq, negate=True)
+            rz_new = kernels.cg_xr_norm2(x.local, p.local, r.local, q.local,
+                                         rz, pq)
+            comm.all_reduce_(rz_new)
             if conv_test_iters and (i % conv_test_iters == 0 or i == maxiter - 1):
                 if math.sqrt(max(float(rz_new.item()), 0.0)) < threshold:
                     info = 0
