@@ -50,3 +50,26 @@ def test_mmread_skew_symmetric(tmp_path):
                  "3 3 2\n2 1 1.5\n3 1 -2.0\n")
     A = sio_ours.mmread(str(p))
     assert np.allclose(np.asarray(A.todense()), sio.mmread(str(p)).toarray())
+
+
+def test_mmread_hermitian(tmp_path):
+    """Hermitian coordinate files: the stored lower triangle expands with
+    CONJUGATED mirror entries (reference mtx_to_coo.cc symmetry handling)."""
+    import scipy.io as spio
+    import scipy.sparse as sps
+
+    rng = np.random.default_rng(44)
+    n = 12
+    low = sps.random(n, n, 0.3, random_state=45).astype(np.complex128)
+    low.data = low.data + 1j * rng.random(low.nnz)
+    low = sps.tril(low)
+    # make diagonal real so the matrix is a valid hermitian
+    full = low + low.conj().T
+    full.setdiag(full.diagonal().real)
+    p = str(tmp_path / "herm.mtx")
+    spio.mmwrite(p, full.tocoo(), symmetry="hermitian")
+    ours = sio_ours.mmread(p).tocsr().to_scipy_sparse_csr()
+    ref = spio.mmread(p).tocsr()
+    ref.sort_indices()
+    assert np.allclose(ours.toarray(), ref.toarray())
+    assert np.allclose(ours.toarray(), ours.toarray().conj().T)
