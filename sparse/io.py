@@ -153,21 +153,25 @@ def mmwrite(path, A, comment: str = "") -> None:
     from .parallel import comm
 
     c = A.tocoo() if A.format != "coo" else A
-    rows, cols, vals = c.row, c.col, c.data
-    if comm.rank() != 0:
-        return
-    cplx = np.iscomplexobj(vals)
-    field = "complex" if cplx else "real"
-    with open(path, "w") as f:
-        f.write(f"%%MatrixMarket matrix coordinate {field} general\n")
-        if comment:
-            for ln in comment.splitlines():
-                f.write(f"%{ln}\n")
-        f.write(f"{A.shape[0]} {A.shape[1]} {len(vals)}\n")
-        CH = 1 << 24  # bounded formatting buffers at capacity scale
-        for s in range(0, len(vals), CH):
-            f.write(_format_block(rows[s: s + CH], cols[s: s + CH],
-                                  vals[s: s + CH], cplx))
+    rows, cols, vals = c.row, c.col, c.data  # collective gathers: all ranks
+    if comm.rank() == 0:
+        cplx = np.iscomplexobj(vals)
+        field = "complex" if cplx else "real"
+        with open(path, "w") as f:
+            f.write(f"%%MatrixMarket matrix coordinate {field} general\n")
+            if comment:
+                for ln in comment.splitlines():
+                    f.write(f"%{ln}\n")
+            f.write(f"{A.shape[0]} {A.shape[1]} {len(vals)}\n")
+            CH = 1 << 24  # bounded formatting buffers at capacity scale
+            for s in range(0, len(vals), CH):
+                f.write(_format_block(rows[s: s + CH], cols[s: s + CH],
+                                      vals[s: s + CH], cplx))
+    if comm.initialized():
+        # other ranks may mmread the file right after this returns
+        import torch.distributed as dist
+
+        dist.barrier()
 
 
 def save_npz(file, matrix, compressed=True):
