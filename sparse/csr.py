@@ -567,11 +567,21 @@ class csr_array(CompressedBase, DenseSparseBase):
         ell = None if dm is not None else (
             self._ell() if self._values.dtype == vdt and not precise
             else None)
+        # write straight into out when layouts agree — saves a full
+        # vector pass per matvec(out=) (the CG/GMG hot loops)
+        mloc = self.partition.count(comm.rank())
+
+        def _ybuf():
+            if (out is not None and out.local.dtype == vdt
+                    and out.local.is_contiguous()
+                    and out.local.numel() == mloc):
+                return out.local
+            return torch.empty(mloc, dtype=vdt, device=self._values.device)
+
         if dm is not None:
             from . import kernels
 
-            ylocal = torch.empty(self.partition.count(comm.rank()), dtype=vdt,
-                                 device=self._values.device)
+            ylocal = _ybuf()
             ws_ = plan.hi - plan.lo
             if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
                 h = plan.gather_halos_begin(x.local.to(vdt))
@@ -587,8 +597,7 @@ class csr_array(CompressedBase, DenseSparseBase):
         elif ell is not None:
             from . import kernels
 
-            ylocal = torch.empty(self.partition.count(comm.rank()), dtype=vdt,
-                                 device=self._values.device)
+            ylocal = _ybuf()
             a = b = 0
             if comm.world_size() > 1 and not os.environ.get("SPARSE_NO_OVERLAP"):
                 a, b = self._ell_interior(ell, plan)
@@ -613,10 +622,15 @@ class csr_array(CompressedBase, DenseSparseBase):
             if lc.values.dtype != vdt:
                 lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt),
                                   lc.nrows, lc.ncols, lc.max_row_nnz)
-            ylocal = ops.spmv(lc, xw, col_lo=plan.lo)
+            yb = _ybuf()
+            if yb.is_cuda:
+                ylocal = ops.spmv(lc, xw, col_lo=plan.lo, y=yb)
+            else:
+                ylocal = ops.spmv(lc, xw, col_lo=plan.lo)
         part = self.partition
         if out is not None:
-            out.local.copy_(ylocal.to(out.local.dtype))
+            if ylocal.data_ptr() != out.local.data_ptr():
+                out.local.copy_(ylocal.to(out.local.dtype))
             return out
         return DistArray.from_local(ylocal, part, (self.shape[0],))
 
