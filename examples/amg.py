@@ -253,10 +253,9 @@ class _Shim:
 def build_repl_tail(levels, repl_threshold=1 << 16, smooth_iters=2):
     """Replicated coarse tail at ws>1 (sparse.multigrid; same latency plan
     as gmg.py): returns (ReplicatedCoarseCycle | None, cut index)."""
-    from sparse.multigrid import ReplicatedCoarseCycle
+    from sparse.multigrid import ReplicatedCoarseCycle, find_replication_cut
 
-    ri = next((i for i, l in enumerate(levels)
-               if l["A"].shape[0] <= repl_threshold), len(levels))
+    ri = find_replication_cut(levels, repl_threshold)
     if comm.world_size() == 1 or ri >= len(levels):
         return None, len(levels)
     shims = []

@@ -132,13 +132,11 @@ class GMG:
         self.repl = None
         self.ri = len(self.levels)
         if comm.world_size() > 1 and repl_threshold:
-            for i, lvl in enumerate(self.levels):
-                if lvl.A.shape[0] <= repl_threshold:
-                    self.ri = i
-                    break
-            if self.ri < len(self.levels):
-                from sparse.multigrid import ReplicatedCoarseCycle
+            from sparse.multigrid import (ReplicatedCoarseCycle,
+                                          find_replication_cut)
 
+            self.ri = find_replication_cut(self.levels, repl_threshold)
+            if self.ri < len(self.levels):
                 self.repl = ReplicatedCoarseCycle(
                     self.levels[self.ri:], self.coarse_inv_t, smooth_iters)
         self._graph = None
