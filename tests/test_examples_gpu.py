@@ -43,3 +43,22 @@ def test_gmg_variants_gpu():
     assert "info=0" in out
     out = run("gmg.py", "-N", "127", "-gridop", "injection")
     assert "info=0" in out
+
+
+def test_spmm_micro_bsr_gpu():
+    """dot_microbenchmark -op spmm engages the BSR-MFMA route on the
+    banded matrix (and the off-switch produces the same numbers)."""
+    out = run("dot_microbenchmark.py", "-op", "spmm", "-k", "32",
+              "-n", "500000", "-iters", "10", "-warmup", "2")
+    assert "spmms" in out
+
+
+def test_quantum_kmax_gpu():
+    out = run("quantum_mis.py", "-l", "6", "-T", "1.0", "-kmax", "6",
+              timeout=400)
+    assert "approx ratio" in out
+
+
+def test_gmg_3d_gpu():
+    out = run("gmg.py", "-N", "127", "-dim", "3", timeout=400)
+    assert "info=0" in out
