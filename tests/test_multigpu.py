@@ -82,6 +82,33 @@ def test_gloo_on_gpu_battery():
 
 
 @pytest.mark.gpu
+def test_gloo_on_gpu_gmg_repl_tail():
+    """gmg.py at ws=2 on one GPU (gloo wire): the replicated coarse tail
+    runs hipGraph-captured with device tensors on both ranks, with the
+    fine level distributed (repl_threshold forces a mid-hierarchy cut)."""
+    if _ngpus() < 1:
+        pytest.skip("needs a GPU")
+    argv = [sys.executable, os.path.join(ROOT, "examples", "gmg.py"),
+            "-N", "255", "-repl_threshold", "20000", "-maxiter", "50"]
+    env = dict(os.environ)
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29747",
+                "WORLD_SIZE": "2", "SPARSE_DIST_BACKEND": "gloo",
+                "GLOO_SOCKET_IFNAME": env.get("GLOO_SOCKET_IFNAME", "lo")})
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e.update({"RANK": str(r), "LOCAL_RANK": str(r)})
+        procs.append(subprocess.Popen(argv, env=e, stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT, cwd=ROOT))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=600)
+        outs.append(out.decode(errors="replace"))
+        assert p.returncode == 0, outs[-1][-2500:]
+    assert "info=0" in outs[0], outs[0][-500:]
+
+
+@pytest.mark.gpu
 def test_gloo_on_gpu_bench_ws2():
     """bench.py at ws=2 with both ranks on one GPU (gloo wire): the exact
     multi-rank bench code path (DIA interior/boundary split, halo
