@@ -114,3 +114,33 @@ def test_sddmm_matches_oracle(m, n, k, d, seed):
                                      (coo.row, coo.col)), shape=s.shape))
     assert np.allclose(np.asarray(out.todense()), exp.toarray(),
                        rtol=1e-10, atol=1e-12)
+
+
+@settings(max_examples=12, deadline=None, derandomize=True)
+@given(n=st.integers(5, 50), d=st.floats(0.05, 0.4), seed=st.integers(0, 10**6),
+       solver=st.sampled_from(["cg", "minres", "gmres", "bicgstab"]))
+def test_solvers_converge_on_spd(n, d, seed, solver):
+    """Random SPD systems: every solver must reach the oracle solution."""
+    from sparse import linalg
+
+    r = sps.random(n, n, d, random_state=seed, format="csr")
+    s = (r + r.T + 2 * n * sps.eye(n)).tocsr()
+    b = np.random.default_rng(seed).random(n)
+    fn = getattr(linalg, solver)
+    kw = {"tol": 1e-11, "maxiter": 3000}
+    if solver in ("cg", "bicgstab"):
+        kw["conv_test_iters"] = 1
+    x, info = fn(csr_array(s), b, **kw)
+    assert info == 0, solver
+    assert np.allclose(s @ np.asarray(x), b, atol=1e-6), solver
+
+
+@settings(max_examples=12, deadline=None, derandomize=True)
+@given(m=st.integers(2, 25), n=st.integers(2, 25), d=st.floats(0.1, 0.6),
+       seed=st.integers(0, 10**6))
+def test_transpose_roundtrips(m, n, d, seed):
+    s = _rand(m, n, d, seed, np.float64)
+    A = csr_array(s)
+    assert np.allclose(np.asarray(A.T.T.todense()), s.toarray())
+    assert np.allclose(np.asarray(A.T.tocsr().todense()), s.T.toarray())
+    assert np.allclose(np.asarray(A.tocsc().tocsr().todense()), s.toarray())
