@@ -572,7 +572,12 @@ def minres(A, b, x0=None, tol=1e-5, maxiter=None, M=None, callback=None,
     """Minimum-residual method for symmetric (possibly indefinite) systems
     (Paige & Saunders Lanczos formulation; scipy.sparse.linalg.minres
     surface).  API superset of the reference, which has no indefinite
-    solver — its CG diverges on saddle-point operators."""
+    solver — its CG diverges on saddle-point operators.
+
+    The Lanczos recurrence consumes alfa/beta on the host each step (the
+    Givens chain is inherently sequential), so this solver syncs ~3x per
+    iteration — same as scipy; the asynchrony discipline of cg applies
+    only to its fused path."""
     A = aslinearoperator(A)
     b = _vec(b, A.dtype)
     n = b.shape[0]

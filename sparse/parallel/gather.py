@@ -133,11 +133,13 @@ class WindowGatherPlan:
         import torch.distributed as dist
 
         send_views, recv_views, hlo, own, hhi, p2p_dev = self._halo_ctx(xlocal)
-        if p2p_dev or not any(
-                v is not None and v.numel() for v in send_views) and not any(
-                v is not None for v in recv_views):
+        nothing_to_exchange = (
+            not any(v is not None and v.numel() for v in send_views)
+            and not any(v is not None for v in recv_views))
+        if p2p_dev or nothing_to_exchange:
             # device/host-native wire: prebuilt descriptors, zero per-call
-            # Python construction
+            # Python construction (p2p_dev is empty only when there is
+            # nothing to exchange or the wire needs host staging below)
             reqs = dist.batch_isend_irecv(p2p_dev) if p2p_dev else []
             return ("pending", reqs, {}, recv_views, (hlo, own, hhi))
         # gloo cannot move CUDA tensors: stage the halos through the host
