@@ -672,12 +672,18 @@ class csr_array(CompressedBase, DenseSparseBase):
                         lo16, hi16, B.partition)
                 aplan = self._plan_cache[key]
                 Bw = aplan.gather(B.local.to(vdt)).contiguous()
-                Clocal = torch.empty(
-                    (self.partition.count(comm.rank()), k), dtype=vdt,
-                    device=self._values.device)
+                mloc = self.partition.count(comm.rank())
+                if (out is not None and out.local.dtype == vdt
+                        and out.local.is_contiguous()
+                        and tuple(out.local.shape) == (mloc, k)):
+                    Clocal = out.local  # direct write: no output copy pass
+                else:
+                    Clocal = torch.empty((mloc, k), dtype=vdt,
+                                         device=self._values.device)
                 kernels.bsr_spmm(bm, Bw, Clocal, aplan.lo)
                 if out is not None:
-                    out.local.copy_(Clocal.to(out.local.dtype))
+                    if Clocal.data_ptr() != out.local.data_ptr():
+                        out.local.copy_(Clocal.to(out.local.dtype))
                     return out
                 return DistArray.from_local(Clocal, self.partition,
                                             (self.shape[0], k))
@@ -689,9 +695,15 @@ class csr_array(CompressedBase, DenseSparseBase):
         if lc.values.dtype != vdt:
             lc = ops.LocalCSR(lc.indptr, lc.indices, lc.values.to(vdt),
                               lc.nrows, lc.ncols, lc.max_row_nnz)
-        Clocal = ops.spmm(lc, Bw, col_lo=plan.lo)
+        Cbuf = None
+        if (out is not None and out.local.dtype == vdt
+                and out.local.is_contiguous() and out.local.dim() == 2
+                and out.local.is_cuda):
+            Cbuf = out.local
+        Clocal = ops.spmm(lc, Bw, col_lo=plan.lo, C=Cbuf)
         if out is not None:
-            out.local.copy_(Clocal.to(out.local.dtype))
+            if Clocal.data_ptr() != out.local.data_ptr():
+                out.local.copy_(Clocal.to(out.local.dtype))
             return out
         return DistArray.from_local(Clocal, self.partition, (self.shape[0], B.shape[1]))
 

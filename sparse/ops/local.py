@@ -125,11 +125,15 @@ def spmv(A: LocalCSR, x: torch.Tensor, col_lo: int = 0,
 
 
 # -- SpMM / rSpMM / SDDMM -----------------------------------------------------
-def spmm(A: LocalCSR, B: torch.Tensor, col_lo: int = 0) -> torch.Tensor:
-    """C = A @ B[window rows].  Reference: SPMM_CSR_DENSE (spmm.cu:26-117)."""
+def spmm(A: LocalCSR, B: torch.Tensor, col_lo: int = 0,
+         C: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """C = A @ B[window rows].  Reference: SPMM_CSR_DENSE (spmm.cu:26-117).
+    Pass C to write the result in place (no output copy)."""
     vdt = torch.promote_types(A.dtype, B.dtype)
     if is_gpu(A.values):
-        C = torch.empty((A.nrows, B.shape[1]), dtype=vdt, device=A.device)
+        if C is None or C.dtype != vdt or tuple(C.shape) != (A.nrows,
+                                                             B.shape[1]):
+            C = torch.empty((A.nrows, B.shape[1]), dtype=vdt, device=A.device)
         hip().spmm(A, B.contiguous(), C, col_lo)
         return C
     m = A.to_scipy(col_lo=col_lo, width=B.shape[0])
