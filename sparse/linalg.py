@@ -725,7 +725,13 @@ def gmres(A, b, x0=None, tol=1e-5, restart=None, maxiter=None, M=None,
 def lsqr(A, b, damp=0.0, atol=1e-8, btol=1e-8, conlim=1e8, iter_lim=None,
          show=False, calc_var=False, x0=None):
     """Golub-Kahan LSQR (reference linalg.py:937-1415, itself scipy-lifted).
-    Returns the scipy 10-tuple."""
+    Returns the scipy 10-tuple.
+
+    The bidiagonalization consumes alfa/beta on the host each step (each
+    normalizes the vector the NEXT matvec consumes), so ~2 syncs per
+    iteration are inherent to the algorithm — same as scipy and the
+    reference; the batched-reduction discipline applies to cg/gmres/eigsh
+    where the recurrences permit it."""
     Aop = aslinearoperator(A)
     b = _vec(b, np.promote_types(Aop.dtype, np.float64))
     m, n = Aop.shape
