@@ -121,9 +121,12 @@ class dia_array(CompressedBase):
         hits = np.where(self._offsets == k)[0]
         if hits.size:
             row = self._data[int(hits[0])]
-            # dia data layout: data[d, j] is the value at column j of diag d
+            # dia data layout: data[d, j] is the value at column j of diag d;
+            # the stored width may be SHORTER than s+dlen (scipy truncates
+            # trailing zeros) — missing entries are zero
             s = max(k, 0)
-            out.copy_(row[s: s + dlen])
+            avail = max(0, min(dlen, row.shape[0] - s))
+            out[:avail].copy_(row[s: s + avail])
         return DistArray.from_global(out)
 
     def transpose(self, axes=None, copy=False):
@@ -135,6 +138,8 @@ class dia_array(CompressedBase):
         new_data = np.zeros((ndiag, new_width), dtype=data.dtype)
         for d, k in enumerate(self._offsets):
             length = min(m + min(k, 0), n - max(k, 0))
+            # stored width may truncate trailing zeros (scipy semantics)
+            length = min(length, width - max(k, 0))
             if length <= 0:
                 continue
             # entries of diag k live at data[d, max(k,0) : max(k,0)+length]
@@ -157,6 +162,8 @@ class dia_array(CompressedBase):
         rows_l, cols_l, vals_l = [], [], []
         for d, k in enumerate(self._offsets):
             length = min(m + min(k, 0), n - max(k, 0))
+            # stored width may truncate trailing zeros (scipy semantics)
+            length = min(length, data.shape[1] - max(k, 0))
             if length <= 0:
                 continue
             j = np.arange(max(k, 0), max(k, 0) + length)

@@ -182,3 +182,24 @@ def test_csc_sddmm():
     out = A.sddmm(C, D)
     assert np.allclose(np.asarray(out.todense()),
                        s.multiply(C @ D).toarray())
+
+
+def test_dia_truncated_data_width():
+    """scipy DIA data may be narrower than max(k,0)+diag_length (trailing
+    zeros truncated): tocoo/tocsr/transpose/diagonal must treat missing
+    entries as zero (fuzz-found regression)."""
+    import scipy.sparse as sps
+
+    from sparse import csr_array
+
+    rng = np.random.default_rng(987654)
+    for _ in range(5):
+        rng.integers(1, 50, 3)
+    m = n = 14
+    a = sps.random(m, n, 0.45, random_state=661633, format="csr")
+    a.sort_indices()
+    di = csr_array(a).tocoo().todia()
+    assert np.allclose(np.asarray(di.todense()), a.toarray())
+    assert np.allclose(np.asarray(di.tocsr().todense()), a.toarray())
+    assert np.allclose(np.asarray(di.T.todense()), a.T.toarray())
+    assert np.allclose(np.asarray(di.diagonal()), a.diagonal())
