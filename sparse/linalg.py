@@ -871,6 +871,24 @@ def eigsh(a, k=6, which="LM", ncv=None, maxiter=None, tol=0.0):
         maxiter = 10 * n
     if tol == 0:
         tol = np.sqrt(np.finfo(np.float64).eps)
+    if ncv <= k:
+        # k == n-1 (tiny n): the Lanczos basis has no room past k —
+        # materialize the operator by n matvecs and solve densely
+        cols = []
+        for i in range(n):
+            e = np.zeros(n)
+            e[i] = 1.0
+            cols.append(np.asarray(A.matvec(e)))
+        dense = np.stack(cols, axis=1)
+        w_all, V_all = np.linalg.eigh(dense)
+        if which == "LM":
+            idx = np.argsort(np.abs(w_all))[::-1][:k]
+        elif which == "LA":
+            idx = np.argsort(w_all)[::-1][:k]
+        else:
+            idx = np.argsort(w_all)[:k]
+        order = np.argsort(w_all[idx])
+        return w_all[idx][order], V_all[:, idx][:, order]
 
     alpha = np.zeros(ncv)
     beta = np.zeros(ncv)

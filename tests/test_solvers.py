@@ -241,3 +241,16 @@ def test_svds_matches_scipy():
     # orthonormal factors
     assert np.allclose(Vh @ Vh.conj().T, np.eye(k), atol=1e-8)
     assert np.allclose(U.T @ U, np.eye(k), atol=1e-6)
+
+
+def test_eigsh_k_equals_n_minus_1():
+    """Tiny-n edge (fuzz-found): k = n-1 leaves no Lanczos room (ncv==k);
+    eigsh must fall back to a dense solve instead of overrunning the
+    basis."""
+    n = 4
+    s = spd_csr(n, seed=77)
+    w, V = linalg.eigsh(csr_array(s), k=3, which="LA")
+    ws = np.linalg.eigvalsh(s.toarray())
+    assert np.allclose(np.sort(w), ws[-3:], atol=1e-8)
+    for i in range(3):
+        assert np.linalg.norm(s @ V[:, i] - w[i] * V[:, i]) < 1e-8
