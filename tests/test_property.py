@@ -144,3 +144,15 @@ def test_transpose_roundtrips(m, n, d, seed):
     assert np.allclose(np.asarray(A.T.T.todense()), s.toarray())
     assert np.allclose(np.asarray(A.T.tocsr().todense()), s.T.toarray())
     assert np.allclose(np.asarray(A.tocsc().tocsr().todense()), s.toarray())
+
+
+@settings(max_examples=20, deadline=None, derandomize=True)
+@given(m=st.integers(1, 40), n=st.integers(1, 40), d=st.floats(0.0, 0.6),
+       seed=st.integers(0, 10**6))
+def test_format_chain_roundtrip(m, n, d, seed):
+    """csr -> coo -> dia -> csr preserves the matrix (the dia leg found a
+    truncated-data-width bug in round 2)."""
+    s = _rand(m, n, d, seed, np.float64)
+    A = csr_array(s)
+    back = A.tocoo().todia().tocsr()
+    assert np.allclose(np.asarray(back.todense()), s.toarray())
